@@ -1,0 +1,293 @@
+"""Block-partitioned distributed matrix.
+
+Reference equivalent: MatRel's RDD[((rowBlk, colBlk), MLMatrix)] with
+~1000x1000 blocks (SURVEY.md §2.1 "Block-matrix representation"). The
+MI355X-native design differs deliberately:
+
+  * Blocks are LARGE (default 16384^2: bf16 block = 512 MiB) - one MI355X
+    holds 288 GB of HBM3E, and MFMA GEMM efficiency wants >= 8k tiles.
+    Block size is per-matrix, so block-sparse matrices can use finer grids.
+  * A block is a torch.Tensor resident on this rank's GPU (or CPU in test
+    mode). ABSENT blocks are implicit zeros - this is the block-sparse
+    representation (BASELINE.json config 4) and also what block-granular
+    selection pruning produces.
+  * Ownership is a pure function of (block index, scheme, device grid) -
+    there is no shuffle machinery; movement is explicit RCCL collectives
+    issued by the executor.
+
+Supports ragged edge blocks (m, n need not divide block_size).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Iterator, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..config import get_config
+from ..exec.runtime import get_runtime
+from .schemes import BCAST, GRID, Scheme
+
+BlockIndex = Tuple[int, int]
+
+
+def _torch_dtype(name: str) -> torch.dtype:
+    return {
+        "bf16": torch.bfloat16,
+        "fp16": torch.float16,
+        "fp32": torch.float32,
+        "fp64": torch.float64,
+    }[name]
+
+
+class BlockMatrix:
+    """A (possibly distributed) block-partitioned matrix."""
+
+    def __init__(self, nrows: int, ncols: int, block_size: Optional[int] = None,
+                 scheme: Scheme = GRID, dtype: Optional[torch.dtype] = None,
+                 device: Optional[torch.device] = None) -> None:
+        cfg = get_config()
+        rt = get_runtime()
+        self.nrows = int(nrows)
+        self.ncols = int(ncols)
+        self.block_size = int(block_size or cfg.block_size)
+        self.scheme = scheme
+        self.device = device if device is not None else rt.device
+        if dtype is None:
+            dtype = _torch_dtype(cfg.dtype) if self.device.type == "cuda" \
+                else torch.float32
+        self.dtype = dtype
+        # Local (this-rank) blocks only.
+        self.blocks: Dict[BlockIndex, torch.Tensor] = {}
+
+    # --- geometry -------------------------------------------------------
+
+    @property
+    def grid_rows(self) -> int:
+        return max(1, math.ceil(self.nrows / self.block_size))
+
+    @property
+    def grid_cols(self) -> int:
+        return max(1, math.ceil(self.ncols / self.block_size))
+
+    @property
+    def shape(self) -> Tuple[int, int]:
+        return (self.nrows, self.ncols)
+
+    def block_shape(self, bi: int, bj: int) -> Tuple[int, int]:
+        bs = self.block_size
+        h = min(bs, self.nrows - bi * bs)
+        w = min(bs, self.ncols - bj * bs)
+        return (h, w)
+
+    def block_indices(self) -> Iterator[BlockIndex]:
+        for bi in range(self.grid_rows):
+            for bj in range(self.grid_cols):
+                yield (bi, bj)
+
+    # --- ownership ------------------------------------------------------
+
+    def owner(self, bi: int, bj: int) -> int:
+        rt = get_runtime()
+        return self.scheme.owner(bi, bj, rt.world_size, rt.grid)
+
+    def is_local(self, bi: int, bj: int) -> bool:
+        rt = get_runtime()
+        o = self.owner(bi, bj)
+        return o == -1 or o == rt.rank
+
+    def owned_indices(self) -> Iterator[BlockIndex]:
+        for idx in self.block_indices():
+            if self.is_local(*idx):
+                yield idx
+
+    # --- block access ---------------------------------------------------
+
+    def get_block(self, bi: int, bj: int) -> Optional[torch.Tensor]:
+        """Local block or None if absent (= zero) or remote."""
+        return self.blocks.get((bi, bj))
+
+    def get_block_dense(self, bi: int, bj: int) -> torch.Tensor:
+        """Local block, materializing zeros for absent blocks."""
+        t = self.blocks.get((bi, bj))
+        if t is None:
+            t = torch.zeros(self.block_shape(bi, bj), dtype=self.dtype,
+                            device=self.device)
+        return t
+
+    def set_block(self, bi: int, bj: int, t: torch.Tensor) -> None:
+        assert t.shape == self.block_shape(bi, bj), \
+            f"block {(bi, bj)}: {tuple(t.shape)} != {self.block_shape(bi, bj)}"
+        self.blocks[(bi, bj)] = t.to(device=self.device, dtype=self.dtype)
+
+    def local_items(self) -> Iterator[Tuple[BlockIndex, torch.Tensor]]:
+        return iter(self.blocks.items())
+
+    # --- stats (feed the optimizer's sparsity-aware cost model) ---------
+
+    def local_nnz_blocks(self) -> int:
+        return len(self.blocks)
+
+    def block_density(self) -> float:
+        """Fraction of blocks present, across all ranks."""
+        rt = get_runtime()
+        total = self.grid_rows * self.grid_cols
+        if not rt.is_distributed or self.scheme.is_replicated():
+            return len(self.blocks) / total
+        t = torch.tensor([len(self.blocks)], dtype=torch.int64)
+        dist.all_reduce(t)
+        return int(t.item()) / total
+
+    # --- constructors ---------------------------------------------------
+
+    @classmethod
+    def zeros(cls, m: int, n: int, **kw) -> "BlockMatrix":
+        return cls(m, n, **kw)
+
+    @classmethod
+    def _filled(cls, m: int, n: int, fill, seed: Optional[int], **kw
+                ) -> "BlockMatrix":
+        out = cls(m, n, **kw)
+        for (bi, bj) in out.owned_indices():
+            shape = out.block_shape(bi, bj)
+            if seed is not None:
+                g = torch.Generator(device="cpu")
+                g.manual_seed((seed * 1000003 + bi * 131 + bj) & 0x7FFFFFFF)
+                t = fill(shape, g).to(device=out.device, dtype=out.dtype)
+            else:
+                t = fill(shape, None).to(device=out.device, dtype=out.dtype)
+            out.blocks[(bi, bj)] = t
+        return out
+
+    @classmethod
+    def randn(cls, m: int, n: int, seed: Optional[int] = 0, **kw
+              ) -> "BlockMatrix":
+        """Block-reproducible N(0,1) fill: block (i,j) content depends only
+        on (seed, i, j), so the same logical matrix materializes under any
+        scheme / world size (used by tests and the bench)."""
+        return cls._filled(
+            m, n, lambda s, g: torch.randn(*s, generator=g), seed, **kw)
+
+    @classmethod
+    def rand(cls, m: int, n: int, seed: Optional[int] = 0, **kw
+             ) -> "BlockMatrix":
+        return cls._filled(
+            m, n, lambda s, g: torch.rand(*s, generator=g), seed, **kw)
+
+    @classmethod
+    def from_dense(cls, t: torch.Tensor, **kw) -> "BlockMatrix":
+        """Split a full tensor (available on every rank) into blocks."""
+        assert t.dim() == 2
+        out = cls(t.shape[0], t.shape[1], **kw)
+        bs = out.block_size
+        for (bi, bj) in out.owned_indices():
+            blk = t[bi * bs:(bi + 1) * bs, bj * bs:(bj + 1) * bs]
+            if blk.count_nonzero() == 0:
+                continue  # keep zero blocks implicit
+            out.blocks[(bi, bj)] = blk.contiguous().to(
+                device=out.device, dtype=out.dtype)
+        return out
+
+    @classmethod
+    def block_sparse_random(cls, m: int, n: int, block_density: float,
+                            seed: int = 0, **kw) -> "BlockMatrix":
+        """Random block-sparse matrix: each block present with probability
+        block_density, reproducibly by block index (config 4)."""
+        out = cls(m, n, **kw)
+        for (bi, bj) in out.owned_indices():
+            g = torch.Generator(device="cpu")
+            g.manual_seed((seed * 1000003 + bi * 131 + bj) & 0x7FFFFFFF)
+            if torch.rand(1, generator=g).item() >= block_density:
+                continue
+            t = torch.randn(*out.block_shape(bi, bj), generator=g)
+            out.blocks[(bi, bj)] = t.to(device=out.device, dtype=out.dtype)
+        return out
+
+    # --- materialization (small matrices / tests only) ------------------
+
+    def to_dense(self) -> torch.Tensor:
+        """Assemble the full matrix on every rank (collective when
+        distributed). Only for small results and tests."""
+        rt = get_runtime()
+        full = torch.zeros(self.nrows, self.ncols, dtype=torch.float32,
+                           device=self.device)
+        bs = self.block_size
+        for (bi, bj) in self.block_indices():
+            owner = self.owner(bi, bj)
+            if owner == -1 or not rt.is_distributed:
+                t = self.blocks.get((bi, bj))
+                if t is not None:
+                    full[bi * bs:bi * bs + t.shape[0],
+                         bj * bs:bj * bs + t.shape[1]] = t.float()
+            else:
+                shape = self.block_shape(bi, bj)
+                if rt.rank == owner:
+                    t = self.get_block_dense(bi, bj).float()
+                else:
+                    t = torch.zeros(shape, dtype=torch.float32,
+                                    device=self.device)
+                dist.broadcast(t, src=owner)
+                full[bi * bs:bi * bs + shape[0],
+                     bj * bs:bj * bs + shape[1]] = t
+        return full
+
+    # --- scheme movement ------------------------------------------------
+
+    def repartition(self, scheme: Scheme) -> "BlockMatrix":
+        """Move blocks to the ownership map of `scheme`.
+
+        Reference: RDD repartition/shuffle. Here: per-block broadcast from
+        the current owner (xGMI p2p under RCCL). Replication targets
+        (BroadcastScheme) receive every block everywhere.
+        """
+        rt = get_runtime()
+        if scheme == self.scheme:
+            return self
+        out = BlockMatrix(self.nrows, self.ncols, self.block_size, scheme,
+                          self.dtype, self.device)
+        if not rt.is_distributed:
+            out.blocks = dict(self.blocks)
+            return out
+        for (bi, bj) in self.block_indices():
+            src = self.owner(bi, bj)
+            dst = scheme.owner(bi, bj, rt.world_size, rt.grid)
+            if src == -1:
+                # already replicated: every rank has it locally (or zero)
+                if dst == -1 or dst == rt.rank:
+                    t = self.blocks.get((bi, bj))
+                    if t is not None:
+                        out.blocks[(bi, bj)] = t
+                continue
+            # presence flag must be agreed: broadcast nnz flag first
+            flag = torch.tensor(
+                [1 if (rt.rank == src and (bi, bj) in self.blocks) else 0],
+                dtype=torch.int32)
+            dist.broadcast(flag, src=src)
+            if flag.item() == 0:
+                continue
+            if dst == -1:
+                t = (self.blocks[(bi, bj)] if rt.rank == src else
+                     torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
+                                 device=self.device))
+                dist.broadcast(t, src=src)
+                out.blocks[(bi, bj)] = t
+            elif src == dst:
+                if rt.rank == src:
+                    out.blocks[(bi, bj)] = self.blocks[(bi, bj)]
+            else:
+                if rt.rank == src:
+                    dist.send(self.blocks[(bi, bj)].contiguous(), dst=dst)
+                elif rt.rank == dst:
+                    t = torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
+                                    device=self.device)
+                    dist.recv(t, src=src)
+                    out.blocks[(bi, bj)] = t
+        return out
+
+    def __repr__(self) -> str:
+        return (f"BlockMatrix({self.nrows}x{self.ncols}, bs={self.block_size},"
+                f" grid={self.grid_rows}x{self.grid_cols},"
+                f" scheme={self.scheme.kind}, dtype={self.dtype},"
+                f" local_blocks={len(self.blocks)})")

@@ -1,0 +1,100 @@
+"""Global configuration for the MatRel-AMD engine.
+
+The reference (purduedb/MatRel, a Spark-SQL fork) configures block size and
+partition counts via SparkConf keys (SURVEY.md §5 "Config / flag system").
+Here a single dataclass covers the whole engine: block geometry, dtype,
+device-grid shape, SUMMA panel size and debug switches. Values can be set
+programmatically, via environment variables (MATREL_*), or per-session.
+
+MI355X-first defaults: blocks are LARGE (multi-GB per GPU shard, not Spark's
+1000x1000) because each GPU has 288 GB HBM3E and kernel-launch amortization
+wants >= 8k-square tiles; the CPU default is small so the multi-block code
+paths are exercised in CPU tests (BASELINE.json config 1: 1k x 1k, 2x2 grid).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import os
+from typing import Optional, Tuple
+
+
+def _env_int(name: str, default: int) -> int:
+    v = os.environ.get(name)
+    return int(v) if v else default
+
+
+def _env_str(name: str, default: str) -> str:
+    return os.environ.get(name, default)
+
+
+@dataclasses.dataclass
+class MatrelConfig:
+    # --- block geometry ---
+    # Side length of a square block. A distributed matrix of shape (m, n) is
+    # tiled into ceil(m/bs) x ceil(n/bs) blocks. GPU default 16384 (bf16 block
+    # = 512 MiB); CPU tests override to something small (e.g. 512).
+    block_size: int = 16384
+
+    # --- device grid ---
+    # Shape (Pr, Pc) of the logical GPU grid used by grid partition schemes
+    # and SUMMA. None -> derived from world size (most-square factoring,
+    # e.g. 8 -> 4x2). Single process -> (1, 1).
+    grid: Optional[Tuple[int, int]] = None
+
+    # --- numerics ---
+    # Compute dtype for dense block math on GPU ("bf16" MFMA path).
+    # Accumulation is always fp32. CPU fallback computes in fp32.
+    dtype: str = "bf16"
+    # dtype used for reductions / solves that need extra precision.
+    solve_dtype: str = "fp32"
+
+    # --- sparsity ---
+    # A block whose density is below this threshold is stored sparse (CSR).
+    sparse_threshold: float = 0.05
+
+    # --- SUMMA ---
+    # Number of k-blocks broadcast per SUMMA step (panel depth). 1 panel =
+    # one block column of A / block row of B. Panels are double-buffered so
+    # the broadcast of panel k+1 overlaps the local GEMM of panel k.
+    summa_panel_depth: int = 1
+
+    # --- execution ---
+    # Use a separate HIP stream for collectives so comm overlaps compute.
+    overlap_comm: bool = True
+    # Emit a chrome-trace of the executed DAG (exec/tracing.py).
+    trace: bool = False
+    trace_path: str = "matrel_trace.json"
+    # Deterministic mode: fixed reduction order (disables some overlap).
+    deterministic: bool = False
+
+    # --- optimizer switches (for tests / ablation) ---
+    enable_transpose_pushdown: bool = True
+    enable_chain_reorder: bool = True
+    enable_cse: bool = True
+    enable_scheme_selection: bool = True
+    enable_scalar_folding: bool = True
+
+    @staticmethod
+    def from_env() -> "MatrelConfig":
+        cfg = MatrelConfig()
+        cfg.block_size = _env_int("MATREL_BLOCK_SIZE", cfg.block_size)
+        cfg.dtype = _env_str("MATREL_DTYPE", cfg.dtype)
+        g = os.environ.get("MATREL_GRID")
+        if g:
+            pr, pc = g.lower().split("x")
+            cfg.grid = (int(pr), int(pc))
+        return cfg
+
+
+# The active config. matrel_amd.init() may replace it.
+_config = MatrelConfig.from_env()
+
+
+def get_config() -> MatrelConfig:
+    return _config
+
+
+def set_config(cfg: MatrelConfig) -> None:
+    global _config
+    _config = cfg
