@@ -1,0 +1,255 @@
+// Python bindings for the matrel_amd CDNA4 kernels (torch extension).
+// Host-side only: tensor checks, workspace allocation, kernel dispatch.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <stdexcept>
+
+// launchers from the .hip translation units
+extern "C" {
+void launch_gemm_bt_bf16_acc(float*, const void*, const void*, int, int,
+                             int, hipStream_t);
+void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
+                            int, hipStream_t);
+void launch_ewise_bf16(int, void*, const void*, const void*, long,
+                       hipStream_t);
+void launch_ewise_f32(int, float*, const float*, const float*, long,
+                      hipStream_t);
+void launch_ewise_scalar_bf16(int, void*, const void*, float, long,
+                              hipStream_t);
+void launch_ewise_scalar_f32(int, float*, const float*, float, long,
+                             hipStream_t);
+void launch_mask_value_bf16(void*, const void*, float, float, int, long,
+                            hipStream_t);
+void launch_mask_value_f32(float*, const float*, float, float, int, long,
+                           hipStream_t);
+void launch_mask_rows_cols_bf16(void*, const void*, const unsigned char*,
+                                const unsigned char*, long, long,
+                                hipStream_t);
+void launch_mask_rows_cols_f32(float*, const float*, const unsigned char*,
+                               const unsigned char*, long, long,
+                               hipStream_t);
+void launch_outer_f32(int, float*, const float*, const float*, long, long,
+                      hipStream_t);
+void launch_reduce_full_bf16(float*, float*, const void*, long, int,
+                             hipStream_t);
+void launch_reduce_full_f32(float*, float*, const float*, long, int,
+                            hipStream_t);
+void launch_reduce_rows_bf16(float*, const void*, long, long, int,
+                             hipStream_t);
+void launch_reduce_rows_f32(float*, const float*, long, long, int,
+                            hipStream_t);
+void launch_reduce_cols_bf16(float*, const void*, long, long, int,
+                             hipStream_t);
+void launch_reduce_cols_f32(float*, const float*, long, long, int,
+                            hipStream_t);
+void launch_trace_bf16(float*, float*, const void*, long, long, hipStream_t);
+void launch_trace_f32(float*, float*, const float*, long, long, hipStream_t);
+void launch_transpose_bf16(void*, const void*, long, long, hipStream_t);
+void launch_transpose_f32(float*, const float*, long, long, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.dim() == 2, name, " must be 2D");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+// ---- GEMM: C(f32) += A @ Bt^T; shapes must be padded to the kernel tile
+// (M,N % 128, K % 64 for bf16 / % 32 for f32). Python side pads.
+void gemm_bt_acc(torch::Tensor C, torch::Tensor A, torch::Tensor Bt) {
+  check_2d(C, "C");
+  check_2d(A, "A");
+  check_2d(Bt, "Bt");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32, "C must be fp32");
+  int M = (int)A.size(0), K = (int)A.size(1), N = (int)Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "K mismatch");
+  TORCH_CHECK(C.size(0) == M && C.size(1) == N, "C shape mismatch");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0, "M,N must be multiples of 128");
+  if (is_bf16(A)) {
+    TORCH_CHECK(is_bf16(Bt), "A/Bt dtype mismatch");
+    TORCH_CHECK(K % 64 == 0, "K must be a multiple of 64 (bf16)");
+    launch_gemm_bt_bf16_acc(C.data_ptr<float>(), A.data_ptr(), Bt.data_ptr(),
+                            M, N, K, cur_stream());
+  } else {
+    TORCH_CHECK(A.scalar_type() == torch::kFloat32, "A must be bf16/fp32");
+    TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32 (fp32)");
+    launch_gemm_bt_f32_acc(C.data_ptr<float>(), A.data_ptr<float>(),
+                           Bt.data_ptr<float>(), M, N, K, cur_stream());
+  }
+}
+
+torch::Tensor ewise(int64_t op, torch::Tensor a, torch::Tensor b) {
+  check_2d(a, "a");
+  check_2d(b, "b");
+  TORCH_CHECK(a.sizes() == b.sizes() && a.scalar_type() == b.scalar_type());
+  auto out = torch::empty_like(a);
+  long n = a.numel();
+  if (is_bf16(a))
+    launch_ewise_bf16((int)op, out.data_ptr(), a.data_ptr(), b.data_ptr(), n,
+                      cur_stream());
+  else
+    launch_ewise_f32((int)op, out.data_ptr<float>(), a.data_ptr<float>(),
+                     b.data_ptr<float>(), n, cur_stream());
+  return out;
+}
+
+torch::Tensor ewise_scalar(int64_t op, torch::Tensor a, double s) {
+  check_2d(a, "a");
+  auto out = torch::empty_like(a);
+  long n = a.numel();
+  if (is_bf16(a))
+    launch_ewise_scalar_bf16((int)op, out.data_ptr(), a.data_ptr(), (float)s,
+                             n, cur_stream());
+  else
+    launch_ewise_scalar_f32((int)op, out.data_ptr<float>(),
+                            a.data_ptr<float>(), (float)s, n, cur_stream());
+  return out;
+}
+
+torch::Tensor mask_value(torch::Tensor a, double lo, double hi, bool inside) {
+  check_2d(a, "a");
+  auto out = torch::empty_like(a);
+  long n = a.numel();
+  if (is_bf16(a))
+    launch_mask_value_bf16(out.data_ptr(), a.data_ptr(), (float)lo,
+                           (float)hi, inside ? 1 : 0, n, cur_stream());
+  else
+    launch_mask_value_f32(out.data_ptr<float>(), a.data_ptr<float>(),
+                          (float)lo, (float)hi, inside ? 1 : 0, n,
+                          cur_stream());
+  return out;
+}
+
+torch::Tensor mask_rows_cols(torch::Tensor a,
+                             c10::optional<torch::Tensor> rm,
+                             c10::optional<torch::Tensor> cm) {
+  check_2d(a, "a");
+  auto out = torch::empty_like(a);
+  const unsigned char* rmp = nullptr;
+  const unsigned char* cmp = nullptr;
+  torch::Tensor rmt, cmt;
+  if (rm.has_value()) {
+    rmt = rm->to(a.device(), torch::kUInt8).contiguous();
+    TORCH_CHECK(rmt.numel() == a.size(0), "row mask length");
+    rmp = rmt.data_ptr<unsigned char>();
+  }
+  if (cm.has_value()) {
+    cmt = cm->to(a.device(), torch::kUInt8).contiguous();
+    TORCH_CHECK(cmt.numel() == a.size(1), "col mask length");
+    cmp = cmt.data_ptr<unsigned char>();
+  }
+  if (is_bf16(a))
+    launch_mask_rows_cols_bf16(out.data_ptr(), a.data_ptr(), rmp, cmp,
+                               a.size(0), a.size(1), cur_stream());
+  else
+    launch_mask_rows_cols_f32(out.data_ptr<float>(), a.data_ptr<float>(),
+                              rmp, cmp, a.size(0), a.size(1), cur_stream());
+  return out;
+}
+
+torch::Tensor outer(int64_t op, torch::Tensor u, torch::Tensor v) {
+  TORCH_CHECK(u.is_cuda() && v.is_cuda());
+  auto uf = u.to(torch::kFloat32).contiguous().reshape(-1);
+  auto vf = v.to(torch::kFloat32).contiguous().reshape(-1);
+  long m = uf.numel(), n = vf.numel();
+  auto out = torch::empty({m, n}, uf.options());
+  launch_outer_f32((int)op, out.data_ptr<float>(), uf.data_ptr<float>(),
+                   vf.data_ptr<float>(), m, n, cur_stream());
+  return out;
+}
+
+torch::Tensor reduce2d(torch::Tensor a, int64_t kind, int64_t axis) {
+  check_2d(a, "a");
+  long rows = a.size(0), cols = a.size(1);
+  auto opts = torch::TensorOptions()
+                  .dtype(torch::kFloat32)
+                  .device(a.device());
+  if (axis < 0) {
+    auto out = torch::empty({1}, opts);
+    auto partials = torch::empty({1024}, opts);
+    if (is_bf16(a))
+      launch_reduce_full_bf16(out.data_ptr<float>(),
+                              partials.data_ptr<float>(), a.data_ptr(),
+                              rows * cols, (int)kind, cur_stream());
+    else
+      launch_reduce_full_f32(out.data_ptr<float>(),
+                             partials.data_ptr<float>(),
+                             a.data_ptr<float>(), rows * cols, (int)kind,
+                             cur_stream());
+    return out;
+  }
+  if (axis == 1) {
+    auto out = torch::empty({rows}, opts);
+    if (is_bf16(a))
+      launch_reduce_rows_bf16(out.data_ptr<float>(), a.data_ptr(), rows,
+                              cols, (int)kind, cur_stream());
+    else
+      launch_reduce_rows_f32(out.data_ptr<float>(), a.data_ptr<float>(),
+                             rows, cols, (int)kind, cur_stream());
+    return out;
+  }
+  auto out = torch::empty({cols}, opts);
+  if (is_bf16(a))
+    launch_reduce_cols_bf16(out.data_ptr<float>(), a.data_ptr(), rows, cols,
+                            (int)kind, cur_stream());
+  else
+    launch_reduce_cols_f32(out.data_ptr<float>(), a.data_ptr<float>(), rows,
+                           cols, (int)kind, cur_stream());
+  return out;
+}
+
+torch::Tensor trace(torch::Tensor a) {
+  check_2d(a, "a");
+  long n = std::min(a.size(0), a.size(1));
+  auto opts =
+      torch::TensorOptions().dtype(torch::kFloat32).device(a.device());
+  auto out = torch::empty({1}, opts);
+  auto partials = torch::empty({1024}, opts);
+  if (is_bf16(a))
+    launch_trace_bf16(out.data_ptr<float>(), partials.data_ptr<float>(),
+                      a.data_ptr(), n, a.size(1), cur_stream());
+  else
+    launch_trace_f32(out.data_ptr<float>(), partials.data_ptr<float>(),
+                     a.data_ptr<float>(), n, a.size(1), cur_stream());
+  return out;
+}
+
+torch::Tensor transpose(torch::Tensor a) {
+  check_2d(a, "a");
+  auto out = torch::empty({a.size(1), a.size(0)}, a.options());
+  if (is_bf16(a))
+    launch_transpose_bf16(out.data_ptr(), a.data_ptr(), a.size(0), a.size(1),
+                          cur_stream());
+  else
+    launch_transpose_f32(out.data_ptr<float>(), a.data_ptr<float>(),
+                         a.size(0), a.size(1), cur_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "matrel_amd hand-written CDNA4 (gfx950) kernels";
+  m.def("gemm_bt_acc", &gemm_bt_acc,
+        "C(f32) += A @ Bt^T (bf16 MFMA / exact-f32 MFMA)");
+  m.def("ewise", &ewise, "elementwise a OP b");
+  m.def("ewise_scalar", &ewise_scalar, "elementwise a OP s");
+  m.def("mask_value", &mask_value, "value-predicate mask");
+  m.def("mask_rows_cols", &mask_rows_cols, "index-predicate mask");
+  m.def("outer", &outer, "outer cross-join");
+  m.def("reduce2d", &reduce2d, "full/row/col reduction");
+  m.def("trace", &trace, "diagonal sum");
+  m.def("transpose", &transpose, "materialized transpose");
+}

@@ -1,0 +1,239 @@
+// Dense MFMA GEMM kernels for gfx950 (MI355X).
+//
+// THE hot kernel of the engine (the reference's per-block
+// Breeze/netlib dgemm becomes a hand-written CDNA4 MFMA GEMM).
+//
+// C(f32, M x N) += A(M x K) @ Bt(N x K)^T   - both operands row-major,
+// B pre-transposed so both A- and B-tiles stage identically (contiguous
+// K-rows; the B^T form is what the transpose-pushdown optimizer naturally
+// produces, and summa.py transposes each broadcast panel once).
+//
+// Structure (cdna_hip_programming.md §5, the "step-3" ladder structure +
+// T1/T2): 128x128 tile, BK=64, 4 waves (2x2), 16x16x32 bf16 MFMA, LDS
+// double-buffered via 16-B global_load_lds with the XOR-16 source/read
+// swizzle (rule 21: linear LDS dest, inverse-swizzled global source,
+// swizzled ds_read), XCD-aware block swizzle.
+//
+// Shapes must satisfy M%128==0, N%128==0, K%64==0 (the Python wrapper pads
+// ragged edge blocks).
+#include "common.h"
+
+#define BM 128
+#define BN 128
+#define BKH 64  // bf16 K-step
+#define BKF 32  // f32  K-step
+
+// ---------------------------------------------------------------- bf16 ----
+// LDS tiles: As[128][64] bf16 (16 KiB), Bs[128][64] bf16, x2 buffers = 64 KiB
+// (2 blocks/CU fit in 160 KiB).
+
+__global__ void
+__launch_bounds__(256)
+gemm_bt_bf16_acc_k(float* __restrict__ C, const bf16* __restrict__ A,
+                   const bf16* __restrict__ Bt, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                       // [2][128*64]
+  bf16* Bs = (bf16*)(smem + 2 * BM * BKH * 2);  // [2][128*64]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wr = wave >> 1, wc = wave & 1;
+
+  const u32 ntm = (u32)(M / BM), ntn = (u32)(N / BN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  const u32 tm = bid / ntn, tn = bid % ntn;
+  const long m0 = (long)tm * BM, n0 = (long)tn * BN;
+
+  // staging geometry: tile row = 64 cols * 2 B = 128 B = 8 16-B slots;
+  // thread t stages slot (t%8) of row (t/8); source slot is XOR-swizzled.
+  const u32 srow = tid >> 3, sslot = tid & 7;
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKH;
+    // 4 chunks of 32 rows each (256 threads x 16 B = 4 KiB per chunk)
+    for (int it = 0; it < 4; ++it) {
+      u32 row = srow + it * 32;
+      const bf16* ga = A + (m0 + row) * (long)K + k0 + src_slot * 8;
+      bf16* la = As + buf * BM * BKH + (row * 8u + sslot) * 8u;
+      glds16(ga, la);
+      const bf16* gb = Bt + (n0 + row) * (long)K + k0 + src_slot * 8;
+      bf16* lb = Bs + buf * BM * BKH + (row * 8u + sslot) * 8u;
+      glds16(gb, lb);
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+  bf16x8 af[2][4], bf[2][4];
+
+  const u32 arow_base = wr * 64 + (lane & 15);
+  const u32 brow_base = wc * 64 + (lane & 15);
+  const u32 kgrp = lane >> 4;  // 0..3: k sub-chunk of 8 within the 32-slice
+
+  auto read_frags = [&](int buf) {
+    const char* a_base = (const char*)(As + buf * BM * BKH);
+    const char* b_base = (const char*)(Bs + buf * BM * BKH);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        u32 row = arow_base + m * 16;
+        u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        af[kk][m] = *(const bf16x8*)(a_base + byte);
+        u32 rowb = brow_base + m * 16;
+        u32 byteb = rowb * 128u + (u32)kk * 64u + kgrp * 16u;
+        byteb ^= (rowb & 7) << 4;
+        bf[kk][m] = *(const bf16x8*)(b_base + byteb);
+      }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);
+    read_frags(cur);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[kk][m], bf[kk][n], acc[m][n], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();  // drains glds (vmcnt 0) + barrier
+    cur ^= 1;
+  }
+
+  // epilogue: C(f32) += acc.  C/D map: col = lane&15, row = (lane>>4)*4+reg.
+  const u32 crow0 = wr * 64 + (lane >> 4) * 4;
+  const u32 ccol0 = wc * 64 + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[m][n][r];
+    }
+}
+
+// ----------------------------------------------------------------- f32 ----
+// Exact-f32 path (v_mfma_f32_16x16x4_f32: f32 in/acc at the 157 TF vector
+// rate - cdna_hip_programming.md §3 "FP32-input MFMA"). Used for fp32
+// matrices (solver outputs, OLS small multiplies) where bf16 rounding is
+// not acceptable. Same tile structure; BK=32.
+
+__global__ void
+__launch_bounds__(256)
+gemm_bt_f32_acc_k(float* __restrict__ C, const float* __restrict__ A,
+                  const float* __restrict__ Bt, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* As = (float*)smem;                       // [2][128*32]
+  float* Bs = (float*)(smem + 2 * BM * BKF * 4);  // [2][128*32]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wr = wave >> 1, wc = wave & 1;
+
+  const u32 ntm = (u32)(M / BM), ntn = (u32)(N / BN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  const u32 tm = bid / ntn, tn = bid % ntn;
+  const long m0 = (long)tm * BM, n0 = (long)tn * BN;
+
+  // tile row = 32 f32 = 128 B = 8 slots of 16 B (4 f32)
+  const u32 srow = tid >> 3, sslot = tid & 7;
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKF;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKF;
+    for (int it = 0; it < 4; ++it) {
+      u32 row = srow + it * 32;
+      const float* ga = A + (m0 + row) * (long)K + k0 + src_slot * 4;
+      float* la = As + buf * BM * BKF + (row * 8u + sslot) * 4u;
+      glds16(ga, la);
+      const float* gb = Bt + (n0 + row) * (long)K + k0 + src_slot * 4;
+      float* lb = Bs + buf * BM * BKF + (row * 8u + sslot) * 4u;
+      glds16(gb, lb);
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+  const u32 arow_base = wr * 64 + (lane & 15);
+  const u32 brow_base = wc * 64 + (lane & 15);
+  const u32 kgrp = lane >> 4;  // k = k4*4 + kgrp
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);
+    const char* a_base = (const char*)(As + cur * BM * BKF);
+    const char* b_base = (const char*)(Bs + cur * BM * BKF);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int k4 = 0; k4 < BKF / 4; ++k4) {
+      float a_s[4], b_s[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        u32 row = arow_base + m * 16;
+        u32 byte = row * 128u + (k4 * 4 + kgrp) * 4u;
+        byte ^= (row & 7) << 4;
+        a_s[m] = *(const float*)(a_base + byte);
+        u32 rowb = brow_base + m * 16;
+        u32 byteb = rowb * 128u + (k4 * 4 + kgrp) * 4u;
+        byteb ^= (rowb & 7) << 4;
+        b_s[m] = *(const float*)(b_base + byteb);
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_s[m], b_s[n], acc[m][n], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const u32 crow0 = wr * 64 + (lane >> 4) * 4;
+  const u32 ccol0 = wc * 64 + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[m][n][r];
+    }
+}
+
+// ------------------------------------------------------------- launchers --
+
+extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
+                                        const void* Bt, int M, int N, int K,
+                                        hipStream_t stream) {
+  dim3 grid((M / BM) * (N / BN));
+  size_t lds = 2u * BM * BKH * 2 * 2;  // 2 bufs x (A+B) x 16 KiB
+  hipLaunchKernelGGL(gemm_bt_bf16_acc_k, grid, dim3(256), lds, stream, C,
+                     (const bf16*)A, (const bf16*)Bt, M, N, K);
+}
+
+extern "C" void launch_gemm_bt_f32_acc(float* C, const float* A,
+                                       const float* Bt, int M, int N, int K,
+                                       hipStream_t stream) {
+  dim3 grid((M / BM) * (N / BN));
+  size_t lds = 2u * BM * BKF * 4 * 2;
+  hipLaunchKernelGGL(gemm_bt_f32_acc_k, grid, dim3(256), lds, stream, C, A,
+                     Bt, M, N, K);
+}

@@ -1,0 +1,204 @@
+// Reduction kernels: full / per-row / per-col {sum, max, min, nnz, sqsum}
+// + trace. fp32 accumulation throughout.
+//
+// Reference equivalent: MatRel AggregateOperator executions (sum / rowSum /
+// colSum / count / max / min / trace, SURVEY.md §2.1 "Physical relational
+// ops"). Full reductions run as grid partials + a one-block combine
+// (deterministic - no float atomics).
+#include "common.h"
+#include <float.h>
+
+#define RK_SUM 0
+#define RK_MAX 1
+#define RK_MIN 2
+#define RK_NNZ 3
+#define RK_SQSUM 4
+
+__device__ __forceinline__ float rk_init(int kind) {
+  switch (kind) {
+    case RK_MAX: return -FLT_MAX;
+    case RK_MIN: return FLT_MAX;
+    default:     return 0.0f;
+  }
+}
+
+__device__ __forceinline__ float rk_map(int kind, float v) {
+  switch (kind) {
+    case RK_NNZ:   return v != 0.0f ? 1.0f : 0.0f;
+    case RK_SQSUM: return v * v;
+    default:       return v;
+  }
+}
+
+__device__ __forceinline__ float rk_comb(int kind, float a, float b) {
+  switch (kind) {
+    case RK_MAX: return fmaxf(a, b);
+    case RK_MIN: return fminf(a, b);
+    default:     return a + b;
+  }
+}
+
+template <typename T>
+__device__ __forceinline__ float load_f(const T* p, long i) {
+  return (float)p[i];
+}
+
+// ---- block-level reduce of a thread-local value -> lane0 of wave 0
+__device__ __forceinline__ float block_reduce(int kind, float v) {
+  __shared__ float warp_part[4];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = rk_comb(kind, v, __shfl_down(v, off, 64));
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) warp_part[wave] = v;
+  __syncthreads();
+  if (wave == 0) {
+    v = lane < 4 ? warp_part[lane] : rk_init(kind);
+#pragma unroll
+    for (int off = 2; off > 0; off >>= 1)
+      v = rk_comb(kind, v, __shfl_down(v, off, 64));
+  }
+  return v;
+}
+
+// ---- full reduction: stage 1 (per-block partials)
+template <typename T, int KIND>
+__global__ void reduce_full_stage1_k(float* __restrict__ partials,
+                                     const T* __restrict__ a, long n) {
+  float acc = rk_init(KIND);
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n; i += stride)
+    acc = rk_comb(KIND, acc, rk_map(KIND, load_f(a, i)));
+  float r = block_reduce(KIND, acc);
+  if (threadIdx.x == 0) partials[blockIdx.x] = r;
+}
+
+// ---- stage 2: one block combines partials
+__global__ void reduce_full_stage2_k(float* __restrict__ out,
+                                     const float* __restrict__ partials,
+                                     int nparts, int kind) {
+  float acc = rk_init(kind);
+  for (int i = threadIdx.x; i < nparts; i += blockDim.x)
+    acc = rk_comb(kind, acc, partials[i]);
+  float r = block_reduce(kind, acc);
+  if (threadIdx.x == 0) out[0] = r;
+}
+
+// ---- per-row reduction: one block per row (out[m])
+template <typename T, int KIND>
+__global__ void reduce_rows_k(float* __restrict__ out,
+                              const T* __restrict__ a, long rows,
+                              long cols) {
+  long r = blockIdx.x;
+  if (r >= rows) return;
+  const T* row = a + r * cols;
+  float acc = rk_init(KIND);
+  for (long c = threadIdx.x; c < cols; c += blockDim.x)
+    acc = rk_comb(KIND, acc, rk_map(KIND, load_f(row, c)));
+  float v = block_reduce(KIND, acc);
+  if (threadIdx.x == 0) out[r] = v;
+}
+
+// ---- per-col reduction: thread-per-column (coalesced), out[n]
+template <typename T, int KIND>
+__global__ void reduce_cols_k(float* __restrict__ out,
+                              const T* __restrict__ a, long rows,
+                              long cols) {
+  long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  float acc = rk_init(KIND);
+  for (long r = 0; r < rows; ++r)
+    acc = rk_comb(KIND, acc, rk_map(KIND, load_f(a, r * cols + c)));
+  out[c] = acc;
+}
+
+// ---- trace: sum of diagonal
+template <typename T>
+__global__ void trace_stage1_k(float* __restrict__ partials,
+                               const T* __restrict__ a, long n, long cols) {
+  float acc = 0.0f;
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n; i += stride) acc += load_f(a, i * cols + i);
+  float r = block_reduce(RK_SUM, acc);
+  if (threadIdx.x == 0) partials[blockIdx.x] = r;
+}
+
+// ------------------------------------------------------------- launchers --
+
+static inline int rgrid(long n) {
+  long b = (n + 255) / 256;
+  return (int)(b < 1 ? 1 : (b > 1024 ? 1024 : b));
+}
+
+#define K_DISPATCH(T, fn, ...)                                   \
+  switch (kind) {                                                \
+    case 0: hipLaunchKernelGGL((fn<T, 0>), __VA_ARGS__); break;  \
+    case 1: hipLaunchKernelGGL((fn<T, 1>), __VA_ARGS__); break;  \
+    case 2: hipLaunchKernelGGL((fn<T, 2>), __VA_ARGS__); break;  \
+    case 3: hipLaunchKernelGGL((fn<T, 3>), __VA_ARGS__); break;  \
+    default: hipLaunchKernelGGL((fn<T, 4>), __VA_ARGS__); break; \
+  }
+
+extern "C" void launch_reduce_full_bf16(float* out, float* partials,
+                                        const void* a, long n, int kind,
+                                        hipStream_t s) {
+  int g = rgrid(n);
+  K_DISPATCH(bf16, reduce_full_stage1_k, dim3(g), dim3(256), 0, s, partials,
+             (const bf16*)a, n);
+  hipLaunchKernelGGL(reduce_full_stage2_k, dim3(1), dim3(256), 0, s, out,
+                     partials, g, kind);
+}
+
+extern "C" void launch_reduce_full_f32(float* out, float* partials,
+                                       const float* a, long n, int kind,
+                                       hipStream_t s) {
+  int g = rgrid(n);
+  K_DISPATCH(float, reduce_full_stage1_k, dim3(g), dim3(256), 0, s, partials,
+             a, n);
+  hipLaunchKernelGGL(reduce_full_stage2_k, dim3(1), dim3(256), 0, s, out,
+                     partials, g, kind);
+}
+
+extern "C" void launch_reduce_rows_bf16(float* out, const void* a, long rows,
+                                        long cols, int kind, hipStream_t s) {
+  K_DISPATCH(bf16, reduce_rows_k, dim3((unsigned)rows), dim3(256), 0, s, out,
+             (const bf16*)a, rows, cols);
+}
+
+extern "C" void launch_reduce_rows_f32(float* out, const float* a, long rows,
+                                       long cols, int kind, hipStream_t s) {
+  K_DISPATCH(float, reduce_rows_k, dim3((unsigned)rows), dim3(256), 0, s,
+             out, a, rows, cols);
+}
+
+extern "C" void launch_reduce_cols_bf16(float* out, const void* a, long rows,
+                                        long cols, int kind, hipStream_t s) {
+  K_DISPATCH(bf16, reduce_cols_k, dim3((unsigned)((cols + 255) / 256)),
+             dim3(256), 0, s, out, (const bf16*)a, rows, cols);
+}
+
+extern "C" void launch_reduce_cols_f32(float* out, const float* a, long rows,
+                                       long cols, int kind, hipStream_t s) {
+  K_DISPATCH(float, reduce_cols_k, dim3((unsigned)((cols + 255) / 256)),
+             dim3(256), 0, s, out, a, rows, cols);
+}
+
+extern "C" void launch_trace_bf16(float* out, float* partials, const void* a,
+                                  long n, long cols, hipStream_t s) {
+  int g = rgrid(n);
+  hipLaunchKernelGGL(trace_stage1_k<bf16>, dim3(g), dim3(256), 0, s,
+                     partials, (const bf16*)a, n, cols);
+  hipLaunchKernelGGL(reduce_full_stage2_k, dim3(1), dim3(256), 0, s, out,
+                     partials, g, RK_SUM);
+}
+
+extern "C" void launch_trace_f32(float* out, float* partials, const float* a,
+                                 long n, long cols, hipStream_t s) {
+  int g = rgrid(n);
+  hipLaunchKernelGGL(trace_stage1_k<float>, dim3(g), dim3(256), 0, s,
+                     partials, a, n, cols);
+  hipLaunchKernelGGL(reduce_full_stage2_k, dim3(1), dim3(256), 0, s, out,
+                     partials, g, RK_SUM);
+}

@@ -275,15 +275,13 @@ class Executor:
                 if kind == "trace":
                     if bi != bj:
                         continue
-                    d = torch.diagonal(t.float()).sum()
-                    partial += d
+                    partial += blockops.trace_block(t)
                     continue
-                tin = t
                 if kind == "norm_fro":
-                    v = (t.float() ** 2).sum()
-                    partial += v
+                    partial += blockops.reduce2d(t, blockops.RED_SQSUM,
+                                                 None)
                     continue
-                v = blockops.reduce2d(tin, red_kind, None)
+                v = blockops.reduce2d(t, red_kind, None)
                 if kind in ("sum", "avg", "count", "nnz"):
                     partial += v
                 else:  # max/min over blocks
@@ -369,10 +367,7 @@ class Executor:
             h, w = out.block_shape(bi, bj)
             asl = av[bi * bs: bi * bs + h]
             bsl = bv[bj * bs: bj * bs + w]
-            if n.op == "mul":
-                t = torch.outer(asl, bsl)
-            else:
-                t = asl.unsqueeze(1) + bsl.unsqueeze(0)
+            t = blockops.outer(n.op, asl, bsl)
             out.blocks[(bi, bj)] = t.to(out.dtype)
         return out
 

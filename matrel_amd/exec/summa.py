@@ -45,23 +45,42 @@ def _presence_map(m: BlockMatrix) -> torch.Tensor:
     return pm
 
 
+def _pre_t(b: torch.Tensor) -> torch.Tensor:
+    """B operand for the B^T GEMM kernel: materialized transpose on GPU
+    (the bandwidth-bound HIP transpose, amortized across every A-row block
+    that reuses it), a free .t() view on CPU."""
+    return blockops.transpose(b) if b.is_cuda else b.t()
+
+
 def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     """Pure-local block multiply: every needed A/B block is already on the
     rank that owns the corresponding C block (single process, or
-    partition-aware alignment: A row-partitioned + B replicated, etc.)."""
-    for (bi, bj) in out.owned_indices():
-        acc: Optional[torch.Tensor] = None
+    partition-aware alignment: A row-partitioned + B replicated, etc.).
+    Loops (j, k) outer so each B block is transposed once for the B^T MFMA
+    kernel and reused across the block-row loop."""
+    my_is = sorted({bi for (bi, bj) in out.owned_indices()})
+    my_js = sorted({bj for (bi, bj) in out.owned_indices()})
+    acc: dict = {}
+    for bj in my_js:
         for k in range(a.grid_cols):
-            ab = a.get_block(bi, k)
             bb = b.get_block(k, bj)
-            if ab is None or bb is None:
+            if bb is None:
                 continue
-            if acc is None:
-                acc = torch.zeros(out.block_shape(bi, bj),
-                                  dtype=torch.float32, device=out.device)
-            blockops.gemm_acc(acc, ab, bb)
-        if acc is not None:
-            out.blocks[(bi, bj)] = acc.to(out.dtype)
+            bt = _pre_t(bb)
+            for bi in my_is:
+                if not out.is_local(bi, bj):
+                    continue
+                ab = a.get_block(bi, k)
+                if ab is None:
+                    continue
+                c = acc.get((bi, bj))
+                if c is None:
+                    c = torch.zeros(out.block_shape(bi, bj),
+                                    dtype=torch.float32, device=out.device)
+                    acc[(bi, bj)] = c
+                blockops.gemm_bt_acc(c, ab, bt)
+    for (bi, bj), c in acc.items():
+        out.blocks[(bi, bj)] = c.to(out.dtype)
 
 
 def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
@@ -136,18 +155,19 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         a_stack, b_stack, wa, wb = cur
         wa.wait()
         wb.wait()
-        for n_i, i in enumerate(my_is):
-            if not pa[i, k]:
+        for n_j, j in enumerate(my_js):
+            if not pb[k, j]:
                 continue
-            for n_j, j in enumerate(my_js):
-                if not pb[k, j]:
+            bt = _pre_t(b_stack[n_j])
+            for n_i, i in enumerate(my_is):
+                if not pa[i, k]:
                     continue
                 c = acc.get((i, j))
                 if c is None:
                     c = torch.zeros(bs, bs, dtype=torch.float32,
                                     device=out.device)
                     acc[(i, j)] = c
-                blockops.gemm_acc(c, a_stack[n_i], b_stack[n_j])
+                blockops.gemm_bt_acc(c, a_stack[n_i], bt)
         cur = nxt
 
     for (i, j), c in acc.items():

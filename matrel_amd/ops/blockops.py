@@ -7,9 +7,14 @@ GPU-side op dispatches to the hand-written CDNA4 HIP extension
 no-GPU config-1 slice.
 
 Conventions:
-  * Dense blocks are 2D contiguous row-major torch tensors.
-  * GEMM accumulates in fp32 (`acc` tensors are fp32); inputs bf16 on GPU.
-  * Elementwise ops are fused single-pass kernels on GPU (HBM-bound).
+  * Dense blocks are 2D contiguous row-major torch tensors (bf16 or fp32
+    on GPU; fp32 on CPU).
+  * GEMM accumulates in fp32. The HIP GEMM takes B TRANSPOSED
+    (C += A @ Bt^T) so both operands stage identically into LDS; callers
+    that reuse a B block across many A blocks pre-transpose once via
+    `transpose()` and call `gemm_bt_acc` (summa.py, local_matmul).
+  * Kernel tiles need M,N % 128 == 0 and K % 64 (bf16) / % 32 (f32):
+    `_pad2` pads ragged blocks transparently.
 """
 
 from __future__ import annotations
@@ -24,19 +29,61 @@ from .backend import require_ext, use_hip
 EW_ADD, EW_SUB, EW_MUL, EW_DIV = 0, 1, 2, 3
 # scalar op ids
 SC_ADD, SC_MUL, SC_POW = 0, 1, 2
-# reduction kinds
-RED_SUM, RED_MAX, RED_MIN, RED_NNZ = 0, 1, 2, 3
+# reduction kinds shared with csrc/reduce.hip
+RED_SUM, RED_MAX, RED_MIN, RED_NNZ, RED_SQSUM = 0, 1, 2, 3, 4
 
 _EW_TORCH = {
     EW_ADD: torch.add, EW_SUB: torch.sub,
     EW_MUL: torch.mul, EW_DIV: torch.div,
 }
 
+_TILE_MN = 128
+
+
+def _kmult(dtype: torch.dtype) -> int:
+    return 64 if dtype == torch.bfloat16 else 32
+
+
+def _ceil(x: int, m: int) -> int:
+    return (x + m - 1) // m * m
+
+
+def _pad2(t: torch.Tensor, rows: int, cols: int) -> torch.Tensor:
+    if t.shape[0] == rows and t.shape[1] == cols:
+        return t.contiguous()
+    out = torch.zeros(rows, cols, dtype=t.dtype, device=t.device)
+    out[: t.shape[0], : t.shape[1]] = t
+    return out
+
+
+def gemm_bt_acc(acc: torch.Tensor, a: torch.Tensor,
+                bt: torch.Tensor) -> None:
+    """acc(fp32, MxN) += a(MxK) @ bt(NxK)^T - the hot MFMA kernel."""
+    if not use_hip(a):
+        acc.add_(a.float() @ bt.float().t())
+        return
+    ext = require_ext()
+    if a.dtype != bt.dtype:
+        a, bt = a.float(), bt.float()
+    m, k = a.shape
+    n = bt.shape[0]
+    km = _kmult(a.dtype)
+    mp, np_, kp = _ceil(m, _TILE_MN), _ceil(n, _TILE_MN), _ceil(k, km)
+    if (mp, np_, kp) == (m, n, k):
+        ext.gemm_bt_acc(acc, a.contiguous(), bt.contiguous())
+        return
+    ap = _pad2(a, mp, kp)
+    btp = _pad2(bt, np_, kp)
+    cp = torch.zeros(mp, np_, dtype=torch.float32, device=a.device)
+    ext.gemm_bt_acc(cp, ap, btp)
+    acc.add_(cp[:m, :n])
+
 
 def gemm_acc(acc: torch.Tensor, a: torch.Tensor, b: torch.Tensor) -> None:
-    """acc(fp32) += a @ b. The hot kernel (MFMA bf16 on GPU)."""
+    """acc(fp32) += a @ b (transposes b internally on GPU; prefer
+    gemm_bt_acc + a cached transpose when b is reused)."""
     if use_hip(a):
-        require_ext().gemm_bf16_acc(acc, a.contiguous(), b.contiguous())
+        gemm_bt_acc(acc, a, transpose(b))
     else:
         acc.add_(a.float() @ b.float())
 
@@ -51,6 +98,8 @@ def gemm(a: torch.Tensor, b: torch.Tensor,
 
 def ewise(op: int, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     if use_hip(a):
+        if a.dtype != b.dtype:
+            a, b = a.float(), b.float()
         return require_ext().ewise(op, a.contiguous(), b.contiguous())
     return _EW_TORCH[op](a, b)
 
@@ -68,7 +117,7 @@ def ewise_scalar(op: int, a: torch.Tensor, s: float) -> torch.Tensor:
 
 
 def transpose(a: torch.Tensor) -> torch.Tensor:
-    """Materialized transpose (fused tiled kernel on GPU)."""
+    """Materialized transpose (tiled LDS kernel on GPU)."""
     if use_hip(a):
         return require_ext().transpose(a.contiguous())
     return a.t().contiguous()
@@ -76,8 +125,8 @@ def transpose(a: torch.Tensor) -> torch.Tensor:
 
 def reduce2d(a: torch.Tensor, kind: int, axis: Optional[int]
              ) -> torch.Tensor:
-    """Reduce a block. axis=None -> scalar (0-dim is returned as 1-elem),
-    axis=0 -> per-column vector, axis=1 -> per-row vector. fp32 output."""
+    """Reduce a block. axis=None -> 1-elem tensor, axis=0 -> per-column,
+    axis=1 -> per-row. fp32 output."""
     if use_hip(a):
         return require_ext().reduce2d(a.contiguous(), kind,
                                       -1 if axis is None else axis)
@@ -91,9 +140,30 @@ def reduce2d(a: torch.Tensor, kind: int, axis: Optional[int]
     elif kind == RED_NNZ:
         nz = (af != 0).float()
         r = nz.sum() if axis is None else nz.sum(dim=axis)
+    elif kind == RED_SQSUM:
+        sq = af * af
+        r = sq.sum() if axis is None else sq.sum(dim=axis)
     else:
         raise ValueError(kind)
     return r.reshape(-1) if axis is None else r
+
+
+def trace_block(a: torch.Tensor) -> torch.Tensor:
+    """Sum of the main diagonal (1-elem fp32 tensor)."""
+    if use_hip(a):
+        return require_ext().trace(a.contiguous())
+    return torch.diagonal(a.float()).sum().reshape(1)
+
+
+def outer(op: str, u: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """Cross-join outer op: out[i,j] = u[i] * v[j] (op='mul') or + ."""
+    if use_hip(u):
+        return require_ext().outer(0 if op == "mul" else 1,
+                                   u.contiguous(), v.contiguous())
+    uf, vf = u.float().reshape(-1), v.float().reshape(-1)
+    if op == "mul":
+        return torch.outer(uf, vf)
+    return uf.unsqueeze(1) + vf.unsqueeze(0)
 
 
 def mask_rows_cols(a: torch.Tensor, row_mask: Optional[torch.Tensor],
