@@ -40,10 +40,11 @@ void launch_reduce_rows_bf16(float*, const void*, long, long, int,
                              hipStream_t);
 void launch_reduce_rows_f32(float*, const float*, long, long, int,
                             hipStream_t);
-void launch_reduce_cols_bf16(float*, const void*, long, long, int,
+void launch_reduce_cols_bf16(float*, float*, const void*, long, long, int,
                              hipStream_t);
-void launch_reduce_cols_f32(float*, const float*, long, long, int,
+void launch_reduce_cols_f32(float*, float*, const float*, long, long, int,
                             hipStream_t);
+int reduce_cols_nchunks(long, long);
 void launch_trace_bf16(float*, float*, const void*, long, long, hipStream_t);
 void launch_trace_f32(float*, float*, const float*, long, long, hipStream_t);
 void launch_transpose_bf16(void*, const void*, long, long, hipStream_t);
@@ -201,12 +202,16 @@ torch::Tensor reduce2d(torch::Tensor a, int64_t kind, int64_t axis) {
     return out;
   }
   auto out = torch::empty({cols}, opts);
+  int nc = reduce_cols_nchunks(rows, cols);
+  auto partials = torch::empty({(long)nc * cols}, opts);
   if (is_bf16(a))
-    launch_reduce_cols_bf16(out.data_ptr<float>(), a.data_ptr(), rows, cols,
-                            (int)kind, cur_stream());
+    launch_reduce_cols_bf16(out.data_ptr<float>(),
+                            partials.data_ptr<float>(), a.data_ptr(), rows,
+                            cols, (int)kind, cur_stream());
   else
-    launch_reduce_cols_f32(out.data_ptr<float>(), a.data_ptr<float>(), rows,
-                           cols, (int)kind, cur_stream());
+    launch_reduce_cols_f32(out.data_ptr<float>(),
+                           partials.data_ptr<float>(), a.data_ptr<float>(),
+                           rows, cols, (int)kind, cur_stream());
   return out;
 }
 

@@ -218,11 +218,121 @@ gemm_bt_f32_acc_k(float* __restrict__ C, const float* __restrict__ A,
     }
 }
 
+// --------------------------------------------------- bf16, 256^2 tile ----
+// 8 waves (2M x 4N), per-wave 128x64 output (8x4 16x16 frags), BK=64,
+// 2 LDS buffers x (A 256x64 + B 256x64) = 128 KiB -> 1 block/CU.
+// The cdna_hip_programming.md §5 table row: "glds, 2 LDS buffers, BK=64,
+// vmcnt(0) + plain __syncthreads()" - the top BK=64 tier for this tile.
+
+#define BM2 256
+#define BN2 256
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
+                       const bf16* __restrict__ Bt, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                        // [2][256*64]
+  bf16* Bs = (bf16*)(smem + 2 * BM2 * BKH * 2);  // [2][256*64]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;  // 2 x 4 wave grid
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  const u32 tm = bid / ntn, tn = bid % ntn;
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  // staging: tile = 256 rows x 64 cols x 2 B; row = 8 x 16-B slots.
+  // 512 threads x 16 B = 8 KiB per pass; 4 passes per operand tile.
+  const u32 srow = tid >> 3, sslot = tid & 7;  // rows 0..63 per pass
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKH;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      u32 row = srow + it * 64;
+      const bf16* ga = A + (m0 + row) * (long)K + k0 + src_slot * 8;
+      glds16(ga, As + buf * BM2 * BKH + (row * 8u + sslot) * 8u);
+      const bf16* gb = Bt + (n0 + row) * (long)K + k0 + src_slot * 8;
+      glds16(gb, Bs + buf * BM2 * BKH + (row * 8u + sslot) * 8u);
+    }
+  };
+
+  f32x4 acc[8][4] = {};
+  const u32 arow_base = wm * 128 + (lane & 15);
+  const u32 brow_base = wn * 64 + (lane & 15);
+  const u32 kgrp = lane >> 4;
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);
+    const char* a_base = (const char*)(As + cur * BM2 * BKH);
+    const char* b_base = (const char*)(Bs + cur * BM2 * BKH);
+    bf16x8 af[8], bf[4];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+        u32 row = arow_base + m * 16;
+        u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        af[m] = *(const bf16x8*)(a_base + byte);
+      }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        u32 rowb = brow_base + n * 16;
+        u32 byteb = rowb * 128u + (u32)kk * 64u + kgrp * 16u;
+        byteb ^= (rowb & 7) << 4;
+        bf[n] = *(const bf16x8*)(b_base + byteb);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[m], bf[n], acc[m][n], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const u32 crow0 = wm * 128 + (lane >> 4) * 4;
+  const u32 ccol0 = wn * 64 + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < 8; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[m][n][r];
+    }
+}
+
 // ------------------------------------------------------------- launchers --
 
 extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
                                         const void* Bt, int M, int N, int K,
                                         hipStream_t stream) {
+  // 256^2 tile only when the grid still fills the chip (>= ~192 of 256
+  // CUs at 1 block/CU); smaller problems keep the 128^2 tile (2 blocks/CU,
+  // 4x the block count).
+  if (M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 192) {
+    dim3 grid((M / BM2) * (N / BN2));
+    size_t lds = 2u * BM2 * BKH * 2 * 2;  // 128 KiB
+    hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds, stream,
+                       C, (const bf16*)A, (const bf16*)Bt, M, N, K);
+    return;
+  }
   dim3 grid((M / BM) * (N / BN));
   size_t lds = 2u * BM * BKH * 2 * 2;  // 2 bufs x (A+B) x 16 KiB
   hipLaunchKernelGGL(gemm_bt_bf16_acc_k, grid, dim3(256), lds, stream, C,

@@ -61,15 +61,25 @@ __device__ __forceinline__ float block_reduce(int kind, float v) {
   return v;
 }
 
-// ---- full reduction: stage 1 (per-block partials)
-template <typename T, int KIND>
+// ---- full reduction: stage 1 (per-block partials), 16-B vector loads
+template <typename T, int KIND, int V>
 __global__ void reduce_full_stage1_k(float* __restrict__ partials,
                                      const T* __restrict__ a, long n) {
+  typedef T vecT __attribute__((ext_vector_type(V)));
   float acc = rk_init(KIND);
+  long nv = n / V;
   long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
-  for (long i = i0; i < n; i += stride)
+  for (long i = i0; i < nv; i += stride) {
+    vecT v = *(const vecT*)(a + i * V);
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      acc = rk_comb(KIND, acc, rk_map(KIND, (float)v[j]));
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (n - nv * V)) {
+    long i = nv * V + threadIdx.x;
     acc = rk_comb(KIND, acc, rk_map(KIND, load_f(a, i)));
+  }
   float r = block_reduce(KIND, acc);
   if (threadIdx.x == 0) partials[blockIdx.x] = r;
 }
@@ -85,31 +95,72 @@ __global__ void reduce_full_stage2_k(float* __restrict__ out,
   if (threadIdx.x == 0) out[0] = r;
 }
 
-// ---- per-row reduction: one block per row (out[m])
-template <typename T, int KIND>
+// ---- per-row reduction: one block per row (out[m]), 16-B vector loads
+template <typename T, int KIND, int V>
 __global__ void reduce_rows_k(float* __restrict__ out,
                               const T* __restrict__ a, long rows,
                               long cols) {
+  typedef T vecT __attribute__((ext_vector_type(V)));
   long r = blockIdx.x;
   if (r >= rows) return;
   const T* row = a + r * cols;
   float acc = rk_init(KIND);
-  for (long c = threadIdx.x; c < cols; c += blockDim.x)
-    acc = rk_comb(KIND, acc, rk_map(KIND, load_f(row, c)));
+  long cv = cols / V;
+  for (long c = threadIdx.x; c < cv; c += blockDim.x) {
+    vecT v = *(const vecT*)(row + c * V);
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      acc = rk_comb(KIND, acc, rk_map(KIND, (float)v[j]));
+  }
+  if (threadIdx.x < (cols - cv * V))
+    acc = rk_comb(KIND, acc, rk_map(KIND, load_f(row, cv * V + threadIdx.x)));
   float v = block_reduce(KIND, acc);
   if (threadIdx.x == 0) out[r] = v;
 }
 
-// ---- per-col reduction: thread-per-column (coalesced), out[n]
-template <typename T, int KIND>
-__global__ void reduce_cols_k(float* __restrict__ out,
-                              const T* __restrict__ a, long rows,
-                              long cols) {
+// ---- per-col reduction, two-stage: grid (col-tiles x row-chunks), each
+// block reduces its row chunk for 2048 columns (8 cols/thread, 16-B
+// vector loads) into partials[chunk][cols]; stage 2 combines chunks.
+#define CR_VEC 8  // bf16 elems per load; f32 uses 4
+
+template <typename T, int KIND, int V>
+__global__ void reduce_cols_stage1_k(float* __restrict__ partials,
+                                     const T* __restrict__ a, long rows,
+                                     long cols, int nchunks) {
+  typedef T vecT __attribute__((ext_vector_type(V)));
+  long c0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * V;
+  int chunk = blockIdx.y;
+  long r0 = rows * chunk / nchunks, r1 = rows * (chunk + 1) / nchunks;
+  float acc[V];
+#pragma unroll
+  for (int j = 0; j < V; ++j) acc[j] = rk_init(KIND);
+  if (c0 + V <= cols) {
+    for (long r = r0; r < r1; ++r) {
+      vecT v = *(const vecT*)(a + r * cols + c0);
+#pragma unroll
+      for (int j = 0; j < V; ++j)
+        acc[j] = rk_comb(KIND, acc[j], rk_map(KIND, (float)v[j]));
+    }
+  } else if (c0 < cols) {
+    for (long r = r0; r < r1; ++r)
+      for (int j = 0; c0 + j < cols; ++j)
+        acc[j] = rk_comb(KIND, acc[j],
+                         rk_map(KIND, load_f(a, r * cols + c0 + j)));
+  } else {
+    return;
+  }
+  float* p = partials + (long)chunk * cols + c0;
+  for (int j = 0; j < V && c0 + j < cols; ++j) p[j] = acc[j];
+}
+
+__global__ void reduce_cols_stage2_k(float* __restrict__ out,
+                                     const float* __restrict__ partials,
+                                     long cols, int nchunks, int kind) {
   long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= cols) return;
-  float acc = rk_init(KIND);
-  for (long r = 0; r < rows; ++r)
-    acc = rk_comb(KIND, acc, rk_map(KIND, load_f(a, r * cols + c)));
+  float acc = rk_init(kind);
+  for (int ch = 0; ch < nchunks; ++ch)
+    acc = rk_comb(kind, acc, partials[(long)ch * cols + c]);
   out[c] = acc;
 }
 
@@ -132,20 +183,20 @@ static inline int rgrid(long n) {
   return (int)(b < 1 ? 1 : (b > 1024 ? 1024 : b));
 }
 
-#define K_DISPATCH(T, fn, ...)                                   \
-  switch (kind) {                                                \
-    case 0: hipLaunchKernelGGL((fn<T, 0>), __VA_ARGS__); break;  \
-    case 1: hipLaunchKernelGGL((fn<T, 1>), __VA_ARGS__); break;  \
-    case 2: hipLaunchKernelGGL((fn<T, 2>), __VA_ARGS__); break;  \
-    case 3: hipLaunchKernelGGL((fn<T, 3>), __VA_ARGS__); break;  \
-    default: hipLaunchKernelGGL((fn<T, 4>), __VA_ARGS__); break; \
+#define K_DISPATCH(T, V, fn, ...)                                   \
+  switch (kind) {                                                   \
+    case 0: hipLaunchKernelGGL((fn<T, 0, V>), __VA_ARGS__); break;  \
+    case 1: hipLaunchKernelGGL((fn<T, 1, V>), __VA_ARGS__); break;  \
+    case 2: hipLaunchKernelGGL((fn<T, 2, V>), __VA_ARGS__); break;  \
+    case 3: hipLaunchKernelGGL((fn<T, 3, V>), __VA_ARGS__); break;  \
+    default: hipLaunchKernelGGL((fn<T, 4, V>), __VA_ARGS__); break; \
   }
 
 extern "C" void launch_reduce_full_bf16(float* out, float* partials,
                                         const void* a, long n, int kind,
                                         hipStream_t s) {
   int g = rgrid(n);
-  K_DISPATCH(bf16, reduce_full_stage1_k, dim3(g), dim3(256), 0, s, partials,
+  K_DISPATCH(bf16, 8, reduce_full_stage1_k, dim3(g), dim3(256), 0, s, partials,
              (const bf16*)a, n);
   hipLaunchKernelGGL(reduce_full_stage2_k, dim3(1), dim3(256), 0, s, out,
                      partials, g, kind);
@@ -155,7 +206,7 @@ extern "C" void launch_reduce_full_f32(float* out, float* partials,
                                        const float* a, long n, int kind,
                                        hipStream_t s) {
   int g = rgrid(n);
-  K_DISPATCH(float, reduce_full_stage1_k, dim3(g), dim3(256), 0, s, partials,
+  K_DISPATCH(float, 4, reduce_full_stage1_k, dim3(g), dim3(256), 0, s, partials,
              a, n);
   hipLaunchKernelGGL(reduce_full_stage2_k, dim3(1), dim3(256), 0, s, out,
                      partials, g, kind);
@@ -163,26 +214,60 @@ extern "C" void launch_reduce_full_f32(float* out, float* partials,
 
 extern "C" void launch_reduce_rows_bf16(float* out, const void* a, long rows,
                                         long cols, int kind, hipStream_t s) {
-  K_DISPATCH(bf16, reduce_rows_k, dim3((unsigned)rows), dim3(256), 0, s, out,
+  K_DISPATCH(bf16, 8, reduce_rows_k, dim3((unsigned)rows), dim3(256), 0, s, out,
              (const bf16*)a, rows, cols);
 }
 
 extern "C" void launch_reduce_rows_f32(float* out, const float* a, long rows,
                                        long cols, int kind, hipStream_t s) {
-  K_DISPATCH(float, reduce_rows_k, dim3((unsigned)rows), dim3(256), 0, s,
+  K_DISPATCH(float, 4, reduce_rows_k, dim3((unsigned)rows), dim3(256), 0, s,
              out, a, rows, cols);
 }
 
-extern "C" void launch_reduce_cols_bf16(float* out, const void* a, long rows,
-                                        long cols, int kind, hipStream_t s) {
-  K_DISPATCH(bf16, reduce_cols_k, dim3((unsigned)((cols + 255) / 256)),
-             dim3(256), 0, s, out, (const bf16*)a, rows, cols);
+// partials workspace supplied by the binding: [nchunks][cols] f32.
+static inline int col_chunks(long rows, long cols) {
+  // enough blocks to fill 256 CUs: col_tiles * chunks >= ~1024
+  long col_tiles = (cols + 2047) / 2048;
+  long want = (2048 + col_tiles - 1) / col_tiles;
+  long mx = rows < 64 ? rows : 64;
+  long c = want < 1 ? 1 : (want > mx ? mx : want);
+  return (int)(c < 1 ? 1 : c);
 }
 
-extern "C" void launch_reduce_cols_f32(float* out, const float* a, long rows,
-                                       long cols, int kind, hipStream_t s) {
-  K_DISPATCH(float, reduce_cols_k, dim3((unsigned)((cols + 255) / 256)),
-             dim3(256), 0, s, out, a, rows, cols);
+#define KC_DISPATCH(T, V, ...)                                              \
+  switch (kind) {                                                           \
+    case 0: hipLaunchKernelGGL((reduce_cols_stage1_k<T, 0, V>), __VA_ARGS__); break; \
+    case 1: hipLaunchKernelGGL((reduce_cols_stage1_k<T, 1, V>), __VA_ARGS__); break; \
+    case 2: hipLaunchKernelGGL((reduce_cols_stage1_k<T, 2, V>), __VA_ARGS__); break; \
+    case 3: hipLaunchKernelGGL((reduce_cols_stage1_k<T, 3, V>), __VA_ARGS__); break; \
+    default: hipLaunchKernelGGL((reduce_cols_stage1_k<T, 4, V>), __VA_ARGS__); break; \
+  }
+
+extern "C" int reduce_cols_nchunks(long rows, long cols) {
+  return col_chunks(rows, cols);
+}
+
+extern "C" void launch_reduce_cols_bf16(float* out, float* partials,
+                                        const void* a, long rows, long cols,
+                                        int kind, hipStream_t s) {
+  int nc = col_chunks(rows, cols);
+  dim3 g1((unsigned)((cols + 2047) / 2048), (unsigned)nc);
+  KC_DISPATCH(bf16, 8, g1, dim3(256), 0, s, partials, (const bf16*)a, rows,
+              cols, nc);
+  hipLaunchKernelGGL(reduce_cols_stage2_k,
+                     dim3((unsigned)((cols + 255) / 256)), dim3(256), 0, s,
+                     out, partials, cols, nc, kind);
+}
+
+extern "C" void launch_reduce_cols_f32(float* out, float* partials,
+                                       const float* a, long rows, long cols,
+                                       int kind, hipStream_t s) {
+  int nc = col_chunks(rows, cols);
+  dim3 g1((unsigned)((cols + 1023) / 1024), (unsigned)nc);
+  KC_DISPATCH(float, 4, g1, dim3(256), 0, s, partials, a, rows, cols, nc);
+  hipLaunchKernelGGL(reduce_cols_stage2_k,
+                     dim3((unsigned)((cols + 255) / 256)), dim3(256), 0, s,
+                     out, partials, cols, nc, kind);
 }
 
 extern "C" void launch_trace_bf16(float* out, float* partials, const void* a,
