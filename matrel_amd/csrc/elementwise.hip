@@ -11,6 +11,8 @@
 #define EW_SUB 1
 #define EW_MUL 2
 #define EW_DIV 3
+#define EW_MIN 4
+#define EW_MAX 5
 #define SC_ADD 0
 #define SC_MUL 1
 #define SC_POW 2
@@ -23,7 +25,9 @@ __device__ __forceinline__ float ew_apply(int op, float a, float b) {
     case EW_ADD: return a + b;
     case EW_SUB: return a - b;
     case EW_MUL: return a * b;
-    default:     return a / b;
+    case EW_DIV: return a / b;
+    case EW_MIN: return fminf(a, b);
+    default:     return fmaxf(a, b);
   }
 }
 
@@ -166,6 +170,29 @@ __global__ void outer_k(float* __restrict__ out, const float* __restrict__ u,
   }
 }
 
+// ---- value semi-join: keep a[i] iff its value appears in the SORTED
+// needle array (binary search per element); else zero. Used by the
+// relational value-join (SURVEY.md §2.1 "joins on matrix indices/values").
+template <typename T>
+__global__ void mask_isin_k(T* __restrict__ out, const T* __restrict__ a,
+                            const float* __restrict__ needles, int nn,
+                            long n) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n; i += stride) {
+    float v = (float)a[i];
+    int lo = 0, hi = nn - 1;
+    bool found = false;
+    while (lo <= hi) {
+      int mid = (lo + hi) >> 1;
+      float m = needles[mid];
+      if (m == v) { found = true; break; }
+      if (m < v) lo = mid + 1; else hi = mid - 1;
+    }
+    out[i] = found ? a[i] : (T)0.0f;
+  }
+}
+
 // ------------------------------------------------------------- launchers --
 
 static inline int ew_grid(long n, int per_thread) {
@@ -180,7 +207,9 @@ extern "C" void launch_ewise_bf16(int op, void* out, const void* a,
     case 0: hipLaunchKernelGGL(ewise_bf16_k<0>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
     case 1: hipLaunchKernelGGL(ewise_bf16_k<1>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
     case 2: hipLaunchKernelGGL(ewise_bf16_k<2>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
-    default: hipLaunchKernelGGL(ewise_bf16_k<3>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
+    case 3: hipLaunchKernelGGL(ewise_bf16_k<3>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
+    case 4: hipLaunchKernelGGL(ewise_bf16_k<4>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
+    default: hipLaunchKernelGGL(ewise_bf16_k<5>, g, blk, 0, s, (bf16*)out, (const bf16*)a, (const bf16*)b, n); break;
   }
 }
 
@@ -191,7 +220,9 @@ extern "C" void launch_ewise_f32(int op, float* out, const float* a,
     case 0: hipLaunchKernelGGL(ewise_f32_k<0>, g, blk, 0, s, out, a, b, n); break;
     case 1: hipLaunchKernelGGL(ewise_f32_k<1>, g, blk, 0, s, out, a, b, n); break;
     case 2: hipLaunchKernelGGL(ewise_f32_k<2>, g, blk, 0, s, out, a, b, n); break;
-    default: hipLaunchKernelGGL(ewise_f32_k<3>, g, blk, 0, s, out, a, b, n); break;
+    case 3: hipLaunchKernelGGL(ewise_f32_k<3>, g, blk, 0, s, out, a, b, n); break;
+    case 4: hipLaunchKernelGGL(ewise_f32_k<4>, g, blk, 0, s, out, a, b, n); break;
+    default: hipLaunchKernelGGL(ewise_f32_k<5>, g, blk, 0, s, out, a, b, n); break;
   }
 }
 
@@ -254,4 +285,18 @@ extern "C" void launch_outer_f32(int op, float* out, const float* u,
     hipLaunchKernelGGL(outer_k<0>, g, blk, 0, s, out, u, v, m, n);
   else
     hipLaunchKernelGGL(outer_k<1>, g, blk, 0, s, out, u, v, m, n);
+}
+
+extern "C" void launch_mask_isin_bf16(void* out, const void* a,
+                                      const float* needles, int nn, long n,
+                                      hipStream_t s) {
+  hipLaunchKernelGGL(mask_isin_k<bf16>, dim3(ew_grid(n, 1)), dim3(256), 0, s,
+                     (bf16*)out, (const bf16*)a, needles, nn, n);
+}
+
+extern "C" void launch_mask_isin_f32(float* out, const float* a,
+                                     const float* needles, int nn, long n,
+                                     hipStream_t s) {
+  hipLaunchKernelGGL(mask_isin_k<float>, dim3(ew_grid(n, 1)), dim3(256), 0,
+                     s, out, a, needles, nn, n);
 }

@@ -32,6 +32,10 @@ void launch_mask_rows_cols_f32(float*, const float*, const unsigned char*,
                                hipStream_t);
 void launch_outer_f32(int, float*, const float*, const float*, long, long,
                       hipStream_t);
+void launch_mask_isin_bf16(void*, const void*, const float*, int, long,
+                           hipStream_t);
+void launch_mask_isin_f32(float*, const float*, const float*, int, long,
+                          hipStream_t);
 void launch_reduce_full_bf16(float*, float*, const void*, long, int,
                              hipStream_t);
 void launch_reduce_full_f32(float*, float*, const float*, long, int,
@@ -171,6 +175,22 @@ torch::Tensor outer(int64_t op, torch::Tensor u, torch::Tensor v) {
   return out;
 }
 
+torch::Tensor mask_isin(torch::Tensor a, torch::Tensor needles_sorted) {
+  check_2d(a, "a");
+  auto nd = needles_sorted.to(a.device(), torch::kFloat32).contiguous();
+  auto out = torch::empty_like(a);
+  long n = a.numel();
+  if (is_bf16(a))
+    launch_mask_isin_bf16(out.data_ptr(), a.data_ptr(),
+                          nd.data_ptr<float>(), (int)nd.numel(), n,
+                          cur_stream());
+  else
+    launch_mask_isin_f32(out.data_ptr<float>(), a.data_ptr<float>(),
+                         nd.data_ptr<float>(), (int)nd.numel(), n,
+                         cur_stream());
+  return out;
+}
+
 torch::Tensor reduce2d(torch::Tensor a, int64_t kind, int64_t axis) {
   check_2d(a, "a");
   long rows = a.size(0), cols = a.size(1);
@@ -254,6 +274,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mask_value", &mask_value, "value-predicate mask");
   m.def("mask_rows_cols", &mask_rows_cols, "index-predicate mask");
   m.def("outer", &outer, "outer cross-join");
+  m.def("mask_isin", &mask_isin, "value semi-join mask");
   m.def("reduce2d", &reduce2d, "full/row/col reduction");
   m.def("trace", &trace, "diagonal sum");
   m.def("transpose", &transpose, "materialized transpose");

@@ -150,6 +150,32 @@ class Matrix:
     def cross_join(self, other: "Matrix", op: str = "mul") -> "Matrix":
         return Matrix(L.CrossJoin(self.plan, other.plan, op))
 
+    def join_elem(self, other: "Matrix", op: str = "mul") -> "Matrix":
+        """Index-equality join on (i,j): combine co-located entries with
+        op in {mul, add, min, max} (MatRel join on indices)."""
+        return Matrix(L.Ewise(self.plan, other.plan, op))
+
+    def join_rows(self, other: "Matrix") -> "Matrix":
+        """Join on row index with multiply-aggregate combiner:
+        C[j,k] = sum_i A[i,j]*B[i,k] == A^T @ B (lowered to the MFMA
+        matmul path; chain/scheme optimizations apply)."""
+        return Matrix(L.MatMul(L.Transpose(self.plan), other.plan))
+
+    def join_cols(self, other: "Matrix") -> "Matrix":
+        """Join on column index: C[i,k] = sum_j A[i,j]*B[k,j] == A @ B^T."""
+        return Matrix(L.MatMul(self.plan, L.Transpose(other.plan)))
+
+    def semi_join_values(self, other: "Matrix") -> "Matrix":
+        """Keep entries of self whose value occurs among other's nonzero
+        values (MatRel join on values, semi-join form)."""
+        return Matrix(L.SemiJoinValue(self.plan, other.plan))
+
+    def elem_min(self, other: "Matrix") -> "Matrix":
+        return Matrix(L.Ewise(self.plan, other.plan, "min"))
+
+    def elem_max(self, other: "Matrix") -> "Matrix":
+        return Matrix(L.Ewise(self.plan, other.plan, "max"))
+
     # --- aggregates ------------------------------------------------------
 
     def _agg(self, kind: str, axis: Optional[int]) -> "Matrix":

@@ -27,8 +27,8 @@ from ..blocks.matrix import BlockMatrix
 from ..blocks.schemes import BCAST, COL, GRID, ROW, Scheme
 from ..ops import blockops
 from ..plan.logical import (Agg, CrossJoin, Ewise, Inverse, Leaf, MatMul,
-                            Plan, ScalarOp, Select, SelectPredicate, Solve,
-                            Transpose)
+                            Plan, ScalarOp, Select, SelectPredicate,
+                            SemiJoinValue, Solve, Transpose)
 from ..plan.physical import PhysicalPlan, plan_physical
 from .runtime import get_runtime
 from .summa import local_matmul, summa_matmul
@@ -79,6 +79,8 @@ class Executor:
             return self._inverse(n)
         if isinstance(n, CrossJoin):
             return self._crossjoin(n)
+        if isinstance(n, SemiJoinValue):
+            return self._semijoin_value(n)
         raise TypeError(f"cannot execute {type(n).__name__}")
 
     # --- operators ------------------------------------------------------
@@ -152,8 +154,9 @@ class Executor:
         out = BlockMatrix(a.nrows, a.ncols, a.block_size, a.scheme,
                           a.dtype, a.device)
         op = {"add": blockops.EW_ADD, "sub": blockops.EW_SUB,
-              "mul": blockops.EW_MUL, "div": blockops.EW_DIV}[n.op]
-        if n.op in ("add", "sub"):
+              "mul": blockops.EW_MUL, "div": blockops.EW_DIV,
+              "min": blockops.EW_MIN, "max": blockops.EW_MAX}[n.op]
+        if n.op in ("add", "sub", "min", "max"):
             keys = set(a.blocks) | set(b.blocks)
         elif n.op == "mul":
             keys = set(a.blocks) & set(b.blocks)
@@ -350,6 +353,18 @@ class Executor:
         x = torch.linalg.inv(a_full)
         ref = self._eval(n.children[0])
         return _vector_result(x.float(), ref, tuple(x.shape))
+
+    def _semijoin_value(self, n: SemiJoinValue) -> BlockMatrix:
+        a = self._eval(n.children[0])
+        b = self._eval(n.children[1])
+        # gather B's distinct nonzero values (small by contract)
+        bv = b.to_dense().reshape(-1)
+        needles = torch.unique(bv[bv != 0].float())
+        out = BlockMatrix(a.nrows, a.ncols, a.block_size, a.scheme,
+                          a.dtype, a.device)
+        for idx, t in a.local_items():
+            out.blocks[idx] = blockops.mask_isin(t, needles)
+        return out
 
     def _crossjoin(self, n: CrossJoin) -> BlockMatrix:
         a = self._eval(n.children[0])  # m x 1

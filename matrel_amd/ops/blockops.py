@@ -26,7 +26,7 @@ import torch
 from .backend import require_ext, use_hip
 
 # elementwise op ids shared with csrc/elementwise.hip
-EW_ADD, EW_SUB, EW_MUL, EW_DIV = 0, 1, 2, 3
+EW_ADD, EW_SUB, EW_MUL, EW_DIV, EW_MIN, EW_MAX = 0, 1, 2, 3, 4, 5
 # scalar op ids
 SC_ADD, SC_MUL, SC_POW = 0, 1, 2
 # reduction kinds shared with csrc/reduce.hip
@@ -35,6 +35,7 @@ RED_SUM, RED_MAX, RED_MIN, RED_NNZ, RED_SQSUM = 0, 1, 2, 3, 4
 _EW_TORCH = {
     EW_ADD: torch.add, EW_SUB: torch.sub,
     EW_MUL: torch.mul, EW_DIV: torch.div,
+    EW_MIN: torch.minimum, EW_MAX: torch.maximum,
 }
 
 _TILE_MN = 128
@@ -193,5 +194,14 @@ def mask_value(a: torch.Tensor, lo: float, hi: float,
                                         bool(keep_inside))
     inside = (a >= lo) & (a <= hi)
     keep = inside if keep_inside else ~inside
+    return torch.where(keep, a, torch.zeros((), dtype=a.dtype,
+                                            device=a.device))
+
+
+def mask_isin(a: torch.Tensor, needles_sorted: torch.Tensor) -> torch.Tensor:
+    """Keep entries of `a` whose value is in the sorted needle set."""
+    if use_hip(a):
+        return require_ext().mask_isin(a.contiguous(), needles_sorted)
+    keep = torch.isin(a.float(), needles_sorted.float())
     return torch.where(keep, a, torch.zeros((), dtype=a.dtype,
                                             device=a.device))

@@ -168,10 +168,12 @@ class ScalarOp(Plan):
 
 
 class Ewise(Plan):
-    """Elementwise A op B; shapes must match."""
+    """Elementwise A op B; shapes must match. Ops min/max realize the
+    reference's index-equality join with min/max combiners (an index join
+    on (i,j)=(i,j) IS an elementwise combine - SURVEY.md §2.1 joins)."""
 
     def __init__(self, l: Plan, r: Plan, op: str) -> None:
-        assert op in ("add", "sub", "mul", "div")
+        assert op in ("add", "sub", "mul", "div", "min", "max")
         if l.shape != r.shape:
             raise ValueError(f"ewise {op}: shape {l.shape} vs {r.shape}")
         super().__init__([l, r])
@@ -184,7 +186,7 @@ class Ewise(Plan):
     @property
     def density(self) -> float:
         da, db = self.children[0].density, self.children[1].density
-        if self.op in ("add", "sub"):
+        if self.op in ("add", "sub", "min", "max"):
             return min(1.0, da + db - da * db)
         if self.op == "mul":
             return da * db
@@ -192,7 +194,7 @@ class Ewise(Plan):
 
     def key(self) -> tuple:
         k = (self.children[0].key(), self.children[1].key())
-        if self.op in ("add", "mul"):  # commutative: canonical order
+        if self.op in ("add", "mul", "min", "max"):  # commutative
             k = tuple(sorted(k))
         return ("e" + self.op,) + k
 
@@ -449,3 +451,27 @@ class CrossJoin(Plan):
 
     def with_children(self, ch: List[Plan]) -> Plan:
         return CrossJoin(ch[0], ch[1], self.op)
+
+
+class SemiJoinValue(Plan):
+    """Value join (semi-join form): keep entries of A whose VALUE appears
+    among the nonzero values of B; zero the rest. Reference: MatRel's join
+    on matrix values. B must be small enough to gather
+    its distinct values host-side (the common analytics case)."""
+
+    def __init__(self, l: Plan, r: Plan) -> None:
+        super().__init__([l, r])
+
+    @property
+    def shape(self):
+        return self.children[0].shape
+
+    @property
+    def density(self) -> float:
+        return self.children[0].density * 0.5
+
+    def key(self) -> tuple:
+        return ("semijoinv", self.children[0].key(), self.children[1].key())
+
+    def with_children(self, ch):
+        return SemiJoinValue(ch[0], ch[1])

@@ -175,7 +175,7 @@ def push_select(p: Plan) -> Plan:
                                               col_hi=pr.col_hi))
                 r = rule(r)
             return MatMul(l, r)
-        if isinstance(c, Ewise) and c.op in ("add", "sub", "mul"):
+        if isinstance(c, Ewise) and c.op in ("add", "sub", "mul", "min", "max"):
             l = rule(Select(c.children[0], pr))
             r = rule(Select(c.children[1], pr))
             return Ewise(l, r, c.op)

@@ -27,7 +27,7 @@ from typing import Dict, Optional
 from ..blocks.schemes import (BCAST, COL, GRID, ROW, Scheme)
 from ..exec.runtime import get_runtime
 from .logical import (Agg, CrossJoin, Ewise, Inverse, Leaf, MatMul, Plan,
-                      ScalarOp, Select, Solve, Transpose)
+                      ScalarOp, Select, SemiJoinValue, Solve, Transpose)
 
 # Operands at or below this many bytes are candidates for replication
 # (broadcast-MM). 288 GB HBM3E per GPU means replication is cheap in memory;
@@ -74,7 +74,7 @@ def plan_physical(root: Plan) -> PhysicalPlan:
             pi = PhysInfo(scheme=n.data.scheme)
         elif isinstance(n, Transpose):
             pi = PhysInfo(scheme=ch[0].scheme.transpose())
-        elif isinstance(n, (ScalarOp, Select)):
+        elif isinstance(n, (ScalarOp, Select, SemiJoinValue)):
             pi = PhysInfo(scheme=ch[0].scheme)
         elif isinstance(n, Ewise):
             # align the smaller operand to the bigger one's scheme
