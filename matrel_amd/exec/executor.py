@@ -31,6 +31,7 @@ from ..plan.logical import (Agg, CrossJoin, Ewise, Inverse, Leaf, MatMul,
                             SemiJoinValue, Solve, Transpose)
 from ..plan.physical import PhysicalPlan, plan_physical
 from .runtime import get_runtime
+from .metrics import record_op
 from .summa import local_matmul, summa_matmul
 from .tracing import trace_op
 
@@ -50,7 +51,10 @@ class Executor:
     def _eval(self, n: Plan) -> BlockMatrix:
         if id(n) in self.cache:
             return self.cache[id(n)]
-        with trace_op(type(n).__name__, shape=n.shape):
+        pi = self.physical.get(n)
+        with trace_op(type(n).__name__, shape=n.shape), \
+                record_op(type(n).__name__, pi.est_flops,
+                          pi.est_comm_bytes):
             out = self._dispatch(n)
         tgt = self.physical.get(n).scheme
         if out.scheme != tgt and get_runtime().is_distributed:

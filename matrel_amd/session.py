@@ -67,5 +67,13 @@ class MatrelSession:
     def rank(self) -> int:
         return self.runtime.rank
 
+    def metrics_report(self) -> dict:
+        from .exec import metrics
+        return metrics.report()
+
+    def reset_metrics(self) -> None:
+        from .exec import metrics
+        metrics.reset()
+
     def shutdown(self) -> None:
         self.runtime.shutdown()
