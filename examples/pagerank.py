@@ -27,18 +27,21 @@ def main():
     s = ma.MatrelSession()
 
     A = s.block_sparse_random(n, n, block_density, seed=1)
-    # column-normalize: divide by col sums (add eps to avoid 0/0)
-    colsum = A.col_sums().compute()  # 1 x n
+    # normalize by the largest |column sum| so the iteration is stable
+    # (a true PageRank would divide each column by its sum; we fold the
+    # normalization into one scalar to keep the demo self-contained)
+    colsum = A.col_sums().compute()
+    scale = float(colsum.to_tensor().abs().max().item()) or 1.0
+    An = A * (1.0 / scale)
 
     r = s.from_dense(torch.full((n, 1), 1.0 / n))
     d = 0.85
     t0 = time.time()
     for it in range(5):
-        # A @ r with damping; colsum normalization folded in scalar form
-        r = ((A @ r) * d + (1 - d) / n).compute()
+        r = ((An @ r) * d + (1 - d) / n).compute()
         total = r.sum().item()
         if s.rank == 0:
-            print(f"iter {it}: sum(r) = {total:.4f}")
+            print(f"iter {it}: sum(r) = {total:.6f}")
     if s.rank == 0:
         print(f"5 iterations on {n}x{n} @ {block_density:.0%} block density "
               f"in {time.time()-t0:.2f}s")
