@@ -12,6 +12,7 @@ void launch_gemm_bt_bf16_acc(float*, const void*, const void*, int, int,
                              int, hipStream_t);
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
+void set_gemm_variant(int);
 void launch_ewise_bf16(int, void*, const void*, const void*, long,
                        hipStream_t);
 void launch_ewise_f32(int, float*, const float*, const float*, long,
@@ -269,6 +270,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "matrel_amd hand-written CDNA4 (gfx950) kernels";
   m.def("gemm_bt_acc", &gemm_bt_acc,
         "C(f32) += A @ Bt^T (bf16 MFMA / exact-f32 MFMA)");
+  m.def("set_gemm_variant", [](int64_t v) { set_gemm_variant((int)v); },
+        "0=auto 1=simple256 2=phased256 3=tile128 (A/B testing)");
   m.def("ewise", &ewise, "elementwise a OP b");
   m.def("ewise_scalar", &ewise_scalar, "elementwise a OP s");
   m.def("mask_value", &mask_value, "value-predicate mask");

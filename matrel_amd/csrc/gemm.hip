@@ -318,7 +318,172 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
     }
 }
 
+// ------------------------------------- bf16, 256^2 tile, 4-phase pipe ----
+// The counted-vmcnt phase pipeline (cdna_hip_programming.md §5 "The 256^2
+// 8-phase template", T3+T4+T5): per K-tile, the 8 waves TOGETHER compute
+// one 128x128 C-quadrant per phase over the full K=64, so every wave
+// consumes the tile's four 16-KiB half-tiles (A0,B0,B1,A1) in the SAME
+// order. Each phase stages ONE half-tile of tile t+1 (2 glds/thread) and
+// ends with s_waitcnt vmcnt(4) + raw s_barrier - the glds queue is never
+// drained to 0 inside the loop (the last two staged halves stay in
+// flight across every barrier). Quadrant snake order (0,0)->(0,1)->
+// (1,1)->(1,0) reuses one operand's fragments at each transition.
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
+                        const bf16* __restrict__ Bt, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: buf b (2) x half h (A0,A1,B0,B1) x 16 KiB
+  bf16* half_base[2][4];
+#pragma unroll
+  for (int b = 0; b < 2; ++b)
+#pragma unroll
+    for (int h = 0; h < 4; ++h)
+      half_base[b][h] = (bf16*)(smem + (b * 4 + h) * (128 * BKH * 2));
+  // h: 0=A0 1=A1 2=B0 3=B1
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  // within a quadrant: wave w -> 32-row x 64-col slice
+  const u32 qr = (wave & 3) * 32;   // row offset in the 128x128 quadrant
+  const u32 qc = (wave >> 2) * 64;  // col offset
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  const u32 tm = bid / ntn, tn = bid % ntn;
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  // staging: one half-tile = 128 rows x 64 cols = 1024 x 16 B; 512 threads
+  // x 2 glds. Same XOR-16 source swizzle as the other kernels.
+  const u32 srow = tid >> 3, sslot = tid & 7;  // rows 0..63
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  // stage half h (0=A0,1=A1,2=B0,3=B1) of K-tile t into buffer buf
+  auto stage_half = [&](int buf, int h, int t) {
+    const long k0 = (long)t * BKH;
+    const bf16* src = (h < 2)
+        ? A + (m0 + (h & 1) * 128) * (long)K
+        : Bt + (n0 + (h & 1) * 128) * (long)K;
+    bf16* dst = half_base[buf][h];
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      u32 row = srow + it * 64;
+      glds16(src + row * (long)K + k0 + src_slot * 8,
+             dst + (row * 8u + sslot) * 8u);
+    }
+  };
+
+  f32x4 acc[4][2][4] = {};  // [quadrant][m:2][n:4]
+  bf16x8 af[2][2], bfr[2][4];  // [kk][m] / [kk][n]
+
+  const u32 arow_base = qr + (lane & 15);
+  const u32 brow_base = qc + (lane & 15);
+  const u32 kgrp = lane >> 4;
+
+  auto read_a = [&](const bf16* base) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        u32 row = arow_base + m * 16;
+        u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        af[kk][m] = *(const bf16x8*)((const char*)base + byte);
+      }
+  };
+  auto read_b = [&](const bf16* base) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        u32 row = brow_base + n * 16;
+        u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        bfr[kk][n] = *(const bf16x8*)((const char*)base + byte);
+      }
+  };
+
+#define MFMA_QUAD(q)                                                       \
+  __builtin_amdgcn_s_setprio(1);                                          \
+  _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                         \
+      _Pragma("unroll") for (int m = 0; m < 2; ++m)                        \
+          _Pragma("unroll") for (int n = 0; n < 4; ++n)                    \
+              acc[q][m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(      \
+                  af[kk][m], bfr[kk][n], acc[q][m][n], 0, 0, 0);           \
+  __builtin_amdgcn_s_setprio(0);
+
+  // prologue: stage tile 0 completely in consumption-priority order
+  stage_half(0, 0, 0);  // A0
+  stage_half(0, 2, 0);  // B0
+  stage_half(0, 3, 0);  // B1
+  stage_half(0, 1, 0);  // A1
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // A0,B0 landed
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < NT; ++t) {
+    const int cur = t & 1, nxt = cur ^ 1;
+    const bool pre = (t + 1 < NT);
+    // ---- phase 1: Q(0,0) - reads A0,B0; stages A0 of t+1
+    read_a(half_base[cur][0]);
+    read_b(half_base[cur][2]);
+    if (pre) stage_half(nxt, 0, t + 1);
+    __builtin_amdgcn_s_barrier();
+    MFMA_QUAD(0)
+    // drain B1(t); leave {A1(t), A0'} (or {A1} when not staging)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 2: Q(0,1) - reads B1 (A frags reused); stages B0'
+    read_b(half_base[cur][3]);
+    if (pre) stage_half(nxt, 2, t + 1);
+    __builtin_amdgcn_s_barrier();
+    MFMA_QUAD(1)
+    // drain A1(t)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 3: Q(1,1) - reads A1 (B frags reused); stages B1'
+    read_a(half_base[cur][1]);
+    if (pre) stage_half(nxt, 3, t + 1);
+    __builtin_amdgcn_s_barrier();
+    MFMA_QUAD(2)
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 4: Q(1,0) - re-reads B0 (landed long ago); stages A1'
+    read_b(half_base[cur][2]);
+    if (pre) stage_half(nxt, 1, t + 1);
+    __builtin_amdgcn_s_barrier();
+    MFMA_QUAD(3)
+    // end of tile: drain up to {B1', A1'} for next P1
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: C(f32) += acc; quadrant (mh,nh) at rows mh*128, cols nh*128
+  const u32 crow0 = qr + (lane >> 4) * 4;
+  const u32 ccol0 = qc + (lane & 15);
+  const int QMH[4] = {0, 0, 1, 1};
+  const int QNH[4] = {0, 1, 1, 0};
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        long row = m0 + QMH[q] * 128 + crow0 + m * 16;
+        long col = n0 + QNH[q] * 128 + ccol0 + n * 16;
+        float* cp = C + row * (long)N + col;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[q][m][n][r];
+      }
+}
+#undef MFMA_QUAD
+
 // ------------------------------------------------------------- launchers --
+
+static int g_gemm_variant = 0;  // 0=auto, 1=simple 256, 2=phased 256, 3=128
+extern "C" void set_gemm_variant(int v) { g_gemm_variant = v; }
 
 extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
                                         const void* Bt, int M, int N, int K,
@@ -326,11 +491,20 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
   // 256^2 tile only when the grid still fills the chip (>= ~192 of 256
   // CUs at 1 block/CU); smaller problems keep the 128^2 tile (2 blocks/CU,
   // 4x the block count).
-  if (M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 192) {
+  bool big = M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 192;
+  int v = g_gemm_variant;
+  if (v == 0) v = big ? 2 : 3;
+  if (v != 3 && M % BM2 == 0 && N % BN2 == 0) {
     dim3 grid((M / BM2) * (N / BN2));
     size_t lds = 2u * BM2 * BKH * 2 * 2;  // 128 KiB
-    hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds, stream,
-                       C, (const bf16*)A, (const bf16*)Bt, M, N, K);
+    if (v == 2)
+      hipLaunchKernelGGL(gemm_bt_bf16_acc_256p_k, grid, dim3(512), lds,
+                         stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+                         K);
+    else
+      hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
+                         stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+                         K);
     return;
   }
   dim3 grid((M / BM) * (N / BN));
