@@ -334,14 +334,9 @@ __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
                         const bf16* __restrict__ Bt, int M, int N, int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: buf b (2) x half h (A0,A1,B0,B1) x 16 KiB
-  bf16* half_base[2][4];
-#pragma unroll
-  for (int b = 0; b < 2; ++b)
-#pragma unroll
-    for (int h = 0; h < 4; ++h)
-      half_base[b][h] = (bf16*)(smem + (b * 4 + h) * (128 * BKH * 2));
-  // h: 0=A0 1=A1 2=B0 3=B1
+  // layout: buf b (2) x half h (0=A0 1=A1 2=B0 3=B1) x 16 KiB; all LDS
+  // addressing is 32-bit offsets from one base (keeps VGPR pressure down).
+#define HOFF(b, h) (((b) * 4u + (h)) * (128u * BKH * 2u))
 
   const u32 tid = threadIdx.x;
   const u32 lane = tid & 63, wave = tid >> 6;
@@ -360,13 +355,14 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   const u32 src_slot = sslot ^ (srow & 7);
   const int NT = K / BKH;
 
-  // stage half h (0=A0,1=A1,2=B0,3=B1) of K-tile t into buffer buf
-  auto stage_half = [&](int buf, int h, int t) {
+  // stage half h (0=A0,1=A1,2=B0,3=B1) of K-tile t into buffer buf.
+  // h is a literal at every call site, so src/dst fold at compile time.
+  auto stage_half = [&](u32 bufoff, int h, int t) {
     const long k0 = (long)t * BKH;
     const bf16* src = (h < 2)
         ? A + (m0 + (h & 1) * 128) * (long)K
         : Bt + (n0 + (h & 1) * 128) * (long)K;
-    bf16* dst = half_base[buf][h];
+    bf16* dst = (bf16*)(smem + bufoff + (u32)h * (128u * BKH * 2u));
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       u32 row = srow + it * 64;
@@ -382,7 +378,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   const u32 brow_base = qc + (lane & 15);
   const u32 kgrp = lane >> 4;
 
-  auto read_a = [&](const bf16* base) {
+  auto read_a = [&](u32 off) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -390,10 +386,10 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
         u32 row = arow_base + m * 16;
         u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
         byte ^= (row & 7) << 4;
-        af[kk][m] = *(const bf16x8*)((const char*)base + byte);
+        af[kk][m] = *(const bf16x8*)(smem + off + byte);
       }
   };
-  auto read_b = [&](const bf16* base) {
+  auto read_b = [&](u32 off) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -401,7 +397,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
         u32 row = brow_base + n * 16;
         u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
         byte ^= (row & 7) << 4;
-        bfr[kk][n] = *(const bf16x8*)((const char*)base + byte);
+        bfr[kk][n] = *(const bf16x8*)(smem + off + byte);
       }
   };
 
@@ -415,19 +411,20 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   __builtin_amdgcn_s_setprio(0);
 
   // prologue: stage tile 0 completely in consumption-priority order
-  stage_half(0, 0, 0);  // A0
-  stage_half(0, 2, 0);  // B0
-  stage_half(0, 3, 0);  // B1
-  stage_half(0, 1, 0);  // A1
+  stage_half(0u, 0, 0);  // A0
+  stage_half(0u, 2, 0);  // B0
+  stage_half(0u, 3, 0);  // B1
+  stage_half(0u, 1, 0);  // A1
   asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // A0,B0 landed
   __builtin_amdgcn_s_barrier();
 
+  const u32 BUFSZ = 4u * 128u * BKH * 2u;
   for (int t = 0; t < NT; ++t) {
-    const int cur = t & 1, nxt = cur ^ 1;
+    const u32 cur = (u32)(t & 1) * BUFSZ, nxt = cur ^ BUFSZ;
     const bool pre = (t + 1 < NT);
     // ---- phase 1: Q(0,0) - reads A0,B0; stages A0 of t+1
-    read_a(half_base[cur][0]);
-    read_b(half_base[cur][2]);
+    read_a(cur + HOFF(0, 0));
+    read_b(cur + HOFF(0, 2));
     if (pre) stage_half(nxt, 0, t + 1);
     __builtin_amdgcn_s_barrier();
     MFMA_QUAD(0)
@@ -436,7 +433,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     // ---- phase 2: Q(0,1) - reads B1 (A frags reused); stages B0'
-    read_b(half_base[cur][3]);
+    read_b(cur + HOFF(0, 3));
     if (pre) stage_half(nxt, 2, t + 1);
     __builtin_amdgcn_s_barrier();
     MFMA_QUAD(1)
@@ -445,13 +442,13 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     // ---- phase 3: Q(1,1) - reads A1 (B frags reused); stages B1'
-    read_a(half_base[cur][1]);
+    read_a(cur + HOFF(0, 1));
     if (pre) stage_half(nxt, 3, t + 1);
     __builtin_amdgcn_s_barrier();
     MFMA_QUAD(2)
     __builtin_amdgcn_s_barrier();
     // ---- phase 4: Q(1,0) - re-reads B0 (landed long ago); stages A1'
-    read_b(half_base[cur][2]);
+    read_b(cur + HOFF(0, 2));
     if (pre) stage_half(nxt, 1, t + 1);
     __builtin_amdgcn_s_barrier();
     MFMA_QUAD(3)
@@ -479,6 +476,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
       }
 }
 #undef MFMA_QUAD
+#undef HOFF
 
 // ------------------------------------------------------------- launchers --
 
