@@ -329,6 +329,7 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
 // flight across every barrier). Quadrant snake order (0,0)->(0,1)->
 // (1,1)->(1,0) reuses one operand's fragments at each transition.
 
+template <bool MIDBAR>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
@@ -426,7 +427,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     read_a(cur + HOFF(0, 0));
     read_b(cur + HOFF(0, 2));
     if (pre) stage_half(nxt, 0, t + 1);
-    __builtin_amdgcn_s_barrier();
+    if (MIDBAR) __builtin_amdgcn_s_barrier();
     MFMA_QUAD(0)
     // drain B1(t); leave {A1(t), A0'} (or {A1} when not staging)
     if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
@@ -435,7 +436,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     // ---- phase 2: Q(0,1) - reads B1 (A frags reused); stages B0'
     read_b(cur + HOFF(0, 3));
     if (pre) stage_half(nxt, 2, t + 1);
-    __builtin_amdgcn_s_barrier();
+    if (MIDBAR) __builtin_amdgcn_s_barrier();
     MFMA_QUAD(1)
     // drain A1(t)
     if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
@@ -444,13 +445,13 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     // ---- phase 3: Q(1,1) - reads A1 (B frags reused); stages B1'
     read_a(cur + HOFF(0, 1));
     if (pre) stage_half(nxt, 3, t + 1);
-    __builtin_amdgcn_s_barrier();
+    if (MIDBAR) __builtin_amdgcn_s_barrier();
     MFMA_QUAD(2)
     __builtin_amdgcn_s_barrier();
     // ---- phase 4: Q(1,0) - re-reads B0 (landed long ago); stages A1'
     read_b(cur + HOFF(0, 2));
     if (pre) stage_half(nxt, 1, t + 1);
-    __builtin_amdgcn_s_barrier();
+    if (MIDBAR) __builtin_amdgcn_s_barrier();
     MFMA_QUAD(3)
     // end of tile: drain up to {B1', A1'} for next P1
     if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
@@ -489,16 +490,20 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
   // 256^2 tile only when the grid still fills the chip (>= ~192 of 256
   // CUs at 1 block/CU); smaller problems keep the 128^2 tile (2 blocks/CU,
   // 4x the block count).
-  bool big = M % BM2 == 0 && N % BN2 == 0 && (M / BM2) * (N / BN2) >= 192;
+  int nb = (M / BM2) * (N / BN2);
   int v = g_gemm_variant;
-  if (v == 0) v = big ? 2 : 3;
+  if (v == 0) v = (nb >= 512) ? 4 : (nb >= 192 ? 1 : 3);
   if (v != 3 && M % BM2 == 0 && N % BN2 == 0) {
-    dim3 grid((M / BM2) * (N / BN2));
+    dim3 grid((unsigned)nb);
     size_t lds = 2u * BM2 * BKH * 2 * 2;  // 128 KiB
     if (v == 2)
-      hipLaunchKernelGGL(gemm_bt_bf16_acc_256p_k, grid, dim3(512), lds,
-                         stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
-                         K);
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<true>), grid, dim3(512),
+                         lds, stream, C, (const bf16*)A, (const bf16*)Bt, M,
+                         N, K);
+    else if (v == 4)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false>), grid, dim3(512),
+                         lds, stream, C, (const bf16*)A, (const bf16*)Bt, M,
+                         N, K);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
