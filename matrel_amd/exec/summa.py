@@ -122,46 +122,50 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         return
 
     def issue_bcast(k: int):
-        """Allocate/stack panels for step k and issue async broadcasts."""
-        # A panel: blocks (i, k) for i in my_is; src rank = (p, k % pc)
+        """Stack panels for step k and issue async broadcasts.
+
+        Only PRESENT blocks enter the panel (the presence maps are
+        globally synced, so every rank in the sub-group derives the same
+        stack layout) - a 1%-block-density operand broadcasts 1% of the
+        bytes (config 4).
+        """
+        # A panel: present blocks (i, k) for i in my_is; src = (p, k % pc)
         a_src = rt.rank_of(p, k % pc)
-        a_stack = torch.zeros(len(my_is), bs, bs, dtype=a.dtype,
+        a_rows = [i for i in my_is if pa[i, k]]
+        a_stack = torch.zeros(len(a_rows), bs, bs, dtype=a.dtype,
                               device=a.device)
         if rt.rank == a_src:
-            for n, i in enumerate(my_is):
+            for n, i in enumerate(a_rows):
                 t = a.get_block(i, k)
                 if t is not None:
                     h, w = t.shape
                     a_stack[n, :h, :w] = t
         wa = dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
                             async_op=True)
-        # B panel: blocks (k, j) for j in my_js; src rank = (k % pr, q)
+        # B panel: present blocks (k, j) for j in my_js; src = (k % pr, q)
         b_src = rt.rank_of(k % pr, q)
-        b_stack = torch.zeros(len(my_js), bs, bs, dtype=b.dtype,
+        b_cols = [j for j in my_js if pb[k, j]]
+        b_stack = torch.zeros(len(b_cols), bs, bs, dtype=b.dtype,
                               device=b.device)
         if rt.rank == b_src:
-            for n, j in enumerate(my_js):
+            for n, j in enumerate(b_cols):
                 t = b.get_block(k, j)
                 if t is not None:
                     h, w = t.shape
                     b_stack[n, :h, :w] = t
         wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
                             async_op=True)
-        return a_stack, b_stack, wa, wb
+        return a_rows, a_stack, b_cols, b_stack, wa, wb
 
     cur = issue_bcast(ks[0])
     for step, k in enumerate(ks):
         nxt = issue_bcast(ks[step + 1]) if step + 1 < len(ks) else None
-        a_stack, b_stack, wa, wb = cur
+        a_rows, a_stack, b_cols, b_stack, wa, wb = cur
         wa.wait()
         wb.wait()
-        for n_j, j in enumerate(my_js):
-            if not pb[k, j]:
-                continue
+        for n_j, j in enumerate(b_cols):
             bt = _pre_t(b_stack[n_j])
-            for n_i, i in enumerate(my_is):
-                if not pa[i, k]:
-                    continue
+            for n_i, i in enumerate(a_rows):
                 c = acc.get((i, j))
                 if c is None:
                     c = torch.zeros(bs, bs, dtype=torch.float32,
