@@ -80,9 +80,10 @@ def main():
     sync()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks
+    # MAX over ranks (RCCL needs device tensors; gloo needs CPU)
     if rt.is_distributed:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        dev = rt.device if rt.backend == "nccl" else torch.device("cpu")
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

@@ -121,6 +121,32 @@ class BlockMatrix:
         assert t.shape == self.block_shape(bi, bj), \
             f"block {(bi, bj)}: {tuple(t.shape)} != {self.block_shape(bi, bj)}"
         self.blocks[(bi, bj)] = t.to(device=self.device, dtype=self.dtype)
+        self._presence = None  # invalidate cached presence map
+
+    def coll_device(self) -> torch.device:
+        """Device for small bookkeeping collectives: RCCL needs GPU
+        tensors, gloo needs CPU tensors."""
+        rt = get_runtime()
+        return self.device if rt.backend == "nccl" else torch.device("cpu")
+
+    def presence(self) -> torch.Tensor:
+        """Global (grid_rows x grid_cols) uint8 map of present blocks
+        (one all_reduce; cached). Every rank agrees on the result, which
+        lets repartition/transpose/SUMMA skip absent blocks without
+        per-block flag exchanges."""
+        cached = getattr(self, "_presence", None)
+        if cached is not None:
+            return cached
+        rt = get_runtime()
+        pm = torch.zeros(self.grid_rows, self.grid_cols, dtype=torch.uint8)
+        for (bi, bj) in self.blocks:
+            pm[bi, bj] = 1
+        if rt.is_distributed and not self.scheme.is_replicated():
+            dev_pm = pm.to(self.coll_device())
+            dist.all_reduce(dev_pm, op=dist.ReduceOp.MAX)
+            pm = dev_pm.cpu()
+        self._presence = pm
+        return pm
 
     def local_items(self) -> Iterator[Tuple[BlockIndex, torch.Tensor]]:
         return iter(self.blocks.items())
@@ -132,13 +158,8 @@ class BlockMatrix:
 
     def block_density(self) -> float:
         """Fraction of blocks present, across all ranks."""
-        rt = get_runtime()
         total = self.grid_rows * self.grid_cols
-        if not rt.is_distributed or self.scheme.is_replicated():
-            return len(self.blocks) / total
-        t = torch.tensor([len(self.blocks)], dtype=torch.int64)
-        dist.all_reduce(t)
-        return int(t.item()) / total
+        return float(self.presence().sum().item()) / total
 
     # --- constructors ---------------------------------------------------
 
@@ -250,6 +271,7 @@ class BlockMatrix:
         if not rt.is_distributed:
             out.blocks = dict(self.blocks)
             return out
+        pm = self.presence()
         for (bi, bj) in self.block_indices():
             src = self.owner(bi, bj)
             dst = scheme.owner(bi, bj, rt.world_size, rt.grid)
@@ -260,12 +282,7 @@ class BlockMatrix:
                     if t is not None:
                         out.blocks[(bi, bj)] = t
                 continue
-            # presence flag must be agreed: broadcast nnz flag first
-            flag = torch.tensor(
-                [1 if (rt.rank == src and (bi, bj) in self.blocks) else 0],
-                dtype=torch.int32)
-            dist.broadcast(flag, src=src)
-            if flag.item() == 0:
+            if not pm[bi, bj]:
                 continue
             if dst == -1:
                 t = (self.blocks[(bi, bj)] if rt.rank == src else

@@ -105,15 +105,12 @@ class Executor:
         for (bi, bj), t in a.local_items():
             tmp.blocks[(bj, bi)] = blockops.transpose(t)
         # blocks now live where A(j,i)'s owner was; redistribute explicitly
+        pm = a.presence()
         out_sch = out.scheme
         for (bi, bj) in tmp.block_indices():
             src = a.owner(bj, bi)  # rank holding transposed block (bi,bj)
             dst = out_sch.owner(bi, bj, rt.world_size, rt.grid)
-            flag = torch.tensor(
-                [1 if (rt.rank == src and (bi, bj) in tmp.blocks) else 0],
-                dtype=torch.int32)
-            dist.broadcast(flag, src=src)
-            if flag.item() == 0:
+            if not pm[bj, bi]:
                 continue
             if src == dst:
                 if rt.rank == src:

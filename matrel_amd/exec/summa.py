@@ -30,19 +30,7 @@ from .runtime import Runtime, get_runtime
 
 
 def _presence_map(m: BlockMatrix) -> torch.Tensor:
-    """Global (grid_rows x grid_cols) bool map of present blocks.
-    One all_reduce; cached on the matrix."""
-    cached = getattr(m, "_presence", None)
-    if cached is not None:
-        return cached
-    rt = get_runtime()
-    pm = torch.zeros(m.grid_rows, m.grid_cols, dtype=torch.uint8)
-    for (bi, bj) in m.blocks:
-        pm[bi, bj] = 1
-    if rt.is_distributed and not m.scheme.is_replicated():
-        dist.all_reduce(pm, op=dist.ReduceOp.MAX)
-    m._presence = pm
-    return pm
+    return m.presence()
 
 
 def _pre_t(b: torch.Tensor) -> torch.Tensor:
