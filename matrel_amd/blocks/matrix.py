@@ -30,6 +30,10 @@ from ..config import get_config
 from ..exec.runtime import get_runtime
 from .schemes import BCAST, GRID, Scheme
 
+
+def _densify(t: torch.Tensor) -> torch.Tensor:
+    return t.to_dense() if t.layout == torch.sparse_csr else t
+
 BlockIndex = Tuple[int, int]
 
 
@@ -110,12 +114,13 @@ class BlockMatrix:
         return self.blocks.get((bi, bj))
 
     def get_block_dense(self, bi: int, bj: int) -> torch.Tensor:
-        """Local block, materializing zeros for absent blocks."""
+        """Local block as a DENSE tensor (zeros for absent blocks, CSR
+        blocks densified)."""
         t = self.blocks.get((bi, bj))
         if t is None:
             t = torch.zeros(self.block_shape(bi, bj), dtype=self.dtype,
                             device=self.device)
-        return t
+        return _densify(t)
 
     def set_block(self, bi: int, bj: int, t: torch.Tensor) -> None:
         assert t.shape == self.block_shape(bi, bj), \
@@ -198,17 +203,38 @@ class BlockMatrix:
             m, n, lambda s, g: torch.rand(*s, generator=g), seed, **kw)
 
     @classmethod
-    def from_dense(cls, t: torch.Tensor, **kw) -> "BlockMatrix":
-        """Split a full tensor (available on every rank) into blocks."""
+    def from_dense(cls, t: torch.Tensor, sparsify: bool = True,
+                   **kw) -> "BlockMatrix":
+        """Split a full tensor (available on every rank) into blocks.
+        Blocks below the config sparse_threshold are stored CSR
+        (reference: auto dense/sparse per block by nnz)."""
         assert t.dim() == 2
+        from ..ops.blockops import maybe_sparsify
+        thr = get_config().sparse_threshold if sparsify else 0.0
         out = cls(t.shape[0], t.shape[1], **kw)
         bs = out.block_size
         for (bi, bj) in out.owned_indices():
             blk = t[bi * bs:(bi + 1) * bs, bj * bs:(bj + 1) * bs]
             if blk.count_nonzero() == 0:
                 continue  # keep zero blocks implicit
-            out.blocks[(bi, bj)] = blk.contiguous().to(
-                device=out.device, dtype=out.dtype)
+            dense = blk.contiguous().to(device=out.device, dtype=out.dtype)
+            out.blocks[(bi, bj)] = maybe_sparsify(dense, thr)
+        return out
+
+    @classmethod
+    def sparse_random(cls, m: int, n: int, density: float, seed: int = 0,
+                      **kw) -> "BlockMatrix":
+        """Element-level sparse random matrix: every block stored CSR with
+        ~density fraction of nonzeros, reproducible by block index."""
+        out = cls(m, n, **kw)
+        for (bi, bj) in out.owned_indices():
+            g = torch.Generator(device="cpu")
+            g.manual_seed((seed * 999983 + bi * 131 + bj) & 0x7FFFFFFF)
+            shape = out.block_shape(bi, bj)
+            mask = torch.rand(*shape, generator=g) < density
+            vals = torch.randn(*shape, generator=g) * mask
+            t = vals.to(device=out.device, dtype=out.dtype)
+            out.blocks[(bi, bj)] = t.to_sparse_csr()
         return out
 
     @classmethod
@@ -240,6 +266,7 @@ class BlockMatrix:
             if owner == -1 or not rt.is_distributed:
                 t = self.blocks.get((bi, bj))
                 if t is not None:
+                    t = _densify(t)
                     full[bi * bs:bi * bs + t.shape[0],
                          bj * bs:bj * bs + t.shape[1]] = t.float()
             else:
@@ -285,7 +312,8 @@ class BlockMatrix:
             if not pm[bi, bj]:
                 continue
             if dst == -1:
-                t = (self.blocks[(bi, bj)] if rt.rank == src else
+                t = (_densify(self.blocks[(bi, bj)]).contiguous()
+                     if rt.rank == src else
                      torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
                                  device=self.device))
                 dist.broadcast(t, src=src)
@@ -295,7 +323,8 @@ class BlockMatrix:
                     out.blocks[(bi, bj)] = self.blocks[(bi, bj)]
             else:
                 if rt.rank == src:
-                    dist.send(self.blocks[(bi, bj)].contiguous(), dst=dst)
+                    dist.send(_densify(self.blocks[(bi, bj)]).contiguous(),
+                              dst=dst)
                 elif rt.rank == dst:
                     t = torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
                                     device=self.device)

@@ -53,6 +53,10 @@ int reduce_cols_nchunks(long, long);
 void launch_trace_bf16(float*, float*, const void*, long, long, hipStream_t);
 void launch_trace_f32(float*, float*, const float*, long, long, hipStream_t);
 void launch_transpose_bf16(void*, const void*, long, long, hipStream_t);
+void launch_spmm_csr_acc_bf16(float*, const int*, const int*, const void*,
+                              const void*, long, long, hipStream_t);
+void launch_spmm_csr_acc_f32(float*, const int*, const int*, const float*,
+                             const float*, long, long, hipStream_t);
 void launch_transpose_f32(float*, const float*, long, long, hipStream_t);
 }
 
@@ -252,6 +256,32 @@ torch::Tensor trace(torch::Tensor a) {
   return out;
 }
 
+// C(f32, MxN) += CSR(M x K) @ B(K x N)
+void spmm_csr_acc(torch::Tensor C, torch::Tensor crow, torch::Tensor cols,
+                  torch::Tensor vals, torch::Tensor B) {
+  check_2d(C, "C");
+  check_2d(B, "B");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32, "C must be fp32");
+  TORCH_CHECK(crow.scalar_type() == torch::kInt32 &&
+              cols.scalar_type() == torch::kInt32, "indices must be int32");
+  long M = C.size(0), N = C.size(1);
+  TORCH_CHECK(B.size(1) == N, "N mismatch");
+  TORCH_CHECK(crow.numel() == M + 1, "crow length");
+  auto cr = crow.contiguous();
+  auto cl = cols.contiguous();
+  auto vl = vals.contiguous();
+  if (vals.scalar_type() == torch::kBFloat16) {
+    TORCH_CHECK(is_bf16(B), "vals/B dtype mismatch");
+    launch_spmm_csr_acc_bf16(C.data_ptr<float>(), cr.data_ptr<int>(),
+                             cl.data_ptr<int>(), vl.data_ptr(), B.data_ptr(),
+                             M, N, cur_stream());
+  } else {
+    launch_spmm_csr_acc_f32(C.data_ptr<float>(), cr.data_ptr<int>(),
+                            cl.data_ptr<int>(), vl.data_ptr<float>(),
+                            B.data_ptr<float>(), M, N, cur_stream());
+  }
+}
+
 torch::Tensor transpose(torch::Tensor a) {
   check_2d(a, "a");
   auto out = torch::empty({a.size(1), a.size(0)}, a.options());
@@ -281,4 +311,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce2d", &reduce2d, "full/row/col reduction");
   m.def("trace", &trace, "diagonal sum");
   m.def("transpose", &transpose, "materialized transpose");
+  m.def("spmm_csr_acc", &spmm_csr_acc, "C(f32) += CSR @ B (within-block sparse)");
 }

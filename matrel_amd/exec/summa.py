@@ -54,7 +54,8 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             bb = b.get_block(k, bj)
             if bb is None:
                 continue
-            bt = _pre_t(bb)
+            bbd = blockops.densify(bb)
+            bt = None  # transposed lazily; only dense A blocks need it
             for bi in my_is:
                 if not out.is_local(bi, bj):
                     continue
@@ -66,7 +67,12 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                     c = torch.zeros(out.block_shape(bi, bj),
                                     dtype=torch.float32, device=out.device)
                     acc[(bi, bj)] = c
-                blockops.gemm_bt_acc(c, ab, bt)
+                if blockops.is_sparse(ab):
+                    blockops.spmm_acc(c, ab, bbd)
+                else:
+                    if bt is None:
+                        bt = _pre_t(bbd)
+                    blockops.gemm_bt_acc(c, ab, bt)
     for (bi, bj), c in acc.items():
         out.blocks[(bi, bj)] = c.to(out.dtype)
 
@@ -126,6 +132,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             for n, i in enumerate(a_rows):
                 t = a.get_block(i, k)
                 if t is not None:
+                    t = blockops.densify(t)
                     h, w = t.shape
                     a_stack[n, :h, :w] = t
         wa = dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
@@ -139,6 +146,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             for n, j in enumerate(b_cols):
                 t = b.get_block(k, j)
                 if t is not None:
+                    t = blockops.densify(t)
                     h, w = t.shape
                     b_stack[n, :h, :w] = t
         wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),

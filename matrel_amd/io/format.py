@@ -57,6 +57,8 @@ def save_matrix(m: BlockMatrix, path: str) -> None:
     for (bi, bj), t in m.local_items():
         if m.owner(bi, bj) == -1 and rt.rank != 0:
             continue  # replicated blocks written once
+        if t.layout == torch.sparse_csr:
+            t = t.to_dense()
         cpu = t.detach().cpu().contiguous()
         raw = cpu.view(torch.uint16) if cpu.dtype == torch.bfloat16 else cpu
         arr = raw.numpy()

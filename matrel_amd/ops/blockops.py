@@ -41,6 +41,43 @@ _EW_TORCH = {
 _TILE_MN = 128
 
 
+def is_sparse(t: torch.Tensor) -> bool:
+    return t.layout == torch.sparse_csr
+
+
+def densify(t: torch.Tensor) -> torch.Tensor:
+    return t.to_dense() if is_sparse(t) else t
+
+
+def maybe_sparsify(t: torch.Tensor, threshold: float) -> torch.Tensor:
+    """Store a block CSR when its density is below `threshold` (the
+    reference's per-block auto dense/sparse selection by nnz,
+    SURVEY.md §2.1 "Block-matrix representation")."""
+    if is_sparse(t) or threshold <= 0:
+        return t
+    nnz = int((t != 0).sum().item())
+    if t.numel() and nnz / t.numel() < threshold:
+        return t.to_sparse_csr()
+    return t
+
+
+def spmm_acc(acc: torch.Tensor, a_csr: torch.Tensor,
+             b: torch.Tensor) -> None:
+    """acc(fp32, MxN) += a_csr(M x K) @ b(K x N) - within-block sparse
+    product (HIP CSR kernel on GPU)."""
+    if acc.is_cuda:
+        ext = require_ext()
+        crow = a_csr.crow_indices().to(torch.int32)
+        cols = a_csr.col_indices().to(torch.int32)
+        vals = a_csr.values()
+        if vals.dtype != b.dtype:
+            vals, b = vals.float(), b.float()
+        ext.spmm_csr_acc(acc, crow, cols, vals.contiguous(),
+                         b.contiguous())
+    else:
+        acc.add_(a_csr.to_dense().float() @ b.float())
+
+
 def _kmult(dtype: torch.dtype) -> int:
     return 64 if dtype == torch.bfloat16 else 32
 
@@ -98,6 +135,7 @@ def gemm(a: torch.Tensor, b: torch.Tensor,
 
 
 def ewise(op: int, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    a, b = densify(a), densify(b)
     if use_hip(a):
         if a.dtype != b.dtype:
             a, b = a.float(), b.float()
@@ -106,6 +144,7 @@ def ewise(op: int, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
 
 
 def ewise_scalar(op: int, a: torch.Tensor, s: float) -> torch.Tensor:
+    a = densify(a)
     if use_hip(a):
         return require_ext().ewise_scalar(op, a.contiguous(), float(s))
     if op == SC_ADD:
@@ -119,6 +158,7 @@ def ewise_scalar(op: int, a: torch.Tensor, s: float) -> torch.Tensor:
 
 def transpose(a: torch.Tensor) -> torch.Tensor:
     """Materialized transpose (tiled LDS kernel on GPU)."""
+    a = densify(a)
     if use_hip(a):
         return require_ext().transpose(a.contiguous())
     return a.t().contiguous()
@@ -128,6 +168,14 @@ def reduce2d(a: torch.Tensor, kind: int, axis: Optional[int]
              ) -> torch.Tensor:
     """Reduce a block. axis=None -> 1-elem tensor, axis=0 -> per-column,
     axis=1 -> per-row. fp32 output."""
+    if is_sparse(a):
+        if axis is None and kind in (RED_SUM, RED_NNZ, RED_SQSUM):
+            # reduce over the values vector directly (zeros contribute 0)
+            v = a.values().reshape(1, -1)
+            if v.numel() == 0:
+                return torch.zeros(1, dtype=torch.float32, device=a.device)
+            return reduce2d(v.contiguous(), kind, None)
+        a = a.to_dense()
     if use_hip(a):
         return require_ext().reduce2d(a.contiguous(), kind,
                                       -1 if axis is None else axis)
@@ -151,6 +199,7 @@ def reduce2d(a: torch.Tensor, kind: int, axis: Optional[int]
 
 def trace_block(a: torch.Tensor) -> torch.Tensor:
     """Sum of the main diagonal (1-elem fp32 tensor)."""
+    a = densify(a)
     if use_hip(a):
         return require_ext().trace(a.contiguous())
     return torch.diagonal(a.float()).sum().reshape(1)
@@ -172,6 +221,7 @@ def mask_rows_cols(a: torch.Tensor, row_mask: Optional[torch.Tensor],
     """Zero out rows/cols excluded by boolean masks (boundary blocks of an
     index-predicate selection; fully excluded blocks are pruned at plan
     time and never reach a kernel - SURVEY.md §3.4)."""
+    a = densify(a)
     if use_hip(a):
         e = require_ext()
         rm = row_mask.to(a.device) if row_mask is not None else None
@@ -189,6 +239,7 @@ def mask_value(a: torch.Tensor, lo: float, hi: float,
                keep_inside: bool) -> torch.Tensor:
     """Value-predicate selection: keep entries inside [lo, hi] (or outside),
     zeroing the rest. Fused single-pass on GPU."""
+    a = densify(a)
     if use_hip(a):
         return require_ext().mask_value(a.contiguous(), float(lo), float(hi),
                                         bool(keep_inside))
@@ -200,6 +251,7 @@ def mask_value(a: torch.Tensor, lo: float, hi: float,
 
 def mask_isin(a: torch.Tensor, needles_sorted: torch.Tensor) -> torch.Tensor:
     """Keep entries of `a` whose value is in the sorted needle set."""
+    a = densify(a)
     if use_hip(a):
         return require_ext().mask_isin(a.contiguous(), needles_sorted)
     keep = torch.isin(a.float(), needles_sorted.float())

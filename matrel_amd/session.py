@@ -49,6 +49,12 @@ class MatrelSession:
                                              seed=seed, **kw)
         return Matrix.from_blocks(bm, density=block_density)
 
+    def sparse_random(self, m: int, n: int, density: float,
+                      seed: int = 0, **kw) -> Matrix:
+        """Element-level sparse random matrix (per-block CSR storage)."""
+        bm = BlockMatrix.sparse_random(m, n, density, seed=seed, **kw)
+        return Matrix.from_blocks(bm, density=density)
+
     def load(self, path: str, scheme: Optional[Scheme] = None) -> Matrix:
         bm = load_matrix(path, scheme=scheme)
         return Matrix.from_blocks(

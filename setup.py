@@ -22,6 +22,7 @@ sources = [
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "reduce.hip"),
     os.path.join(CSRC, "transpose.hip"),
+    os.path.join(CSRC, "spmm.hip"),
 ]
 
 setup(
