@@ -106,12 +106,15 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     # fp32 accumulators for owned C blocks
     acc: Dict[Tuple[int, int], torch.Tensor] = {}
 
-    # skip k-steps that contribute nothing anywhere for this rank's tiles
-    def k_useful(k: int) -> bool:
-        return (any(pa[i, k] for i in my_is) and
-                any(pb[k, j] for j in my_js))
-
-    ks = [k for k in range(gk) if k_useful(k)]
+    # Broadcast participation must be UNIFORM across each sub-group
+    # (collectives deadlock otherwise): the A-panel predicate depends only
+    # on the grid ROW (my_is is shared by the whole row group), the
+    # B-panel predicate only on the grid COLUMN. A rank therefore joins
+    # the A broadcast at steps ka, the B broadcast at steps kb, and
+    # computes at ka & kb.
+    ka = {k for k in range(gk) if any(pa[i, k] for i in my_is)}
+    kb = {k for k in range(gk) if any(pb[k, j] for j in my_js)}
+    ks = sorted(ka | kb)
     if not ks:
         return
 
@@ -121,44 +124,52 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         Only PRESENT blocks enter the panel (the presence maps are
         globally synced, so every rank in the sub-group derives the same
         stack layout) - a 1%-block-density operand broadcasts 1% of the
-        bytes (config 4).
+        bytes (config 4). A rank joins the A (B) broadcast only at steps
+        in ka (kb) - group-uniform by construction.
         """
-        # A panel: present blocks (i, k) for i in my_is; src = (p, k % pc)
-        a_src = rt.rank_of(p, k % pc)
-        a_rows = [i for i in my_is if pa[i, k]]
-        a_stack = torch.zeros(len(a_rows), bs, bs, dtype=a.dtype,
-                              device=a.device)
-        if rt.rank == a_src:
-            for n, i in enumerate(a_rows):
-                t = a.get_block(i, k)
-                if t is not None:
-                    t = blockops.densify(t)
-                    h, w = t.shape
-                    a_stack[n, :h, :w] = t
-        wa = dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
-                            async_op=True)
-        # B panel: present blocks (k, j) for j in my_js; src = (k % pr, q)
-        b_src = rt.rank_of(k % pr, q)
-        b_cols = [j for j in my_js if pb[k, j]]
-        b_stack = torch.zeros(len(b_cols), bs, bs, dtype=b.dtype,
-                              device=b.device)
-        if rt.rank == b_src:
-            for n, j in enumerate(b_cols):
-                t = b.get_block(k, j)
-                if t is not None:
-                    t = blockops.densify(t)
-                    h, w = t.shape
-                    b_stack[n, :h, :w] = t
-        wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
-                            async_op=True)
+        a_rows, a_stack, wa = [], None, None
+        if k in ka:
+            a_src = rt.rank_of(p, k % pc)
+            a_rows = [i for i in my_is if pa[i, k]]
+            a_stack = torch.zeros(len(a_rows), bs, bs, dtype=a.dtype,
+                                  device=a.device)
+            if rt.rank == a_src:
+                for n, i in enumerate(a_rows):
+                    t = a.get_block(i, k)
+                    if t is not None:
+                        t = blockops.densify(t)
+                        h, w = t.shape
+                        a_stack[n, :h, :w] = t
+            wa = dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
+                                async_op=True)
+        b_cols, b_stack, wb = [], None, None
+        if k in kb:
+            b_src = rt.rank_of(k % pr, q)
+            b_cols = [j for j in my_js if pb[k, j]]
+            b_stack = torch.zeros(len(b_cols), bs, bs, dtype=b.dtype,
+                                  device=b.device)
+            if rt.rank == b_src:
+                for n, j in enumerate(b_cols):
+                    t = b.get_block(k, j)
+                    if t is not None:
+                        t = blockops.densify(t)
+                        h, w = t.shape
+                        b_stack[n, :h, :w] = t
+            wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
+                                async_op=True)
         return a_rows, a_stack, b_cols, b_stack, wa, wb
 
     cur = issue_bcast(ks[0])
     for step, k in enumerate(ks):
         nxt = issue_bcast(ks[step + 1]) if step + 1 < len(ks) else None
         a_rows, a_stack, b_cols, b_stack, wa, wb = cur
-        wa.wait()
-        wb.wait()
+        if wa is not None:
+            wa.wait()
+        if wb is not None:
+            wb.wait()
+        if not (k in ka and k in kb):
+            cur = nxt
+            continue
         for n_j, j in enumerate(b_cols):
             bt = _pre_t(b_stack[n_j])
             for n_i, i in enumerate(a_rows):

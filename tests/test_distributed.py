@@ -99,6 +99,17 @@ def _body_block_sparse(s, tmpdir):
     assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
 
 
+def _body_sparse_skewed(s, tmpdir):
+    """Sparse x sparse with skewed block presence: exercises the
+    group-uniform broadcast-participation logic (ka/kb) in SUMMA."""
+    for seed in (1, 2, 3, 4):
+        A = s.block_sparse_random(160, 160, 0.3, seed=seed)
+        B = s.block_sparse_random(160, 96, 0.3, seed=seed + 50)
+        C = (A @ B).compute()
+        ref = A.to_tensor() @ B.to_tensor()
+        assert torch.allclose(C.to_tensor(), ref, atol=1e-3), seed
+
+
 def _body_save_load(s, tmpdir):
     A = s.randn(96, 64, seed=10)
     path = os.path.join(tmpdir, "dist_mat")
@@ -120,8 +131,8 @@ def _body_ols(s, tmpdir):
 
 @pytest.mark.parametrize("body", [
     "_body_summa", "_body_bcast_mm", "_body_aggregates", "_body_transpose",
-    "_body_repartition", "_body_block_sparse", "_body_save_load",
-    "_body_ols",
+    "_body_repartition", "_body_block_sparse", "_body_sparse_skewed",
+    "_body_save_load", "_body_ols",
 ])
 def test_distributed(body, tmp_path):
     _run(body, tmp_path)
