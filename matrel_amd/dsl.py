@@ -271,6 +271,13 @@ class Matrix:
         walk(opt, 0, set())
         return "\n".join(lines)
 
+    def compile(self):
+        """Capture this expression's kernel sequence in a hipGraph and
+        return a replayable callable (exec/graphcap.py). Falls back to
+        eager when the plan is not capturable."""
+        from .exec.graphcap import compile_expression
+        return compile_expression(self)
+
     def compute(self) -> "Matrix":
         """Optimize + plan + execute; returns a materialized Matrix."""
         from .exec.executor import Executor

@@ -247,3 +247,16 @@ def test_block_sparse_gpu(gpu_session):
     C = (A @ B).compute()
     ref = A.to_tensor() @ B.to_tensor()
     assert (C.to_tensor() - ref).abs().max().item() < 1.0
+
+
+def test_compile_hipgraph_replay(gpu_session):
+    """hipGraph capture + replay of a small expression (exec/graphcap)."""
+    s = gpu_session
+    A = s.randn(1024, 1024, seed=30)
+    B = s.randn(1024, 1024, seed=31)
+    fn = (A @ B + A).compile()
+    out1 = fn().to_tensor().clone()
+    out2 = fn().to_tensor()
+    ref = A.to_tensor() @ B.to_tensor() + A.to_tensor()
+    assert (out1 - ref).abs().max().item() < 1.0
+    assert torch.equal(out1, out2)
