@@ -30,6 +30,23 @@ __device__ __forceinline__ u32 xcd_swizzle(u32 bid, u32 nwg) {
   return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + bid / NXCD;
 }
 
+// ---- supertile decomposition: virtual id -> (tm, tn) such that ids
+// 0..SM*SN-1 form a 2D SM x SN cluster of output tiles (the ~32 tiles
+// co-resident on one XCD then share SM A-strips + SN B-strips instead of
+// 32 disjoint ones: the GEMM at >=8k is HBM-bound, so strip reuse through
+// the XCD's L2/LLC is the main lever past ~1 TF/s... per-GPU TFLOP).
+// Requires ntm % SM == 0 && ntn % SN == 0 (caller guards).
+__device__ __forceinline__ void supertile_map(u32 id, u32 ntm, u32 ntn,
+                                              u32 SM, u32 SN, u32* tm,
+                                              u32* tn) {
+  u32 per = SM * SN;
+  u32 st = id / per, w = id % per;
+  u32 stn_cnt = ntn / SN;
+  u32 stm = st / stn_cnt, stn = st % stn_cnt;
+  *tm = stm * SM + w / SN;
+  *tn = stn * SN + w % SN;
+}
+
 // ---- ordered-int encoding for float atomic max/min
 __device__ __forceinline__ int f32_order(float x) {
   int i = __float_as_int(x);

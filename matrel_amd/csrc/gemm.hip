@@ -241,7 +241,13 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
 
   const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
   u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
-  const u32 tm = bid / ntn, tn = bid % ntn;
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 8 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 8, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
   const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
 
   // staging: tile = 256 rows x 64 cols x 2 B; row = 8 x 16-B slots.
@@ -347,7 +353,13 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
 
   const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
   u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
-  const u32 tm = bid / ntn, tn = bid % ntn;
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 8 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 8, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
   const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
 
   // staging: one half-tile = 128 rows x 64 cols = 1024 x 16 B; 512 threads

@@ -159,3 +159,18 @@ def test_full_optimize_pipeline_correct(session):
     a, b, c = A.to_tensor(), B.to_tensor(), C.to_tensor()
     ref = 3.0 * (a @ b @ c)
     assert torch.allclose(out, ref, atol=1e-3)
+
+
+def test_supertile_map_bijective():
+    """Python replica of csrc/common.h supertile_map: must be a bijection
+    id -> (tm, tn) for every guarded geometry."""
+    def smap(i, ntm, ntn, SM=4, SN=8):
+        per = SM * SN
+        st, w = divmod(i, per)
+        stm, stn = divmod(st, ntn // SN)
+        return (stm * SM + w // SN, stn * SN + w % SN)
+
+    for ntm, ntn in [(8, 8), (16, 16), (32, 32), (64, 64), (4, 8), (12, 24)]:
+        seen = {smap(i, ntm, ntn) for i in range(ntm * ntn)}
+        assert len(seen) == ntm * ntn
+        assert all(0 <= tm < ntm and 0 <= tn < ntn for tm, tn in seen)
