@@ -176,10 +176,15 @@ class BlockMatrix:
     def _filled(cls, m: int, n: int, fill, seed: Optional[int], **kw
                 ) -> "BlockMatrix":
         out = cls(m, n, **kw)
+        # generate directly on the target device (a 16384^2 block takes
+        # ~ms on GPU vs seconds through host randn + H2D); per-block
+        # seeding keeps the fill reproducible per (seed, bi, bj) within a
+        # device type.
+        gen_dev = "cuda" if out.device.type == "cuda" else "cpu"
         for (bi, bj) in out.owned_indices():
             shape = out.block_shape(bi, bj)
             if seed is not None:
-                g = torch.Generator(device="cpu")
+                g = torch.Generator(device=gen_dev)
                 g.manual_seed((seed * 1000003 + bi * 131 + bj) & 0x7FFFFFFF)
                 t = fill(shape, g).to(device=out.device, dtype=out.dtype)
             else:
@@ -194,13 +199,19 @@ class BlockMatrix:
         on (seed, i, j), so the same logical matrix materializes under any
         scheme / world size (used by tests and the bench)."""
         return cls._filled(
-            m, n, lambda s, g: torch.randn(*s, generator=g), seed, **kw)
+            m, n,
+            lambda s, g: torch.randn(*s, generator=g,
+                                     device=g.device if g else None),
+            seed, **kw)
 
     @classmethod
     def rand(cls, m: int, n: int, seed: Optional[int] = 0, **kw
              ) -> "BlockMatrix":
         return cls._filled(
-            m, n, lambda s, g: torch.rand(*s, generator=g), seed, **kw)
+            m, n,
+            lambda s, g: torch.rand(*s, generator=g,
+                                    device=g.device if g else None),
+            seed, **kw)
 
     @classmethod
     def from_dense(cls, t: torch.Tensor, sparsify: bool = True,

@@ -459,7 +459,9 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     if (pre) stage_half(nxt, 3, t + 1);
     if (MIDBAR) __builtin_amdgcn_s_barrier();
     MFMA_QUAD(2)
-    __builtin_amdgcn_s_barrier();
+    // no barrier here: phase 4 reads B0(cur) - landed and barriered at
+    // the tile boundary - and stages into the OTHER buffer.
+    if (MIDBAR) __builtin_amdgcn_s_barrier();
     // ---- phase 4: Q(1,0) - re-reads B0 (landed long ago); stages A1'
     read_b(cur + HOFF(0, 2));
     if (pre) stage_half(nxt, 1, t + 1);
@@ -491,6 +493,145 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
 #undef MFMA_QUAD
 #undef HOFF
 
+// --------------------------- bf16, 256^2 tile, phased, 32x32x16 MFMA ----
+// Same quadrant-per-phase pipeline as gemm_bt_bf16_acc_256p_k but on
+// v_mfma_f32_32x32x16_bf16 (uarch ceiling 2382 TF vs 2075 for 16x16x32;
+// half the MFMA instructions at 2x the flops each). Per quadrant a wave
+// owns a 32 x 64 slice = 1 M-frag x 2 N-frags of 32x32, K=16 per MFMA
+// (4 kk steps over BK=64). C/D map: col = lane&31,
+// row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_bf16_acc_256p32_k(float* __restrict__ C, const bf16* __restrict__ A,
+                          const bf16* __restrict__ Bt, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#define HOFF32(b, h) (((b) * 4u + (h)) * (128u * BKH * 2u))
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 qr = (wave & 3) * 32;
+  const u32 qc = (wave >> 2) * 64;
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 8 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 8, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  const u32 srow = tid >> 3, sslot = tid & 7;
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  auto stage_half = [&](u32 bufoff, int h, int t) {
+    const long k0 = (long)t * BKH;
+    const bf16* src = (h < 2)
+        ? A + (m0 + (h & 1) * 128) * (long)K
+        : Bt + (n0 + (h & 1) * 128) * (long)K;
+    bf16* dst = (bf16*)(smem + bufoff + (u32)h * (128u * BKH * 2u));
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      u32 row = srow + it * 64;
+      glds16(src + row * (long)K + k0 + src_slot * 8,
+             dst + (row * 8u + sslot) * 8u);
+    }
+  };
+
+  f32x16 acc[4][2] = {};   // [quadrant][n-frag]
+  bf16x8 af[4], bfr[4][2];  // [kk] / [kk][n]
+
+  const u32 arow_base = qr + (lane & 31);
+  const u32 kgrp = lane >> 5;  // 0..1: 8-elem k sub-chunk within K=16
+
+  auto read_a = [&](u32 off) {
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      u32 row = arow_base;
+      u32 byte = row * 128u + (u32)kk * 32u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      af[kk] = *(const bf16x8*)(smem + off + byte);
+    }
+  };
+  auto read_b = [&](u32 off) {
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        u32 row = qc + n * 32 + (lane & 31);
+        u32 byte = row * 128u + (u32)kk * 32u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        bfr[kk][n] = *(const bf16x8*)(smem + off + byte);
+      }
+  };
+
+#define MFMA_QUAD32(q)                                                     \
+  __builtin_amdgcn_s_setprio(1);                                          \
+  _Pragma("unroll") for (int kk = 0; kk < 4; ++kk)                         \
+      _Pragma("unroll") for (int n = 0; n < 2; ++n)                        \
+          acc[q][n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
+              af[kk], bfr[kk][n], acc[q][n], 0, 0, 0);                     \
+  __builtin_amdgcn_s_setprio(0);
+
+  stage_half(0u, 0, 0);
+  stage_half(0u, 2, 0);
+  stage_half(0u, 3, 0);
+  stage_half(0u, 1, 0);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const u32 BUFSZ = 4u * 128u * BKH * 2u;
+  for (int t = 0; t < NT; ++t) {
+    const u32 cur = (u32)(t & 1) * BUFSZ, nxt = cur ^ BUFSZ;
+    const bool pre = (t + 1 < NT);
+    read_a(cur + HOFF32(0, 0));
+    read_b(cur + HOFF32(0, 2));
+    if (pre) stage_half(nxt, 0, t + 1);
+    MFMA_QUAD32(0)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    read_b(cur + HOFF32(0, 3));
+    if (pre) stage_half(nxt, 2, t + 1);
+    MFMA_QUAD32(1)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    read_a(cur + HOFF32(0, 1));
+    if (pre) stage_half(nxt, 3, t + 1);
+    MFMA_QUAD32(2)
+    __builtin_amdgcn_s_barrier();
+    read_b(cur + HOFF32(0, 2));
+    if (pre) stage_half(nxt, 1, t + 1);
+    MFMA_QUAD32(3)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: quadrant (mh,nh); C/D 32x32 map
+  const int QMH32[4] = {0, 0, 1, 1};
+  const int QNH32[4] = {0, 1, 1, 0};
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + QMH32[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        long col = n0 + QNH32[q] * 128 + qc + n * 32 + (lane & 31);
+        C[row * (long)N + col] += acc[q][n][r];
+      }
+    }
+}
+#undef MFMA_QUAD32
+#undef HOFF32
+
 // ------------------------------------------------------------- launchers --
 
 static int g_gemm_variant = 0;  // 0=auto, 1=simple 256, 2=phased 256, 3=128
@@ -516,6 +657,10 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false>), grid, dim3(512),
                          lds, stream, C, (const bf16*)A, (const bf16*)Bt, M,
                          N, K);
+    else if (v == 5)
+      hipLaunchKernelGGL(gemm_bt_bf16_acc_256p32_k, grid, dim3(512), lds,
+                         stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+                         K);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
