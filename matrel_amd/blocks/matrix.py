@@ -21,14 +21,14 @@ Supports ragged edge blocks (m, n need not divide block_size).
 from __future__ import annotations
 
 import math
-from typing import Dict, Iterator, List, Optional, Tuple
+from typing import Dict, Iterator, Optional, Tuple
 
 import torch
 import torch.distributed as dist
 
 from ..config import get_config
 from ..exec.runtime import get_runtime
-from .schemes import BCAST, GRID, Scheme
+from .schemes import GRID, Scheme
 
 
 def _densify(t: torch.Tensor) -> torch.Tensor:

@@ -24,11 +24,11 @@ import torch
 import torch.distributed as dist
 
 from ..blocks.matrix import BlockMatrix
-from ..blocks.schemes import BCAST, COL, GRID, ROW, Scheme
+from ..blocks.schemes import BCAST, COL, GRID, ROW
 from ..ops import blockops
 from ..plan.logical import (Agg, CrossJoin, Ewise, Inverse, Leaf, MatMul,
-                            Plan, ScalarOp, Select, SelectPredicate,
-                            SemiJoinValue, Solve, Transpose)
+                            Plan, ScalarOp, Select, SemiJoinValue, Solve,
+                            Transpose)
 from ..plan.physical import PhysicalPlan, plan_physical
 from .runtime import get_runtime
 from .metrics import record_op

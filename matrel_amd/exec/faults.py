@@ -20,7 +20,7 @@ import json
 import os
 import sys
 import traceback
-from typing import Callable, Dict, Optional
+from typing import Callable, Dict
 
 from ..blocks.matrix import BlockMatrix
 from ..io.format import load_matrix, save_matrix

@@ -18,7 +18,7 @@ config-1 path (BASELINE.json configs[0]) and most unit tests run.
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

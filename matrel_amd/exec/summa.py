@@ -18,15 +18,14 @@ repartitions operands first when the planner chose SUMMA.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 import torch.distributed as dist
 
 from ..blocks.matrix import BlockMatrix
-from ..blocks.schemes import GRID
 from ..ops import blockops
-from .runtime import Runtime, get_runtime
+from .runtime import get_runtime
 
 
 def _presence_map(m: BlockMatrix) -> torch.Tensor:

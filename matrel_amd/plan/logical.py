@@ -17,7 +17,7 @@ plan/optimizer.py, and lowered/executed by exec/executor.py.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional, Tuple
 
 from ..blocks.matrix import BlockMatrix

@@ -18,9 +18,8 @@ from __future__ import annotations
 from typing import Dict, List, Tuple
 
 from ..config import get_config
-from .logical import (Agg, CrossJoin, Ewise, Inverse, Leaf, MatMul, MatPower,
-                      Plan, ScalarOp, Select, SelectPredicate, Solve,
-                      Transpose)
+from .logical import (Ewise, MatMul, MatPower, Plan, ScalarOp, Select,
+                      SelectPredicate, Transpose)
 
 
 # --------------------------------------------------------------------------

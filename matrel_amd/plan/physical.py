@@ -21,7 +21,7 @@ N^2/Pr + N^2/Pc per step).
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, Optional
 
 from ..blocks.schemes import (BCAST, COL, GRID, ROW, Scheme)

@@ -13,10 +13,10 @@ from typing import Optional, Tuple
 import torch
 
 from .blocks.matrix import BlockMatrix
-from .blocks.schemes import GRID, Scheme
+from .blocks.schemes import Scheme
 from .config import MatrelConfig, get_config, set_config
 from .dsl import Matrix
-from .exec.runtime import get_runtime, init_runtime
+from .exec.runtime import init_runtime
 from .io.format import load_coo_text, load_matrix
 
 
