@@ -110,18 +110,30 @@ gemm_bt_bf16_acc_k(float* __restrict__ C, const bf16* __restrict__ A,
   }
 
   // epilogue: C(f32) += acc.  C/D map: col = lane&15, row = (lane>>4)*4+reg.
+  // two-pass (batched loads, one wait) - see the 256p epilogue note.
   const u32 crow0 = wr * 64 + (lane >> 4) * 4;
   const u32 ccol0 = wc * 64 + (lane & 15);
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < 4; ++m) {
+    float cv[4][4];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      const float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cv[n][r] = cp[(long)r * N];
+    }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       long row = m0 + crow0 + m * 16;
       long col = n0 + ccol0 + n * 16;
       float* cp = C + row * (long)N + col;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[m][n][r];
+      for (int r = 0; r < 4; ++r)
+        cp[(long)r * N] = cv[n][r] + acc[m][n][r];
     }
+  }
 }
 
 // ----------------------------------------------------------------- f32 ----
@@ -207,15 +219,26 @@ gemm_bt_f32_acc_k(float* __restrict__ C, const float* __restrict__ A,
   const u32 crow0 = wr * 64 + (lane >> 4) * 4;
   const u32 ccol0 = wc * 64 + (lane & 15);
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < 4; ++m) {
+    float cv[4][4];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      const float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cv[n][r] = cp[(long)r * N];
+    }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       long row = m0 + crow0 + m * 16;
       long col = n0 + ccol0 + n * 16;
       float* cp = C + row * (long)N + col;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[m][n][r];
+      for (int r = 0; r < 4; ++r)
+        cp[(long)r * N] = cv[n][r] + acc[m][n][r];
     }
+  }
 }
 
 // --------------------------------------------------- bf16, 256^2 tile ----
@@ -313,15 +336,26 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
   const u32 crow0 = wm * 128 + (lane >> 4) * 4;
   const u32 ccol0 = wn * 64 + (lane & 15);
 #pragma unroll
-  for (int m = 0; m < 8; ++m)
+  for (int m = 0; m < 8; ++m) {
+    float cv[4][4];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      const float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cv[n][r] = cp[(long)r * N];
+    }
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       long row = m0 + crow0 + m * 16;
       long col = n0 + ccol0 + n * 16;
       float* cp = C + row * (long)N + col;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[m][n][r];
+      for (int r = 0; r < 4; ++r)
+        cp[(long)r * N] = cv[n][r] + acc[m][n][r];
     }
+  }
 }
 
 // ------------------------------------- bf16, 256^2 tile, 4-phase pipe ----
@@ -472,13 +506,27 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
     __builtin_amdgcn_s_barrier();
   }
 
-  // epilogue: C(f32) += acc; quadrant (mh,nh) at rows mh*128, cols nh*128
+  // epilogue: C(f32) += acc; quadrant (mh,nh) at rows mh*128, cols nh*128.
+  // Two-pass per quadrant (gather 32 C values, then add+store) so the
+  // loads batch under ONE vmcnt instead of 128 serialized
+  // load->vmcnt(0)->add->store chains.
   const u32 crow0 = qr + (lane >> 4) * 4;
   const u32 ccol0 = qc + (lane & 15);
   const int QMH[4] = {0, 0, 1, 1};
   const int QNH[4] = {0, 1, 1, 0};
 #pragma unroll
-  for (int q = 0; q < 4; ++q)
+  for (int q = 0; q < 4; ++q) {
+    float cv[2][4][4];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        long row = m0 + QMH[q] * 128 + crow0 + m * 16;
+        long col = n0 + QNH[q] * 128 + ccol0 + n * 16;
+        const float* cp = C + row * (long)N + col;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) cv[m][n][r] = cp[(long)r * N];
+      }
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -487,8 +535,10 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
         long col = n0 + QNH[q] * 128 + ccol0 + n * 16;
         float* cp = C + row * (long)N + col;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) cp[(long)r * N] += acc[q][m][n][r];
+        for (int r = 0; r < 4; ++r)
+          cp[(long)r * N] = cv[m][n][r] + acc[q][m][n][r];
       }
+  }
 }
 #undef MFMA_QUAD
 #undef HOFF
@@ -617,17 +667,27 @@ gemm_bt_bf16_acc_256p32_k(float* __restrict__ C, const bf16* __restrict__ A,
   const int QMH32[4] = {0, 0, 1, 1};
   const int QNH32[4] = {0, 1, 1, 0};
 #pragma unroll
-  for (int q = 0; q < 4; ++q)
+  for (int q = 0; q < 4; ++q) {
+    float cv[2][16];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 2; ++n)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         long row = m0 + QMH32[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
                    4 * (lane >> 5);
         long col = n0 + QNH32[q] * 128 + qc + n * 32 + (lane & 31);
-        C[row * (long)N + col] += acc[q][n][r];
+        cv[n][r] = C[row * (long)N + col];
       }
-    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + QMH32[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        long col = n0 + QNH32[q] * 128 + qc + n * 32 + (lane & 31);
+        C[row * (long)N + col] = cv[n][r] + acc[q][n][r];
+      }
+  }
 }
 #undef MFMA_QUAD32
 #undef HOFF32
