@@ -369,7 +369,7 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
 // flight across every barrier). Quadrant snake order (0,0)->(0,1)->
 // (1,1)->(1,0) reuses one operand's fragments at each transition.
 
-template <bool MIDBAR>
+template <bool MIDBAR, bool STATICPRIO = false>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
@@ -449,13 +449,21 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   };
 
 #define MFMA_QUAD(q)                                                       \
-  __builtin_amdgcn_s_setprio(1);                                          \
+  if (!STATICPRIO) __builtin_amdgcn_s_setprio(1);                          \
   _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                         \
       _Pragma("unroll") for (int m = 0; m < 2; ++m)                        \
           _Pragma("unroll") for (int n = 0; n < 4; ++n)                    \
               acc[q][m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(      \
                   af[kk][m], bfr[kk][n], acc[q][m][n], 0, 0, 0);           \
-  __builtin_amdgcn_s_setprio(0);
+  if (!STATICPRIO) __builtin_amdgcn_s_setprio(0);
+
+  // T5 static form: one priority raise for the second-dispatched half
+  // (the arbitration loser on every segment), no per-cluster flips.
+  // Condition must be provably wave-uniform (readfirstlane) or the
+  // setprio applies to every wave (cdna_hip_programming.md T5 recipe).
+  if (STATICPRIO &&
+      __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
 
   // prologue: stage tile 0 completely in consumption-priority order
   stage_half(0u, 0, 0);  // A0
@@ -705,7 +713,7 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
   // 4x the block count).
   int nb = (M / BM2) * (N / BN2);
   int v = g_gemm_variant;
-  if (v == 0) v = (nb >= 512) ? 4 : (nb >= 192 ? 1 : 3);
+  if (v == 0) v = (nb >= 512) ? 6 : (nb >= 192 ? 1 : 3);
   if (v != 3 && M % BM2 == 0 && N % BN2 == 0) {
     dim3 grid((unsigned)nb);
     size_t lds = 2u * BM2 * BKH * 2 * 2;  // 128 KiB
@@ -721,6 +729,10 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256p32_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
                          K);
+    else if (v == 6)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true>), grid,
+                         dim3(512), lds, stream, C, (const bf16*)A,
+                         (const bf16*)Bt, M, N, K);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,

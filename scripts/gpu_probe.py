@@ -28,7 +28,7 @@ def main():
         a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
         bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
         c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
-        for v, name in ((1, "simple256"), (4, "phased-nb"), (5, "phased-nb-32x32"), (3, "tile128")):
+        for v, name in ((1, "simple256"), (4, "phased-nb"), (6, "phased-nb-sprio"), (3, "tile128")):
             ext.set_gemm_variant(v)
             dt = timeit(lambda: ext.gemm_bt_acc(c, a, bt))
             print(f"gemm {name} {n}^3: {2*n**3/dt/1e12:.0f} TFLOPS")
@@ -36,7 +36,7 @@ def main():
         ext.set_gemm_variant(1)
         c1 = torch.zeros(n, n, dtype=torch.float32, device="cuda")
         ext.gemm_bt_acc(c1, a, bt)
-        for v in (2, 4, 5):
+        for v in (2, 4, 5, 6):
             ext.set_gemm_variant(v)
             c2 = torch.zeros(n, n, dtype=torch.float32, device="cuda")
             ext.gemm_bt_acc(c2, a, bt)
