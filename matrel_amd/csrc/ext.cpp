@@ -13,6 +13,7 @@ void launch_gemm_bt_bf16_acc(float*, const void*, const void*, int, int,
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
 void set_gemm_variant(int);
+void set_supertile(int, int);
 void launch_ewise_bf16(int, void*, const void*, const void*, long,
                        hipStream_t);
 void launch_ewise_f32(int, float*, const float*, const float*, long,
@@ -302,6 +303,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "C(f32) += A @ Bt^T (bf16 MFMA / exact-f32 MFMA)");
   m.def("set_gemm_variant", [](int64_t v) { set_gemm_variant((int)v); },
         "0=auto 1=simple256 2=phased256 3=tile128 (A/B testing)");
+  m.def("set_supertile", [](int64_t sm, int64_t sn) {
+          set_supertile((int)sm, (int)sn);
+        }, "XCD supertile cluster dims (0 disables)");
   m.def("ewise", &ewise, "elementwise a OP b");
   m.def("ewise_scalar", &ewise_scalar, "elementwise a OP s");
   m.def("mask_value", &mask_value, "value-predicate mask");

@@ -373,7 +373,8 @@ template <bool MIDBAR, bool STATICPRIO = false>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
-                        const bf16* __restrict__ Bt, int M, int N, int K) {
+                        const bf16* __restrict__ Bt, int M, int N, int K,
+                        u32 stm_, u32 stn_) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // layout: buf b (2) x half h (0=A0 1=A1 2=B0 3=B1) x 16 KiB; all LDS
   // addressing is 32-bit offsets from one base (keeps VGPR pressure down).
@@ -388,8 +389,8 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
   u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
   u32 tm, tn;
-  if (ntm % 4 == 0 && ntn % 8 == 0) {
-    supertile_map(bid, ntm, ntn, 4, 8, &tm, &tn);
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
   } else {
     tm = bid / ntn;
     tn = bid % ntn;
@@ -704,6 +705,11 @@ gemm_bt_bf16_acc_256p32_k(float* __restrict__ C, const bf16* __restrict__ A,
 
 static int g_gemm_variant = 0;  // 0=auto, 1=simple 256, 2=phased 256, 3=128
 extern "C" void set_gemm_variant(int v) { g_gemm_variant = v; }
+static unsigned g_st_m = 4, g_st_n = 4;  // supertile cluster dims
+extern "C" void set_supertile(int sm, int sn) {
+  g_st_m = (unsigned)sm;
+  g_st_n = (unsigned)sn;
+}
 
 extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
                                         const void* Bt, int M, int N, int K,
@@ -720,11 +726,11 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
     if (v == 2)
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<true>), grid, dim3(512),
                          lds, stream, C, (const bf16*)A, (const bf16*)Bt, M,
-                         N, K);
+                         N, K, g_st_m, g_st_n);
     else if (v == 4)
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false>), grid, dim3(512),
                          lds, stream, C, (const bf16*)A, (const bf16*)Bt, M,
-                         N, K);
+                         N, K, g_st_m, g_st_n);
     else if (v == 5)
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256p32_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
@@ -732,7 +738,7 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
     else if (v == 6)
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true>), grid,
                          dim3(512), lds, stream, C, (const bf16*)A,
-                         (const bf16*)Bt, M, N, K);
+                         (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
