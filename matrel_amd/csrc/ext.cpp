@@ -10,6 +10,8 @@
 extern "C" {
 void launch_gemm_bt_bf16_acc(float*, const void*, const void*, int, int,
                              int, hipStream_t);
+void launch_gemm_bt_bf16_acc_b0(float*, const void*, const void*, int, int,
+                                int, hipStream_t);
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
 void set_gemm_variant(int);
@@ -79,7 +81,20 @@ bool is_bf16(const torch::Tensor& t) {
 
 // ---- GEMM: C(f32) += A @ Bt^T; shapes must be padded to the kernel tile
 // (M,N % 128, K % 64 for bf16 / % 32 for f32). Python side pads.
+void gemm_bt_acc_impl(torch::Tensor& C, torch::Tensor& A,
+                      torch::Tensor& Bt, bool beta0);
+
 void gemm_bt_acc(torch::Tensor C, torch::Tensor A, torch::Tensor Bt) {
+  gemm_bt_acc_impl(C, A, Bt, false);
+}
+
+void gemm_bt_acc_beta0(torch::Tensor C, torch::Tensor A, torch::Tensor Bt) {
+  // C may be UNINITIALIZED (torch.empty): the kernel overwrites it.
+  gemm_bt_acc_impl(C, A, Bt, true);
+}
+
+void gemm_bt_acc_impl(torch::Tensor& C, torch::Tensor& A, torch::Tensor& Bt,
+                      bool beta0) {
   check_2d(C, "C");
   check_2d(A, "A");
   check_2d(Bt, "Bt");
@@ -91,9 +106,14 @@ void gemm_bt_acc(torch::Tensor C, torch::Tensor A, torch::Tensor Bt) {
   if (is_bf16(A)) {
     TORCH_CHECK(is_bf16(Bt), "A/Bt dtype mismatch");
     TORCH_CHECK(K % 64 == 0, "K must be a multiple of 64 (bf16)");
-    launch_gemm_bt_bf16_acc(C.data_ptr<float>(), A.data_ptr(), Bt.data_ptr(),
-                            M, N, K, cur_stream());
+    if (beta0)
+      launch_gemm_bt_bf16_acc_b0(C.data_ptr<float>(), A.data_ptr(),
+                                 Bt.data_ptr(), M, N, K, cur_stream());
+    else
+      launch_gemm_bt_bf16_acc(C.data_ptr<float>(), A.data_ptr(),
+                              Bt.data_ptr(), M, N, K, cur_stream());
   } else {
+    if (beta0) C.zero_();
     TORCH_CHECK(A.scalar_type() == torch::kFloat32, "A must be bf16/fp32");
     TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32 (fp32)");
     launch_gemm_bt_f32_acc(C.data_ptr<float>(), A.data_ptr<float>(),
@@ -301,6 +321,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "matrel_amd hand-written CDNA4 (gfx950) kernels";
   m.def("gemm_bt_acc", &gemm_bt_acc,
         "C(f32) += A @ Bt^T (bf16 MFMA / exact-f32 MFMA)");
+  m.def("gemm_bt_acc_beta0", &gemm_bt_acc_beta0,
+        "C(f32) = A @ Bt^T (C may be uninitialized)");
   m.def("set_gemm_variant", [](int64_t v) { set_gemm_variant((int)v); },
         "0=auto 1=simple256 2=phased256 3=tile128 (A/B testing)");
   m.def("set_supertile", [](int64_t sm, int64_t sn) {

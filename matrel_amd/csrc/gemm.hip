@@ -369,7 +369,7 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
 // flight across every barrier). Quadrant snake order (0,0)->(0,1)->
 // (1,1)->(1,0) reuses one operand's fragments at each transition.
 
-template <bool MIDBAR, bool STATICPRIO = false>
+template <bool MIDBAR, bool STATICPRIO = false, bool BETA0 = false>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
@@ -526,16 +526,18 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
 #pragma unroll
   for (int q = 0; q < 4; ++q) {
     float cv[2][4][4];
+    if (!BETA0) {
 #pragma unroll
-    for (int m = 0; m < 2; ++m)
+      for (int m = 0; m < 2; ++m)
 #pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        long row = m0 + QMH[q] * 128 + crow0 + m * 16;
-        long col = n0 + QNH[q] * 128 + ccol0 + n * 16;
-        const float* cp = C + row * (long)N + col;
+        for (int n = 0; n < 4; ++n) {
+          long row = m0 + QMH[q] * 128 + crow0 + m * 16;
+          long col = n0 + QNH[q] * 128 + ccol0 + n * 16;
+          const float* cp = C + row * (long)N + col;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) cv[m][n][r] = cp[(long)r * N];
-      }
+          for (int r = 0; r < 4; ++r) cv[m][n][r] = cp[(long)r * N];
+        }
+    }
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -545,7 +547,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
         float* cp = C + row * (long)N + col;
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          cp[(long)r * N] = cv[m][n][r] + acc[q][m][n][r];
+          cp[(long)r * N] = (BETA0 ? 0.0f : cv[m][n][r]) + acc[q][m][n][r];
       }
   }
 }
@@ -709,6 +711,28 @@ static unsigned g_st_m = 4, g_st_n = 4;  // supertile cluster dims
 extern "C" void set_supertile(int sm, int sn) {
   g_st_m = (unsigned)sm;
   g_st_n = (unsigned)sn;
+}
+
+extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
+                                        const void* Bt, int M, int N, int K,
+                                        hipStream_t stream);
+
+extern "C" void launch_gemm_bt_bf16_acc_b0(float* C, const void* A,
+                                           const void* Bt, int M, int N,
+                                           int K, hipStream_t stream) {
+  // beta=0 fast path (first k-contribution of a C block): phased kernel,
+  // no C reads. Falls back to the generic path for small grids.
+  int nb = (M / BM2) * (N / BN2);
+  if (M % BM2 == 0 && N % BN2 == 0 && nb >= 192) {
+    size_t lds = 2u * BM2 * BKH * 2 * 2;
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n);
+    return;
+  }
+  hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream);
+  launch_gemm_bt_bf16_acc(C, A, Bt, M, N, K, stream);
 }
 
 extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,

@@ -62,16 +62,19 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 if ab is None:
                     continue
                 c = acc.get((bi, bj))
-                if c is None:
-                    c = torch.zeros(out.block_shape(bi, bj),
+                first = c is None
+                if first:
+                    c = torch.empty(out.block_shape(bi, bj),
                                     dtype=torch.float32, device=out.device)
                     acc[(bi, bj)] = c
                 if blockops.is_sparse(ab):
+                    if first:
+                        c.zero_()
                     blockops.spmm_acc(c, ab, bbd)
                 else:
                     if bt is None:
                         bt = _pre_t(bbd)
-                    blockops.gemm_bt_acc(c, ab, bt)
+                    blockops.gemm_bt_acc(c, ab, bt, beta0=first)
     for (bi, bj), c in acc.items():
         out.blocks[(bi, bj)] = c.to(out.dtype)
 
@@ -173,11 +176,12 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             bt = _pre_t(b_stack[n_j])
             for n_i, i in enumerate(a_rows):
                 c = acc.get((i, j))
-                if c is None:
-                    c = torch.zeros(bs, bs, dtype=torch.float32,
+                first = c is None
+                if first:
+                    c = torch.empty(bs, bs, dtype=torch.float32,
                                     device=out.device)
                     acc[(i, j)] = c
-                blockops.gemm_bt_acc(c, a_stack[n_i], bt)
+                blockops.gemm_bt_acc(c, a_stack[n_i], bt, beta0=first)
         cur = nxt
 
     for (i, j), c in acc.items():

@@ -94,11 +94,17 @@ def _pad2(t: torch.Tensor, rows: int, cols: int) -> torch.Tensor:
     return out
 
 
-def gemm_bt_acc(acc: torch.Tensor, a: torch.Tensor,
-                bt: torch.Tensor) -> None:
-    """acc(fp32, MxN) += a(MxK) @ bt(NxK)^T - the hot MFMA kernel."""
+def gemm_bt_acc(acc: torch.Tensor, a: torch.Tensor, bt: torch.Tensor,
+                beta0: bool = False) -> None:
+    """acc(fp32, MxN) += a(MxK) @ bt(NxK)^T - the hot MFMA kernel.
+    With beta0=True, acc is OVERWRITTEN (may be torch.empty): the kernel
+    skips the C-read pass (first k-contribution of a block)."""
     if not use_hip(a):
-        acc.add_(a.float() @ bt.float().t())
+        prod = a.float() @ bt.float().t()
+        if beta0:
+            acc.copy_(prod)
+        else:
+            acc.add_(prod)
         return
     ext = require_ext()
     if a.dtype != bt.dtype:
@@ -108,13 +114,19 @@ def gemm_bt_acc(acc: torch.Tensor, a: torch.Tensor,
     km = _kmult(a.dtype)
     mp, np_, kp = _ceil(m, _TILE_MN), _ceil(n, _TILE_MN), _ceil(k, km)
     if (mp, np_, kp) == (m, n, k):
-        ext.gemm_bt_acc(acc, a.contiguous(), bt.contiguous())
+        if beta0:
+            ext.gemm_bt_acc_beta0(acc, a.contiguous(), bt.contiguous())
+        else:
+            ext.gemm_bt_acc(acc, a.contiguous(), bt.contiguous())
         return
     ap = _pad2(a, mp, kp)
     btp = _pad2(bt, np_, kp)
-    cp = torch.zeros(mp, np_, dtype=torch.float32, device=a.device)
-    ext.gemm_bt_acc(cp, ap, btp)
-    acc.add_(cp[:m, :n])
+    cp = torch.empty(mp, np_, dtype=torch.float32, device=a.device)
+    ext.gemm_bt_acc_beta0(cp, ap, btp)
+    if beta0:
+        acc.copy_(cp[:m, :n])
+    else:
+        acc.add_(cp[:m, :n])
 
 
 def gemm_acc(acc: torch.Tensor, a: torch.Tensor, b: torch.Tensor) -> None:

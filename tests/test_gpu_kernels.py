@@ -260,3 +260,22 @@ def test_compile_hipgraph_replay(gpu_session):
     ref = A.to_tensor() @ B.to_tensor() + A.to_tensor()
     assert (out1 - ref).abs().max().item() < 1.0
     assert torch.equal(out1, out2)
+
+
+def test_gemm_beta0(ext):
+    """beta0 path overwrites uninitialized C (no C-read epilogue)."""
+    M = N = K = 512
+    a = _randn(M, K, seed=40)
+    bt = _randn(N, K, seed=41)
+    c = torch.full((M, N), 777.0, dtype=torch.float32, device="cuda")
+    ext.gemm_bt_acc_beta0(c, a, bt)
+    ref = a.float() @ bt.float().t()
+    assert (c - ref).abs().max().item() < 0.3
+    # large-grid phased beta0 path
+    M2 = N2 = 4096
+    a2 = _randn(M2, 256, seed=42)
+    bt2 = _randn(N2, 256, seed=43)
+    c2 = torch.full((M2, N2), -5.0, dtype=torch.float32, device="cuda")
+    ext.gemm_bt_acc_beta0(c2, a2, bt2)
+    ref2 = a2.float() @ bt2.float().t()
+    assert (c2 - ref2).abs().max().item() < 0.3
