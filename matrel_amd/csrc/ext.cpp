@@ -12,6 +12,8 @@ void launch_gemm_bt_bf16_acc(float*, const void*, const void*, int, int,
                              int, hipStream_t);
 void launch_gemm_bt_bf16_acc_b0(float*, const void*, const void*, int, int,
                                 int, hipStream_t);
+int launch_gemm_bt_bf16_final(float*, void*, const void*, const void*, int,
+                              int, int, int, hipStream_t);
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
 void set_gemm_variant(int);
@@ -323,6 +325,27 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "C(f32) += A @ Bt^T (bf16 MFMA / exact-f32 MFMA)");
   m.def("gemm_bt_acc_beta0", &gemm_bt_acc_beta0,
         "C(f32) = A @ Bt^T (C may be uninitialized)");
+  m.def("gemm_bt_final",
+        [](torch::Tensor C, torch::Tensor Co, torch::Tensor A,
+           torch::Tensor Bt, bool beta0) -> bool {
+          check_2d(C, "C");
+          check_2d(Co, "Co");
+          check_2d(A, "A");
+          check_2d(Bt, "Bt");
+          TORCH_CHECK(C.scalar_type() == torch::kFloat32 &&
+                      Co.scalar_type() == torch::kBFloat16 &&
+                      is_bf16(A) && is_bf16(Bt));
+          int M = (int)A.size(0), K = (int)A.size(1), N = (int)Bt.size(0);
+          TORCH_CHECK(Co.size(0) == M && Co.size(1) == N &&
+                      C.size(0) == M && C.size(1) == N);
+          TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0);
+          return launch_gemm_bt_bf16_final(
+                     C.data_ptr<float>(), Co.data_ptr(), A.data_ptr(),
+                     Bt.data_ptr(), M, N, K, beta0 ? 1 : 0,
+                     cur_stream()) != 0;
+        },
+        "last-k fused: bf16 out straight from the accumulate epilogue; "
+        "returns false when geometry unsupported (caller falls back)");
   m.def("set_gemm_variant", [](int64_t v) { set_gemm_variant((int)v); },
         "0=auto 1=simple256 2=phased256 3=tile128 (A/B testing)");
   m.def("set_supertile", [](int64_t sm, int64_t sn) {

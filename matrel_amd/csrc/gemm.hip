@@ -369,12 +369,13 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
 // flight across every barrier). Quadrant snake order (0,0)->(0,1)->
 // (1,1)->(1,0) reuses one operand's fragments at each transition.
 
-template <bool MIDBAR, bool STATICPRIO = false, bool BETA0 = false>
+template <bool MIDBAR, bool STATICPRIO = false, bool BETA0 = false,
+          bool BF16OUT = false>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
                         const bf16* __restrict__ Bt, int M, int N, int K,
-                        u32 stm_, u32 stn_) {
+                        u32 stm_, u32 stn_, bf16* __restrict__ Co = nullptr) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // layout: buf b (2) x half h (0=A0 1=A1 2=B0 3=B1) x 16 KiB; all LDS
   // addressing is 32-bit offsets from one base (keeps VGPR pressure down).
@@ -544,10 +545,14 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
       for (int n = 0; n < 4; ++n) {
         long row = m0 + QMH[q] * 128 + crow0 + m * 16;
         long col = n0 + QNH[q] * 128 + ccol0 + n * 16;
-        float* cp = C + row * (long)N + col;
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          cp[(long)r * N] = (BETA0 ? 0.0f : cv[m][n][r]) + acc[q][m][n][r];
+        for (int r = 0; r < 4; ++r) {
+          float v = (BETA0 ? 0.0f : cv[m][n][r]) + acc[q][m][n][r];
+          if (BF16OUT)
+            Co[(row + r) * (long)N + col] = (bf16)v;
+          else
+            C[(row + r) * (long)N + col] = v;
+        }
       }
   }
 }
@@ -717,6 +722,29 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
                                         const void* Bt, int M, int N, int K,
                                         hipStream_t stream);
 
+// last-k fused variant: read fp32 partials from C (unless beta0), write
+// the finished block STRAIGHT to bf16 Co (no fp32 store, no separate
+// conversion pass). Only for the phased geometry; caller guards.
+extern "C" int launch_gemm_bt_bf16_final(float* C, void* Co, const void* A,
+                                         const void* Bt, int M, int N,
+                                         int K, int beta0,
+                                         hipStream_t stream) {
+  int nb = (M / BM2) * (N / BN2);
+  if (!(M % BM2 == 0 && N % BN2 == 0 && nb >= 192)) return 0;
+  size_t lds = 2u * BM2 * BKH * 2 * 2;
+  if (beta0)
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true, true, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n, (bf16*)Co);
+  else
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true, false, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n, (bf16*)Co);
+  return 1;
+}
+
 extern "C" void launch_gemm_bt_bf16_acc_b0(float* C, const void* A,
                                            const void* Bt, int M, int N,
                                            int K, hipStream_t stream) {
@@ -731,7 +759,7 @@ extern "C" void launch_gemm_bt_bf16_acc_b0(float* C, const void* A,
                        g_st_n);
     return;
   }
-  hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream);
+  (void)hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream);
   launch_gemm_bt_bf16_acc(C, A, Bt, M, N, K, stream);
 }
 

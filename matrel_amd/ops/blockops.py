@@ -129,6 +129,25 @@ def gemm_bt_acc(acc: torch.Tensor, a: torch.Tensor, bt: torch.Tensor,
         acc.add_(cp[:m, :n])
 
 
+def gemm_bt_final(acc: torch.Tensor, out_bf16: torch.Tensor,
+                  a: torch.Tensor, bt: torch.Tensor,
+                  beta0: bool = False) -> bool:
+    """Fused last-k call: (acc-partials [+ a @ bt^T]) written STRAIGHT to
+    bf16 `out_bf16` (no fp32 store, no conversion pass). Returns False
+    when the geometry/dtype is unsupported - caller falls back to
+    gemm_bt_acc + .to(bf16)."""
+    if not use_hip(a) or a.dtype != torch.bfloat16 \
+            or bt.dtype != torch.bfloat16 \
+            or out_bf16.dtype != torch.bfloat16:
+        return False
+    m, k = a.shape
+    n = bt.shape[0]
+    if m % 128 or n % 128 or k % 64:
+        return False
+    return bool(require_ext().gemm_bt_final(
+        acc, out_bf16, a.contiguous(), bt.contiguous(), beta0))
+
+
 def gemm_acc(acc: torch.Tensor, a: torch.Tensor, b: torch.Tensor) -> None:
     """acc(fp32) += a @ b (transposes b internally on GPU; prefer
     gemm_bt_acc + a cached transpose when b is reused)."""
