@@ -150,30 +150,32 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         if k in ka:
             a_src = rt.rank_of(p, k % pc)
             a_rows = [i for i in my_is if pa[i, k]]
-            a_stack = torch.zeros(len(a_rows), bs, bs, dtype=a.dtype,
+            # empty, not zeros: the broadcast overwrites it on receivers,
+            # and the source zero-pads only ragged edge blocks
+            a_stack = torch.empty(len(a_rows), bs, bs, dtype=a.dtype,
                                   device=a.device)
             if rt.rank == a_src:
                 for n, i in enumerate(a_rows):
-                    t = a.get_block(i, k)
-                    if t is not None:
-                        t = blockops.densify(t)
-                        h, w = t.shape
-                        a_stack[n, :h, :w] = t
+                    t = blockops.densify(a.get_block_dense(i, k))
+                    h, w = t.shape
+                    if (h, w) != (bs, bs):
+                        a_stack[n].zero_()
+                    a_stack[n, :h, :w] = t
             wa = dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
                                 async_op=True)
         b_cols, b_stack, wb = [], None, None
         if k in kb:
             b_src = rt.rank_of(k % pr, q)
             b_cols = [j for j in my_js if pb[k, j]]
-            b_stack = torch.zeros(len(b_cols), bs, bs, dtype=b.dtype,
+            b_stack = torch.empty(len(b_cols), bs, bs, dtype=b.dtype,
                                   device=b.device)
             if rt.rank == b_src:
                 for n, j in enumerate(b_cols):
-                    t = b.get_block(k, j)
-                    if t is not None:
-                        t = blockops.densify(t)
-                        h, w = t.shape
-                        b_stack[n, :h, :w] = t
+                    t = blockops.densify(b.get_block_dense(k, j))
+                    h, w = t.shape
+                    if (h, w) != (bs, bs):
+                        b_stack[n].zero_()
+                    b_stack[n, :h, :w] = t
             wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
                                 async_op=True)
         return a_rows, a_stack, b_cols, b_stack, wa, wb

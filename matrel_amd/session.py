@@ -82,4 +82,7 @@ class MatrelSession:
         metrics.reset()
 
     def shutdown(self) -> None:
+        from .exec import tracing
+        if self.config.trace:
+            tracing.dump_trace()
         self.runtime.shutdown()
