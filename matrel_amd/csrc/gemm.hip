@@ -791,6 +791,10 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true>), grid,
                          dim3(512), lds, stream, C, (const bf16*)A,
                          (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
+    else if (v == 7)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<true, true>), grid,
+                         dim3(512), lds, stream, C, (const bf16*)A,
+                         (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,

@@ -28,7 +28,7 @@ def main():
         a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
         bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
         c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
-        for v, name in ((1, "simple256"), (4, "phased-nb"), (6, "phased-nb-sprio"), (3, "tile128")):
+        for v, name in ((1, "simple256"), (4, "phased-nb"), (6, "phased-nb-sprio"), (7, "phased-mb-sprio"), (3, "tile128")):
             ext.set_gemm_variant(v)
             dt = timeit(lambda: ext.gemm_bt_acc(c, a, bt))
             print(f"gemm {name} {n}^3: {2*n**3/dt/1e12:.0f} TFLOPS")
