@@ -38,6 +38,9 @@ def main():
     ap.add_argument("--n-size", type=int, default=0,
                     help="override matrix side N")
     ap.add_argument("--block-size", type=int, default=16384)
+    ap.add_argument("--precision", choices=["bf16", "mx8"], default="bf16",
+                    help="GEMM compute precision (mx8 = MX-scaled fp8; "
+                         "NOT the headline config - BASELINE is bf16)")
     args = ap.parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -46,7 +49,8 @@ def main():
     import matrel_amd as ma
 
     block = args.block_size if has_gpu else 512
-    ma.set_config(ma.MatrelConfig(block_size=block))
+    ma.set_config(ma.MatrelConfig(block_size=block,
+                                  gemm_precision=args.precision))
     sess = ma.MatrelSession()
     rt = sess.runtime
     world = rt.world_size
@@ -103,7 +107,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if has_gpu else "fp32",
+            "dtype": ("fp8_mx" if args.precision == "mx8" else "bf16")
+                     if has_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": "NxN distributed dense matmul (BASELINE configs 2/3)",

@@ -14,6 +14,9 @@ void launch_gemm_bt_bf16_acc_b0(float*, const void*, const void*, int, int,
                                 int, hipStream_t);
 int launch_gemm_bt_bf16_final(float*, void*, const void*, const void*, int,
                               int, int, int, hipStream_t);
+void launch_gemm_bt_mx8_acc(float*, const void*, const void*, const void*,
+                            const void*, int, int, int, int, hipStream_t);
+void launch_mx8_convert(void*, void*, const void*, long, long, hipStream_t);
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
 void set_gemm_variant(int);
@@ -305,6 +308,40 @@ void spmm_csr_acc(torch::Tensor C, torch::Tensor crow, torch::Tensor cols,
   }
 }
 
+// bf16 -> (e4m3 bytes, e8m0 scales [rows][cols/32])
+std::vector<torch::Tensor> mx8_convert(torch::Tensor a) {
+  check_2d(a, "a");
+  TORCH_CHECK(is_bf16(a), "mx8_convert needs bf16");
+  long rows = a.size(0), cols = a.size(1);
+  TORCH_CHECK(cols % 32 == 0, "cols must be a multiple of 32");
+  auto opts = torch::TensorOptions().dtype(torch::kUInt8).device(a.device());
+  auto out8 = torch::empty({rows, cols}, opts);
+  auto sc = torch::empty({rows, cols / 32}, opts);
+  launch_mx8_convert(out8.data_ptr(), sc.data_ptr(), a.data_ptr(), rows,
+                     cols, cur_stream());
+  return {out8, sc};
+}
+
+// C(f32) += A8 @ Bt8^T with per-32-block e8m0 scales
+void gemm_bt_mx8_acc(torch::Tensor C, torch::Tensor A8, torch::Tensor Asc,
+                     torch::Tensor Bt8, torch::Tensor Bsc, bool beta0) {
+  check_2d(C, "C");
+  check_2d(A8, "A8");
+  check_2d(Bt8, "Bt8");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(A8.scalar_type() == torch::kUInt8 &&
+              Bt8.scalar_type() == torch::kUInt8);
+  int M = (int)A8.size(0), K = (int)A8.size(1), N = (int)Bt8.size(0);
+  TORCH_CHECK(Bt8.size(1) == K && C.size(0) == M && C.size(1) == N);
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "mx8 GEMM needs M,N % 256 == 0, K % 64 == 0");
+  TORCH_CHECK(Asc.numel() == (long)M * (K / 32) &&
+              Bsc.numel() == (long)N * (K / 32));
+  launch_gemm_bt_mx8_acc(C.data_ptr<float>(), A8.data_ptr(), Asc.data_ptr(),
+                         Bt8.data_ptr(), Bsc.data_ptr(), M, N, K,
+                         beta0 ? 1 : 0, cur_stream());
+}
+
 torch::Tensor transpose(torch::Tensor a) {
   check_2d(a, "a");
   auto out = torch::empty({a.size(1), a.size(0)}, a.options());
@@ -361,4 +398,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("trace", &trace, "diagonal sum");
   m.def("transpose", &transpose, "materialized transpose");
   m.def("spmm_csr_acc", &spmm_csr_acc, "C(f32) += CSR @ B (within-block sparse)");
+  m.def("mx8_convert", &mx8_convert, "bf16 -> (e4m3, e8m0 block scales)");
+  m.def("gemm_bt_mx8_acc", &gemm_bt_mx8_acc,
+        "C(f32) += A8 @ Bt8^T (MX-fp8 scaled MFMA, ~2x bf16 rate)");
 }

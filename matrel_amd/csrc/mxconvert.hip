@@ -1,0 +1,81 @@
+// bf16 -> MX-fp8 conversion: per (row, 32-wide k-block) compute the e8m0
+// shared scale (2^(floor(log2 maxabs) - 7): normalized max in [128,256),
+// inside e4m3's 448 range) and quantize the 32 values to e4m3 (RNE).
+// Feeds csrc/gemm_mx8.hip.
+#include "common.h"
+
+typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ unsigned char f32_to_e4m3(float f) {
+  // OCP e4m3fn, round-to-nearest-even, saturate to +-448
+  unsigned s = (__float_as_uint(f) >> 31) << 7;
+  float a = fabsf(f);
+  if (a != a) return (unsigned char)(s | 0x7F);  // NaN
+  if (a >= 448.0f) return (unsigned char)(s | 0x7E);  // saturate to 448
+  if (a < 0.87890625e-2f * 0.125f) {  // < 2^-10 -> subnormal/zero region
+    // subnormals: m/8 * 2^-6
+    int m = (int)rintf(a * 512.0f);  // a / 2^-9
+    if (m > 7) m = 7;
+    return (unsigned char)(s | m);
+  }
+  int e = (int)floorf(log2f(a));
+  float mant = a / exp2f((float)e) - 1.0f;  // [0,1)
+  int m = (int)rintf(mant * 8.0f);
+  if (m == 8) { m = 0; e += 1; }
+  int be = e + 7;
+  if (be < 1) {  // subnormal
+    int ms = (int)rintf(a * 512.0f);
+    if (ms > 7) ms = 7;
+    return (unsigned char)(s | ms);
+  }
+  if (be > 15) return (unsigned char)(s | 0x7E);
+  return (unsigned char)(s | (be << 3) | m);
+}
+
+// one thread per (row, kblock): 32 bf16 in, 32 e4m3 + 1 e8m0 out
+__global__ void mx8_convert_k(unsigned char* __restrict__ out8,
+                              unsigned char* __restrict__ sc,
+                              const bf16* __restrict__ in, long rows,
+                              long cols) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long kb = cols / 32;
+  if (idx >= rows * kb) return;
+  long r = idx / kb, b = idx - r * kb;
+  const bf16* src = in + r * cols + b * 32;
+  float v[32];
+  float mx = 0.0f;
+#pragma unroll
+  for (int i = 0; i < 32; i += 8) {
+    bf16x8v x = *(const bf16x8v*)(src + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)x[j];
+      v[i + j] = f;
+      float a = fabsf(f);
+      if (a > mx) mx = a;
+    }
+  }
+  int e;
+  if (mx == 0.0f) {
+    e = 0;
+    sc[idx] = 127;  // scale 1.0; all-zero payload
+  } else {
+    e = (int)floorf(log2f(mx)) - 7;  // normalized max in [128, 256)
+    if (e < -127) e = -127;
+    if (e > 127) e = 127;
+    sc[idx] = (unsigned char)(e + 127);
+  }
+  float inv = exp2f((float)-e);
+  unsigned char* dst = out8 + r * cols + b * 32;
+#pragma unroll
+  for (int i = 0; i < 32; ++i) dst[i] = f32_to_e4m3(v[i] * inv);
+}
+
+extern "C" void launch_mx8_convert(void* out8, void* sc, const void* in,
+                                   long rows, long cols, hipStream_t s) {
+  long n = rows * (cols / 32);
+  long blocks = (n + 255) / 256;
+  hipLaunchKernelGGL(mx8_convert_k, dim3((unsigned)blocks), dim3(256), 0, s,
+                     (unsigned char*)out8, (unsigned char*)sc,
+                     (const bf16*)in, rows, cols);
+}

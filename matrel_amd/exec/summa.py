@@ -24,6 +24,7 @@ import torch
 import torch.distributed as dist
 
 from ..blocks.matrix import BlockMatrix
+from ..config import get_config
 from ..ops import blockops
 from .runtime import get_runtime
 
@@ -56,6 +57,25 @@ def _bt_cached(b: BlockMatrix, k: int, bj: int,
     return bt
 
 
+def _mx8_cached(m: BlockMatrix, key, tensor_bf16: torch.Tensor):
+    """Per-matrix cache of MX-fp8 conversions (fp8 payload is half the
+    bf16 bytes; reused across every multiply touching the block)."""
+    cache = getattr(m, "_mx8_cache", None)
+    if cache is None:
+        cache = m._mx8_cache = {}
+    ent = cache.get(key)
+    if ent is None:
+        ent = blockops.to_mx8(tensor_bf16)
+        cache[key] = ent
+    return ent
+
+
+def _mx8_usable(t: torch.Tensor) -> bool:
+    return (t.is_cuda and t.dtype == torch.bfloat16
+            and t.shape[0] % 256 == 0 and t.shape[1] % 64 == 0
+            and t.shape[1] % 32 == 0)
+
+
 def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     """Pure-local block multiply: every needed A/B block is already on the
     rank that owns the corresponding C block (single process, or
@@ -63,6 +83,7 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     Loops (j, k) outer so each B block is transposed once for the B^T MFMA
     kernel and reused across the block-row loop."""
     fused_ok = (out.device.type == "cuda" and out.dtype == torch.bfloat16)
+    mx8 = get_config().gemm_precision == "mx8"
     for (bi, bj) in out.owned_indices():
         ks = [k for k in range(a.grid_cols)
               if (bi, k) in a.blocks and (k, bj) in b.blocks]
@@ -86,6 +107,13 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 blockops.spmm_acc(c, ab, bbd)
                 continue
             bt = _bt_cached(b, k, bj, bbd)
+            if mx8 and _mx8_usable(ab) and _mx8_usable(bt) \
+                    and ab.shape[0] % 256 == 0 and bt.shape[0] % 256 == 0:
+                a8, asc = _mx8_cached(a, (bi, k), ab)
+                bt8, btsc = _mx8_cached(b, ("t", k, bj), bt)
+                blockops.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc,
+                                         beta0=first)
+                continue
             if use_fused and last:
                 fused_done = blockops.gemm_bt_final(c, outblk, ab, bt,
                                                     beta0=first)

@@ -288,3 +288,16 @@ def mask_isin(a: torch.Tensor, needles_sorted: torch.Tensor) -> torch.Tensor:
     keep = torch.isin(a.float(), needles_sorted.float())
     return torch.where(keep, a, torch.zeros((), dtype=a.dtype,
                                             device=a.device))
+
+
+def to_mx8(t_bf16: torch.Tensor):
+    """bf16 -> (e4m3 bytes, e8m0 block scales [rows][cols/32])."""
+    return tuple(require_ext().mx8_convert(t_bf16.contiguous()))
+
+
+def gemm_bt_mx8_acc(acc: torch.Tensor, a8, asc, bt8, btsc,
+                    beta0: bool = False) -> None:
+    """acc(fp32) += dequant(a8,asc) @ dequant(bt8,btsc)^T via the MX-fp8
+    scaled MFMA (~2x the bf16 rate). Shapes must satisfy M,N % 256 == 0,
+    K % 64 == 0 (callers guard and fall back to bf16)."""
+    require_ext().gemm_bt_mx8_acc(acc, a8, asc, bt8, btsc, beta0)

@@ -279,3 +279,60 @@ def test_gemm_beta0(ext):
     ext.gemm_bt_acc_beta0(c2, a2, bt2)
     ref2 = a2.float() @ bt2.float().t()
     assert (c2 - ref2).abs().max().item() < 0.3
+
+
+def test_mx8_convert_roundtrip(ext):
+    """bf16 -> MX-fp8: dequantized values within e4m3 relative error."""
+    a = _randn(256, 256, seed=50)
+    a8, sc = ext.mx8_convert(a)
+    assert a8.shape == (256, 256) and sc.shape == (256, 8)
+    # dequantize on host
+    import torch as T
+    bits = a8.cpu().to(T.int32)
+    s = bits >> 7
+    be = (bits >> 3) & 0xF
+    m = bits & 7
+    val = T.where(be > 0, (2.0 ** (be - 7)) * (1 + m / 8.0),
+                  (2.0 ** -6) * (m / 8.0))
+    val = T.where(s > 0, -val, val)
+    scale = 2.0 ** (sc.cpu().to(T.int32).float() - 127)
+    deq = val.float() * scale.repeat_interleave(32, dim=1)
+    ref = a.float().cpu()
+    err = (deq - ref).abs()
+    tol = ref.abs() * 0.08 + 1e-3
+    assert (err <= tol).all(), f"max rel {((err/(ref.abs()+1e-6)).max())}"
+
+
+def test_mx8_gemm_vs_fp32(ext):
+    M = N = 512
+    K = 512
+    a = _randn(M, K, seed=51)
+    bt = _randn(N, K, seed=52)
+    a8, asc = ext.mx8_convert(a)
+    bt8, btsc = ext.mx8_convert(bt)
+    c = torch.full((M, N), 3.0, dtype=torch.float32, device="cuda")
+    ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True)  # beta0 overwrite
+    ref = a.float() @ bt.float().t()
+    rel = ((c - ref).abs() / (ref.abs() + 1.0)).max().item()
+    assert rel < 0.05, f"mx8 rel err {rel}"
+    # accumulate semantics
+    ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, False)
+    rel2 = ((c - 2 * ref).abs() / (2 * ref.abs() + 1.0)).max().item()
+    assert rel2 < 0.05, f"mx8 acc rel err {rel2}"
+
+
+def test_mx8_end_to_end(gpu_session):
+    import matrel_amd as ma
+    cfg = ma.get_config()
+    old = cfg.gemm_precision
+    cfg.gemm_precision = "mx8"
+    try:
+        s = gpu_session
+        A = s.randn(2048, 2048, seed=53)
+        B = s.randn(2048, 2048, seed=54)
+        C = (A @ B).compute()
+        ref = A.to_tensor() @ B.to_tensor()
+        rel = ((C.to_tensor() - ref).abs() / (ref.abs() + 1.0)).max().item()
+        assert rel < 0.06, f"mx8 e2e rel err {rel}"
+    finally:
+        cfg.gemm_precision = old
