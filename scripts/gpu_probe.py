@@ -44,6 +44,18 @@ def main():
             print(f"  variant {v} vs simple max|diff| = {err}")
         del a, bt, c, c1, c2
     ext.set_gemm_variant(0)
+    # MX-fp8 scaled-MFMA GEMM
+    for n in (8192, 16384):
+        a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
+        bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
+        a8, asc = ext.mx8_convert(a)
+        bt8, btsc = ext.mx8_convert(bt)
+        del a, bt
+        c = torch.empty(n, n, dtype=torch.float32, device="cuda")
+        dt = timeit(lambda: ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True),
+                    iters=6, warmup=2)
+        print(f"gemm mx8 {n}^3: {2*n**3/dt/1e12:.0f} TFLOPS")
+        del a8, asc, bt8, btsc, c
     for n in (2048, 4096, 8192):
         a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
         bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)

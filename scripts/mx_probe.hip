@@ -18,8 +18,19 @@ __global__ void probe_k(const unsigned char* __restrict__ A,
   v8i a = ((const v8i*)A)[lane];
   v8i b = ((const v8i*)B)[lane];
   f32x16 c = {};
-  c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 0, sa,
-                                                      0, sb);
+  // mode>=1000: per-lane scale probes. 1000+x: sa = 127 + (lane&1),
+  // sb=127 -> which C rows double tells us which lanes feed which rows.
+  int sa_eff = sa, sb_eff = sb;
+  if (sa >= 1000) {
+    int mode = sa - 1000;
+    if (mode == 0) sa_eff = 127 + (lane & 1);          // by lane parity
+    else if (mode == 1) sa_eff = 127 + ((lane >> 5) & 1);  // by k-half
+    else if (mode == 2) sa_eff = (lane == 3) ? 128 : 127;  // single lane
+    else if (mode == 1000) sa_eff = (lane == 35) ? 128 : 127;
+    else sa_eff = 127;
+  }
+  c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 0,
+                                                      sa_eff, 0, sb_eff);
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
@@ -195,11 +206,46 @@ int main() {
   run(127, 127);
   printf("2.0s: C[0][0]=%g (want 128)\n", hC[0]);
 
+  // per-lane scale probes (A=B=1.0)
+  memset(hA, f2e4m3(1.0f), sizeof(hA));
+  memset(hB, f2e4m3(1.0f), sizeof(hB));
+  run(1000, 127);  // sa = 127 + (lane&1)
+  printf("perlane parity: C[0][0]=%g C[1][0]=%g C[2][0]=%g C[3][0]=%g\n",
+         hC[0], hC[32], hC[64], hC[96]);
+  run(1001, 127);  // sa = 127 + k-half
+  printf("perlane khalf: C[r][c] all should be 96: C[0][0]=%g C[5][9]=%g\n",
+         hC[0], hC[5 * 32 + 9]);
+  run(1002, 127);  // only lane 3 -> rows/k it feeds doubled
+  printf("single lane3 x2:\n");
+  for (int i = 0; i < 32; ++i)
+    if (hC[i * 32] != 64.0f) printf("  C[%d][0]=%g\n", i, hC[i * 32]);
+
+  // DECISIVE k-half pairing probe: payload only in k-half 0 (lanes<32),
+  // scale doubled only on lane 3 (a k-half-0 lane). If lane-l scale pairs
+  // with lane-l data (same k-half): C[3][0] = 64; if the halves are
+  // swapped: C[3][0] = 32 (the doubled scale lands on the zero half).
+  memset(hA, 0, sizeof(hA));
+  memset(hB, 0, sizeof(hB));
+  for (int l = 0; l < 32; ++l)
+    for (int kk = 0; kk < 32; ++kk) {
+      hA[l * 32 + kk] = f2e4m3(1.0f);
+      hB[l * 32 + kk] = f2e4m3(1.0f);
+    }
+  run(1002, 127);
+  printf("khalf pairing: C[3][0]=%g (64=same-lane, 32=swapped), "
+         "C[4][0]=%g (ctrl, want 32)\n", hC[3 * 32], hC[4 * 32]);
+  // and the mirror: scale doubled on lane 35 (k-half-1 lane), payload
+  // still only in half 0: same-lane -> no effect (32); swapped -> 64.
+  // reuse mode: sa>=2000 -> lane==35 doubled
+  run(2000, 127);
+  printf("khalf mirror: C[3][0]=%g (32=same-lane, 64=swapped)\n",
+         hC[3 * 32]);
+
   // ---- decode-table sweep: A filled with byte v, B = 1.0 (0x38);
   // hardware value of byte v = C[0][0] / 64.
   memset(hB, 0x38, sizeof(hB));
   printf("decode table (byte: hw_value | e4m3_value):\n");
-  for (int v = 0; v < 0x80; v += 4) {
+  for (int v = 0x30; v < 0x48; v += 1) {
     memset(hA, v, sizeof(hA));
     run(120, 120);
     float e4 = 0.f;
@@ -209,6 +255,34 @@ int main() {
               : exp2f(-6.f) * (m / 8.0f);
     }
     printf("  0x%02x: %g | %g\n", v, hC[0] / 64.0f, e4);
+  }
+
+  // random payload through the bare MFMA vs host dequant model
+  srand(123);
+  for (int i = 0; i < (int)sizeof(hA); ++i) hA[i] = rand() & 0x7F;
+  for (int i = 0; i < (int)sizeof(hB); ++i) hB[i] = rand() & 0x7F;
+  run(127, 127);
+  {
+    double maxerr = 0;
+    for (int i = 0; i < 32; ++i)
+      for (int j = 0; j < 32; ++j) {
+        double ref = 0;
+        for (int kk = 0; kk < 64; ++kk) {
+          int la = (kk >= 32 ? 32 : 0) + i, ka = kk & 31;
+          int lb = (kk >= 32 ? 32 : 0) + j, kb2 = kk & 31;
+          auto dec = [](unsigned char v) -> double {
+            int s = v >> 7, be = (v >> 3) & 0xF, m = v & 7;
+            double val = be ? exp2((double)(be - 7)) * (1 + m / 8.0)
+                            : exp2(-6.0) * (m / 8.0);
+            return s ? -val : val;
+          };
+          ref += dec(hA[la * 32 + ka]) * dec(hB[lb * 32 + kb2]);
+        }
+        double e = fabs(hC[i * 32 + j] - ref);
+        if (e > maxerr) maxerr = e;
+      }
+    printf("bare-mfma random-payload vs dequant model: max err %.6f %s\n",
+           maxerr, maxerr < 1e-2 ? "PASS" : "FAIL");
   }
   return 0;
 }

@@ -313,12 +313,17 @@ def test_mx8_gemm_vs_fp32(ext):
     c = torch.full((M, N), 3.0, dtype=torch.float32, device="cuda")
     ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True)  # beta0 overwrite
     ref = a.float() @ bt.float().t()
-    rel = ((c - ref).abs() / (ref.abs() + 1.0)).max().item()
-    assert rel < 0.05, f"mx8 rel err {rel}"
+    # fp8 e4m3 quantization noise: per-term sigma ~3.6%; for N(0,1)
+    # operands the output noise sigma ~ 0.05*sqrt(K); bound at ~6 sigma
+    # for the max over M*N entries, plus a mean-relative sanity check.
+    sigma = 0.05 * math.sqrt(K)
+    err = (c - ref).abs()
+    assert err.max().item() < 6 * sigma, f"mx8 max err {err.max().item()}"
+    assert (err.mean() / ref.abs().mean()).item() < 0.05, "mx8 mean rel"
     # accumulate semantics
     ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, False)
-    rel2 = ((c - 2 * ref).abs() / (2 * ref.abs() + 1.0)).max().item()
-    assert rel2 < 0.05, f"mx8 acc rel err {rel2}"
+    err2 = (c - 2 * ref).abs()
+    assert err2.max().item() < 12 * sigma, "mx8 acc err"
 
 
 def test_mx8_end_to_end(gpu_session):
@@ -332,7 +337,9 @@ def test_mx8_end_to_end(gpu_session):
         B = s.randn(2048, 2048, seed=54)
         C = (A @ B).compute()
         ref = A.to_tensor() @ B.to_tensor()
-        rel = ((C.to_tensor() - ref).abs() / (ref.abs() + 1.0)).max().item()
-        assert rel < 0.06, f"mx8 e2e rel err {rel}"
+        sigma = 0.05 * math.sqrt(2048)
+        err = (C.to_tensor() - ref).abs()
+        assert err.max().item() < 6 * sigma, f"mx8 e2e {err.max().item()}"
+        assert (err.mean() / ref.abs().mean()).item() < 0.05
     finally:
         cfg.gemm_precision = old
