@@ -316,7 +316,7 @@ std::vector<torch::Tensor> mx8_convert(torch::Tensor a) {
   TORCH_CHECK(cols % 32 == 0, "cols must be a multiple of 32");
   auto opts = torch::TensorOptions().dtype(torch::kUInt8).device(a.device());
   auto out8 = torch::empty({rows, cols}, opts);
-  auto sc = torch::empty({rows, cols / 32}, opts);
+  auto sc = torch::empty({cols / 32, rows}, opts);  // block-major
   launch_mx8_convert(out8.data_ptr(), sc.data_ptr(), a.data_ptr(), rows,
                      cols, cur_stream());
   return {out8, sc};
@@ -337,6 +337,8 @@ void gemm_bt_mx8_acc(torch::Tensor C, torch::Tensor A8, torch::Tensor Asc,
               "mx8 GEMM needs M,N % 256 == 0, K % 64 == 0");
   TORCH_CHECK(Asc.numel() == (long)M * (K / 32) &&
               Bsc.numel() == (long)N * (K / 32));
+  TORCH_CHECK(Asc.size(0) == K / 32 && Bsc.size(0) == K / 32,
+              "scales must be block-major [K/32][rows]");
   launch_gemm_bt_mx8_acc(C.data_ptr<float>(), A8.data_ptr(), Asc.data_ptr(),
                          Bt8.data_ptr(), Bsc.data_ptr(), M, N, K,
                          beta0 ? 1 : 0, cur_stream());

@@ -113,7 +113,7 @@ gemm_bt_mx8_acc_k(float* __restrict__ C, const unsigned char* __restrict__ A,
       af[m][5] = *(const int*)(p2 + 4);
       af[m][6] = *(const int*)(p2 + 8);
       af[m][7] = *(const int*)(p2 + 12);
-      sa[m] = Asc[(m0 + row) * (long)KB + kblk];
+      sa[m] = Asc[(long)kblk * M + m0 + row];
     }
     v8i bf[2];
     int sb[2];
@@ -132,7 +132,7 @@ gemm_bt_mx8_acc_k(float* __restrict__ C, const unsigned char* __restrict__ A,
       bf[n][5] = *(const int*)(p2 + 4);
       bf[n][6] = *(const int*)(p2 + 8);
       bf[n][7] = *(const int*)(p2 + 12);
-      sb[n] = Bsc[(n0 + row) * (long)KB + kblk];
+      sb[n] = Bsc[(long)kblk * N + n0 + row];
     }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -241,13 +241,23 @@ gemm_bt_mx8_acc_k128(float* __restrict__ C,
     const unsigned char* a_base = As + cur * MXBM * 128;
     const unsigned char* b_base = Bs + cur * MXBM * 128;
     // one dword of scales per fragment row: blocks [4t, 4t+4)
-    int sa_w[4], sb_w[2];
+    int sa_w[4], sb_w[2];  // byte b of the word = scale of block 4t+b
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
-      sa_w[m] = *(const int*)(Asc + (m0 + arow + m * 32) * (long)KB + t * 4);
+    for (int m = 0; m < 4; ++m) {
+      long r = m0 + arow + m * 32;
+      sa_w[m] = (int)Asc[(long)(t * 4 + 0) * M + r] |
+                ((int)Asc[(long)(t * 4 + 1) * M + r] << 8) |
+                ((int)Asc[(long)(t * 4 + 2) * M + r] << 16) |
+                ((int)Asc[(long)(t * 4 + 3) * M + r] << 24);
+    }
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
-      sb_w[n] = *(const int*)(Bsc + (n0 + brow + n * 32) * (long)KB + t * 4);
+    for (int n = 0; n < 2; ++n) {
+      long r = n0 + brow + n * 32;
+      sb_w[n] = (int)Bsc[(long)(t * 4 + 0) * N + r] |
+                ((int)Bsc[(long)(t * 4 + 1) * N + r] << 8) |
+                ((int)Bsc[(long)(t * 4 + 2) * N + r] << 16) |
+                ((int)Bsc[(long)(t * 4 + 3) * N + r] << 24);
+    }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       v8i af[4], bf[2];

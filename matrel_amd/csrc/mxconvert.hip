@@ -56,14 +56,18 @@ __global__ void mx8_convert_k(unsigned char* __restrict__ out8,
     }
   }
   int e;
+  // scales stored BLOCK-MAJOR: sc[block][row] (contiguous rows per
+  // 32-wide k-block) so GEMM kernels can stage a tile's scales with
+  // global_load_lds
+  long sidx = b * rows + r;
   if (mx == 0.0f) {
     e = 0;
-    sc[idx] = 127;  // scale 1.0; all-zero payload
+    sc[sidx] = 127;  // scale 1.0; all-zero payload
   } else {
     e = (int)floorf(log2f(mx)) - 7;  // normalized max in [128, 256)
     if (e < -127) e = -127;
     if (e > 127) e = 127;
-    sc[idx] = (unsigned char)(e + 127);
+    sc[sidx] = (unsigned char)(e + 127);
   }
   float inv = exp2f((float)-e);
   unsigned char* dst = out8 + r * cols + b * 32;

@@ -45,7 +45,7 @@ def dequant(a8, sc):
     val = torch.where(be > 0, (2.0 ** (be - 7)) * (1 + m / 8.0),
                       (2.0 ** -6) * (m / 8.0))
     val = torch.where(s > 0, -val, val)
-    scale = 2.0 ** (sc.to(torch.int32).float() - 127)
+    scale = 2.0 ** (sc.to(torch.int32).float().t() - 127)
     return val.float() * scale.repeat_interleave(32, dim=1)
 
 

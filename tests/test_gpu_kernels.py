@@ -285,7 +285,7 @@ def test_mx8_convert_roundtrip(ext):
     """bf16 -> MX-fp8: dequantized values within e4m3 relative error."""
     a = _randn(256, 256, seed=50)
     a8, sc = ext.mx8_convert(a)
-    assert a8.shape == (256, 256) and sc.shape == (256, 8)
+    assert a8.shape == (256, 256) and sc.shape == (8, 256)  # block-major
     # dequantize on host
     import torch as T
     bits = a8.cpu().to(T.int32)
@@ -295,7 +295,7 @@ def test_mx8_convert_roundtrip(ext):
     val = T.where(be > 0, (2.0 ** (be - 7)) * (1 + m / 8.0),
                   (2.0 ** -6) * (m / 8.0))
     val = T.where(s > 0, -val, val)
-    scale = 2.0 ** (sc.cpu().to(T.int32).float() - 127)
+    scale = 2.0 ** (sc.cpu().to(T.int32).float().t() - 127)
     deq = val.float() * scale.repeat_interleave(32, dim=1)
     ref = a.float().cpu()
     err = (deq - ref).abs()
