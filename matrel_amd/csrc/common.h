@@ -21,6 +21,12 @@ __device__ __forceinline__ void glds16(const void* gptr, void* lds_ptr) {
       (__attribute__((address_space(3))) u32*)lds_ptr, 16, 0, 0);
 }
 
+__device__ __forceinline__ void glds4(const void* gptr, void* lds_ptr) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) u32*)gptr,
+      (__attribute__((address_space(3))) u32*)lds_ptr, 4, 0, 0);
+}
+
 // ---- XCD-aware block swizzle (bijective form; cdna_hip_programming.md §5):
 // consecutive output tiles land on the same XCD's L2 so neighbouring tiles
 // re-hit shared operand panels.

@@ -24,7 +24,17 @@
 #include "common.h"
 
 typedef int v8i __attribute__((ext_vector_type(8)));
+typedef int v4i __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+__device__ __forceinline__ v8i v8_from(const void* p_lo, const void* p_hi) {
+  v4i lo = *(const v4i*)p_lo;   // one ds_read_b128
+  v4i hi = *(const v4i*)p_hi;   // one ds_read_b128
+  v8i r;
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+  r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+  return r;
+}
 
 #define MXBM 256
 #define MXBN 256
@@ -101,18 +111,10 @@ gemm_bt_mx8_acc_k(float* __restrict__ C, const unsigned char* __restrict__ A,
       // bytes 0-15 <- orig k [kh*16, +16): 16-B slot index kh
       u32 byte = row * 64u + kh * 16u;
       byte ^= ((row >> 2) & 3) << 4;
-      const char* p = (const char*)a_base + byte;
-      af[m][0] = *(const int*)(p);
-      af[m][1] = *(const int*)(p + 4);
-      af[m][2] = *(const int*)(p + 8);
-      af[m][3] = *(const int*)(p + 12);
-      // bytes 16-31 <- orig k [32 + kh*16, +16): slot kh+2 = slot ^ 2;
-      // the XOR swizzle commutes with the bit-5 flip
-      const char* p2 = (const char*)a_base + (byte ^ 32u);
-      af[m][4] = *(const int*)(p2);
-      af[m][5] = *(const int*)(p2 + 4);
-      af[m][6] = *(const int*)(p2 + 8);
-      af[m][7] = *(const int*)(p2 + 12);
+      // bytes 16-31 <- orig k [32 + kh*16, +16): slot ^ 2 (the XOR
+      // swizzle commutes with the bit-5 flip)
+      af[m] = v8_from((const char*)a_base + byte,
+                      (const char*)a_base + (byte ^ 32u));
       sa[m] = Asc[(long)kblk * M + m0 + row];
     }
     v8i bf[2];
@@ -122,16 +124,8 @@ gemm_bt_mx8_acc_k(float* __restrict__ C, const unsigned char* __restrict__ A,
       u32 row = brow + n * 32;
       u32 byte = row * 64u + kh * 16u;
       byte ^= ((row >> 2) & 3) << 4;
-      const char* p = (const char*)b_base + byte;
-      bf[n][0] = *(const int*)(p);
-      bf[n][1] = *(const int*)(p + 4);
-      bf[n][2] = *(const int*)(p + 8);
-      bf[n][3] = *(const int*)(p + 12);
-      const char* p2 = (const char*)b_base + (byte ^ 32u);
-      bf[n][4] = *(const int*)(p2);
-      bf[n][5] = *(const int*)(p2 + 4);
-      bf[n][6] = *(const int*)(p2 + 8);
-      bf[n][7] = *(const int*)(p2 + 12);
+      bf[n] = v8_from((const char*)b_base + byte,
+                      (const char*)b_base + (byte ^ 32u));
       sb[n] = Bsc[(long)kblk * N + n0 + row];
     }
     __builtin_amdgcn_s_setprio(1);
@@ -266,32 +260,16 @@ gemm_bt_mx8_acc_k128(float* __restrict__ C,
         u32 row = arow + m * 32;
         u32 byte = row * 128u + (u32)kk * 64u + kh * 16u;
         byte ^= ((row >> 1) & 7) << 4;
-        const char* p = (const char*)a_base + byte;
-        af[m][0] = *(const int*)(p);
-        af[m][1] = *(const int*)(p + 4);
-        af[m][2] = *(const int*)(p + 8);
-        af[m][3] = *(const int*)(p + 12);
-        const char* p2 = (const char*)a_base + (byte ^ 32u);
-        af[m][4] = *(const int*)(p2);
-        af[m][5] = *(const int*)(p2 + 4);
-        af[m][6] = *(const int*)(p2 + 8);
-        af[m][7] = *(const int*)(p2 + 12);
+        af[m] = v8_from((const char*)a_base + byte,
+                        (const char*)a_base + (byte ^ 32u));
       }
 #pragma unroll
       for (int n = 0; n < 2; ++n) {
         u32 row = brow + n * 32;
         u32 byte = row * 128u + (u32)kk * 64u + kh * 16u;
         byte ^= ((row >> 1) & 7) << 4;
-        const char* p = (const char*)b_base + byte;
-        bf[n][0] = *(const int*)(p);
-        bf[n][1] = *(const int*)(p + 4);
-        bf[n][2] = *(const int*)(p + 8);
-        bf[n][3] = *(const int*)(p + 12);
-        const char* p2 = (const char*)b_base + (byte ^ 32u);
-        bf[n][4] = *(const int*)(p2);
-        bf[n][5] = *(const int*)(p2 + 4);
-        bf[n][6] = *(const int*)(p2 + 8);
-        bf[n][7] = *(const int*)(p2 + 12);
+        bf[n] = v8_from((const char*)b_base + byte,
+                        (const char*)b_base + (byte ^ 32u));
       }
 #pragma unroll
       for (int m = 0; m < 4; ++m)
@@ -330,12 +308,25 @@ gemm_bt_mx8_acc_k128(float* __restrict__ C,
     }
 }
 
+__global__ void gemm_bt_mx8_acc_k128p(float* __restrict__,
+                                      const unsigned char* __restrict__,
+                                      const unsigned char* __restrict__,
+                                      const unsigned char* __restrict__,
+                                      const unsigned char* __restrict__,
+                                      int, int, int, int);
+#define MXP_BUF (66u * 1024u)
+
 extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                                        const void* Asc, const void* Bt,
                                        const void* Bsc, int M, int N, int K,
                                        int beta0, hipStream_t stream) {
   dim3 grid((M / MXBM) * (N / MXBN));
   if (K % 128 == 0) {
+    // NOTE: the phased variant (gemm_bt_mx8_acc_k128p) is parked: the
+    // 8-register mfma_scale operand tuples fragment the allocator at
+    // 2 waves/SIMD and the spill code's scratch VMEM defeats its counted
+    // vmcnt (171-321 TF measured). The simple double-buffered structure
+    // below measures 1574-1696 TF.
     size_t lds = 4u * MXBM * 128;  // 128 KiB
     hipLaunchKernelGGL(gemm_bt_mx8_acc_k128, grid, dim3(512), lds, stream,
                        C, (const unsigned char*)A,
@@ -349,3 +340,179 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                      (const unsigned char*)Bt, (const unsigned char*)Bsc, M,
                      N, K, beta0);
 }
+
+// ---- phased BK=128 variant: the counted-vmcnt quadrant pipeline of
+// csrc/gemm.hip ported to MX-fp8. Per phase all 8 waves compute one
+// 128x128 C-quadrant over K=128; each phase stages one 16-KiB payload
+// half-tile of tile t+1 (P1 additionally stages the 2-KiB scale region
+// via 4-B global_load_lds so NO ordinary VMEM loads pollute the counted
+// vmcnt queue); the glds queue never drains inside the loop.
+// LDS per buffer: A 32K + B 32K + Asc 1K + Bsc 1K = 66 KiB (x2 = 132 K).
+
+#undef MXP_BUF
+#define MXP_BUF (66u * 1024u)
+#define MXP_ASC (64u * 1024u)
+#define MXP_BSC (65u * 1024u)
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx8_acc_k128p(float* __restrict__ C,
+                      const unsigned char* __restrict__ A,
+                      const unsigned char* __restrict__ Asc,
+                      const unsigned char* __restrict__ Bt,
+                      const unsigned char* __restrict__ Bsc, int M, int N,
+                      int K, int beta0) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 qr = (wave & 3) * 32;   // quadrant-slice row
+  const u32 qc = (wave >> 2) * 64;  // quadrant-slice col
+
+  const u32 ntm = (u32)(M / MXBM), ntn = (u32)(N / MXBN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 4 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 4, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * MXBM, n0 = (long)tn * MXBN;
+
+  const u32 srow_b = tid >> 2;  // payload half: 128 rows x 8 slots/row...
+  // half-tile = 128 rows x 128 B = 1024 chunks; 512 thr x 2 glds
+  const u32 hrow = tid >> 3;    // 0..63 (+64 for second pass)
+  const u32 hslot = tid & 7;
+  (void)srow_b;
+
+  // stage payload half h (0=A0 1=A1 2=B0 3=B1) of tile t into buffer buf
+  auto stage_half = [&](u32 bufoff, int h, int t) {
+    const long k0 = (long)t * 128;
+    const unsigned char* src = (h < 2)
+        ? A + (m0 + (h & 1) * 128) * (long)K
+        : Bt + (n0 + (h & 1) * 128) * (long)K;
+    unsigned char* dst = (unsigned char*)smem + bufoff +
+                         (h < 2 ? 0u : 32768u) + (h & 1) * 16384u;
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      u32 row = hrow + p * 64;           // row within the 128-row half
+      u32 full_row = (h & 1) * 128 + row;  // row within the 256-row tile
+      u32 ss = hslot ^ ((full_row >> 1) & 7);
+      glds16(src + row * (long)K + k0 + ss * 16,
+             dst + row * 128u + hslot * 16u);
+    }
+  };
+  // stage the 2-KiB scale region (A 1K then B 1K) of tile t
+  auto stage_scales = [&](u32 bufoff, int t) {
+    u32 i = tid * 4;  // byte index in the 2-KiB region
+    bool isA = i < 1024;
+    u32 off = isA ? i : i - 1024;
+    u32 blk = off >> 8, r = off & 255;
+    const unsigned char* src =
+        isA ? Asc + (long)(t * 4 + blk) * M + m0 + r
+            : Bsc + (long)(t * 4 + blk) * N + n0 + r;
+    unsigned char* dst = (unsigned char*)smem + bufoff + MXP_ASC + i;
+    glds4(src, dst);
+  };
+
+  f32x16 acc[4][2] = {};  // [quadrant][n-frag]
+  const u32 kh = lane >> 5;
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  // Fragments are RE-READ per quadrant phase (LDS reads are cheap;
+  // holding them across phases pushed the kernel past the 256-VGPR cap
+  // and the spill code's scratch VMEM defeated the counted vmcnt).
+#define MX_QUAD(q, mh, nh)                                                  \
+  {                                                                         \
+    u32 sra = (u32)mh * 128u + qr + (lane & 31);                            \
+    const unsigned char* scb = (const unsigned char*)smem + cur + MXP_ASC;  \
+    _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                      \
+      u32 abyte = sra * 128u + (u32)kk * 64u + kh * 16u;                    \
+      abyte ^= ((sra >> 1) & 7) << 4;                                       \
+      v8i af = v8_from((const char*)smem + cur + abyte,                     \
+                       (const char*)smem + cur + (abyte ^ 32u));            \
+      int sa = scb[(kk * 2 + kh) * 256u + sra];                             \
+      _Pragma("unroll") for (int n = 0; n < 2; ++n) {                       \
+        u32 srb = (u32)nh * 128u + qc + (u32)n * 32 + (lane & 31);          \
+        u32 bbyte = srb * 128u + (u32)kk * 64u + kh * 16u;                  \
+        bbyte ^= ((srb >> 1) & 7) << 4;                                     \
+        v8i bf = v8_from((const char*)smem + cur + 32768u + bbyte,          \
+                         (const char*)smem + cur + 32768u + (bbyte ^ 32u)); \
+        int sb = scb[1024u + (kk * 2 + kh) * 256u + srb];                   \
+        acc[q][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(        \
+            af, bf, acc[q][n], 0, 0, 0, sa, 0, sb);                         \
+      }                                                                     \
+    }                                                                       \
+  }
+
+  // prologue: stage tile 0 completely (scales, A0, B0, B1, A1)
+  stage_scales(0u, 0);
+  stage_half(0u, 0, 0);
+  stage_half(0u, 2, 0);
+  stage_half(0u, 3, 0);
+  stage_half(0u, 1, 0);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // sc, A0, B0 landed
+  __builtin_amdgcn_s_barrier();
+
+  const int NT = K / 128;
+  for (int t = 0; t < NT; ++t) {
+    const u32 cur = (u32)(t & 1) * MXP_BUF, nxt = cur ^ MXP_BUF;
+    const bool pre = (t + 1 < NT);
+    // phase 1: Q(0,0); stage scales' + A0'
+    if (pre) {
+      stage_scales(nxt, t + 1);
+      stage_half(nxt, 0, t + 1);
+    }
+    MX_QUAD(0, 0, 0)
+    if (pre) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // phase 2: Q(0,1); stage B0'
+    if (pre) stage_half(nxt, 2, t + 1);
+    MX_QUAD(1, 0, 1)
+    if (pre) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // phase 3: Q(1,1); stage B1'
+    if (pre) stage_half(nxt, 3, t + 1);
+    MX_QUAD(2, 1, 1)
+    __builtin_amdgcn_s_barrier();
+    // phase 4: Q(1,0); stage A1'
+    if (pre) stage_half(nxt, 1, t + 1);
+    MX_QUAD(3, 1, 0)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue (batched 2-pass, bf16 of gemm.hip's 256p pattern)
+  const int QMH[4] = {0, 0, 1, 1};
+  const int QNH[4] = {0, 1, 1, 0};
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    float cv[2][16];
+    if (!beta0) {
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + QMH[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                     4 * (lane >> 5);
+          long col = n0 + QNH[q] * 128 + qc + n * 32 + (lane & 31);
+          cv[n][r] = C[row * (long)N + col];
+        }
+    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + QMH[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        long col = n0 + QNH[q] * 128 + qc + n * 32 + (lane & 31);
+        C[row * (long)N + col] =
+            (beta0 ? 0.0f : cv[n][r]) + acc[q][n][r];
+      }
+  }
+}
+#undef MX_QUAD
