@@ -195,7 +195,9 @@ class Ewise(Plan):
     def key(self) -> tuple:
         k = (self.children[0].key(), self.children[1].key())
         if self.op in ("add", "mul", "min", "max"):  # commutative
-            k = tuple(sorted(k))
+            # keys can contain non-orderable payloads (SelectPredicate):
+            # canonicalize by repr, which is deterministic for these tuples
+            k = tuple(sorted(k, key=repr))
         return ("e" + self.op,) + k
 
     def with_children(self, ch: List[Plan]) -> Plan:
