@@ -356,6 +356,57 @@ torch::Tensor transpose(torch::Tensor a) {
   return out;
 }
 
+// ---- native batch executor -------------------------------------------
+// Executes a whole block-matmul task list without returning to Python
+// between launches (the reference's DAGScheduler-task-loop equivalent for
+// the per-block inner loop; SURVEY.md §2.2 "host-side DAG executor").
+// Pays off in the fine-block regime (block-sparse plans with thousands of
+// sub-millisecond GEMMs) where per-call Python dispatch is measurable.
+//
+// tasks: int64 [n][4] rows (acc_idx, a_idx, bt_idx, beta0).
+void gemm_bt_batch(std::vector<torch::Tensor> accs,
+                   std::vector<torch::Tensor> as,
+                   std::vector<torch::Tensor> bts, torch::Tensor tasks) {
+  TORCH_CHECK(tasks.dim() == 2 && tasks.size(1) == 4,
+              "tasks must be [n][4]");
+  TORCH_CHECK(tasks.scalar_type() == torch::kInt64);
+  auto t = tasks.cpu().contiguous();
+  const int64_t* p = t.data_ptr<int64_t>();
+  int64_t n = t.size(0);
+  hipStream_t stream = cur_stream();
+  for (int64_t i = 0; i < n; ++i) {
+    auto& C = accs.at((size_t)p[i * 4 + 0]);
+    auto& A = as.at((size_t)p[i * 4 + 1]);
+    auto& Bt = bts.at((size_t)p[i * 4 + 2]);
+    bool beta0 = p[i * 4 + 3] != 0;
+    int M = (int)A.size(0), K = (int)A.size(1), N = (int)Bt.size(0);
+    TORCH_CHECK(C.size(0) == M && C.size(1) == N && Bt.size(1) == K);
+    TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
+                "batch tasks must be tile-aligned");
+    if (is_bf16(A)) {
+      if (beta0)
+        launch_gemm_bt_bf16_acc_b0(C.data_ptr<float>(), A.data_ptr(),
+                                   Bt.data_ptr(), M, N, K, stream);
+      else
+        launch_gemm_bt_bf16_acc(C.data_ptr<float>(), A.data_ptr(),
+                                Bt.data_ptr(), M, N, K, stream);
+    } else {
+      if (beta0) C.zero_();
+      TORCH_CHECK(K % 32 == 0);
+      launch_gemm_bt_f32_acc(C.data_ptr<float>(), A.data_ptr<float>(),
+                             Bt.data_ptr<float>(), M, N, K, stream);
+    }
+  }
+}
+
+// batched transpose (B-panel preparation without Python round trips)
+std::vector<torch::Tensor> transpose_batch(std::vector<torch::Tensor> ts) {
+  std::vector<torch::Tensor> out;
+  out.reserve(ts.size());
+  for (auto& a : ts) out.push_back(transpose(a));
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -399,6 +450,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce2d", &reduce2d, "full/row/col reduction");
   m.def("trace", &trace, "diagonal sum");
   m.def("transpose", &transpose, "materialized transpose");
+  m.def("gemm_bt_batch", &gemm_bt_batch,
+        "execute a block-GEMM task list natively (no Python per launch)");
+  m.def("transpose_batch", &transpose_batch, "batched transpose");
   m.def("spmm_csr_acc", &spmm_csr_acc, "C(f32) += CSR @ B (within-block sparse)");
   m.def("mx8_convert", &mx8_convert, "bf16 -> (e4m3, e8m0 block scales)");
   m.def("gemm_bt_mx8_acc", &gemm_bt_mx8_acc,

@@ -76,12 +76,73 @@ def _mx8_usable(t: torch.Tensor) -> bool:
             and t.shape[1] % 32 == 0)
 
 
+def _try_batch_local(a: BlockMatrix, b: BlockMatrix,
+                     out: BlockMatrix) -> bool:
+    """Fine-block fast path: hand the whole per-block GEMM task list to
+    the native C++ batch executor (csrc/ext.cpp gemm_bt_batch) - one
+    Python->native transition for the entire plan instead of one per
+    block pair. Used for tile-aligned all-dense bf16 plans with many
+    small blocks (block-sparse workloads); falls back otherwise."""
+    if out.device.type != "cuda" or out.dtype != torch.bfloat16:
+        return False
+    if get_config().gemm_precision == "mx8":
+        return False
+    bs = out.block_size
+    if bs % 256 != 0 or bs > 8192:
+        return False
+    if (a.nrows % bs or a.ncols % bs or b.ncols % bs
+            or a.block_size != bs or b.block_size != bs):
+        return False
+    for _, t in a.local_items():
+        if blockops.is_sparse(t) or t.dtype != torch.bfloat16:
+            return False
+    for _, t in b.local_items():
+        if blockops.is_sparse(t) or t.dtype != torch.bfloat16:
+            return False
+    accs, acc_keys = [], {}
+    a_list, a_idx = [], {}
+    bt_list, bt_idx = [], {}
+    tasks = []
+    for (bi, bj) in out.owned_indices():
+        ks = [k for k in range(a.grid_cols)
+              if (bi, k) in a.blocks and (k, bj) in b.blocks]
+        if not ks:
+            continue
+        ai = acc_keys.get((bi, bj))
+        if ai is None:
+            ai = len(accs)
+            acc_keys[(bi, bj)] = ai
+            accs.append(torch.empty(out.block_shape(bi, bj),
+                                    dtype=torch.float32,
+                                    device=out.device))
+        for pos, k in enumerate(ks):
+            if (bi, k) not in a_idx:
+                a_idx[(bi, k)] = len(a_list)
+                a_list.append(a.blocks[(bi, k)].contiguous())
+            if (k, bj) not in bt_idx:
+                bt_idx[(k, bj)] = len(bt_list)
+                bt_list.append(_bt_cached(b, k, bj, b.blocks[(k, bj)]))
+            tasks.append((ai, a_idx[(bi, k)], bt_idx[(k, bj)],
+                          1 if pos == 0 else 0))
+    if len(tasks) < 16:
+        return False
+    from ..ops.backend import require_ext
+    require_ext().gemm_bt_batch(
+        accs, a_list, bt_list,
+        torch.tensor(tasks, dtype=torch.int64))
+    for (bi, bj), ai in acc_keys.items():
+        out.blocks[(bi, bj)] = accs[ai].to(out.dtype)
+    return True
+
+
 def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     """Pure-local block multiply: every needed A/B block is already on the
     rank that owns the corresponding C block (single process, or
     partition-aware alignment: A row-partitioned + B replicated, etc.).
     Loops (j, k) outer so each B block is transposed once for the B^T MFMA
     kernel and reused across the block-row loop."""
+    if _try_batch_local(a, b, out):
+        return
     fused_ok = (out.device.type == "cuda" and out.dtype == torch.bfloat16)
     mx8 = get_config().gemm_precision == "mx8"
     for (bi, bj) in out.owned_indices():

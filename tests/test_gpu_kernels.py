@@ -343,3 +343,18 @@ def test_mx8_end_to_end(gpu_session):
         assert (err.mean() / ref.abs().mean()).item() < 0.05
     finally:
         cfg.gemm_precision = old
+
+
+def test_batch_executor_fine_blocks():
+    """Native C++ batch executor path (fine tile-aligned blocks)."""
+    import matrel_amd as ma
+    from matrel_amd.exec.runtime import reset_runtime
+    reset_runtime()
+    ma.set_config(ma.MatrelConfig(block_size=512))
+    s = ma.MatrelSession()
+    A = s.randn(4096, 4096, seed=60)
+    B = s.randn(4096, 2048, seed=61)
+    C = (A @ B).compute()  # 8x4 C blocks x 8 k = 256 tasks -> batch path
+    ref = A.to_tensor() @ B.to_tensor()
+    assert (C.to_tensor() - ref).abs().max().item() < 1.5
+    reset_runtime()
