@@ -254,6 +254,10 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                                 async_op=True)
         b_cols, b_stack, wb = [], None, None
         if k in kb:
+            # B panel is broadcast PRE-TRANSPOSED: the owner transposes
+            # each block once (the B^T GEMM operand), so the other pc-1
+            # ranks in the column group skip the transpose entirely and
+            # receivers feed panels straight into the MFMA kernel.
             b_src = rt.rank_of(k % pr, q)
             b_cols = [j for j in my_js if pb[k, j]]
             b_stack = torch.empty(len(b_cols), bs, bs, dtype=b.dtype,
@@ -263,8 +267,14 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                     t = blockops.densify(b.get_block_dense(k, j))
                     h, w = t.shape
                     if (h, w) != (bs, bs):
-                        b_stack[n].zero_()
-                    b_stack[n, :h, :w] = t
+                        pad = torch.zeros(bs, bs, dtype=t.dtype,
+                                          device=t.device)
+                        pad[:h, :w] = t
+                        t = pad
+                    if t.is_cuda:
+                        b_stack[n] = blockops.transpose(t)
+                    else:
+                        b_stack[n] = t.t()
             wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
                                 async_op=True)
         return a_rows, a_stack, b_cols, b_stack, wa, wb
@@ -296,7 +306,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             cur = nxt
             continue
         for n_j, j in enumerate(b_cols):
-            bt = _pre_t(b_stack[n_j])
+            bt = b_stack[n_j]  # already transposed by the panel owner
             for n_i, i in enumerate(a_rows):
                 c = acc.get((i, j))
                 first = c is None
