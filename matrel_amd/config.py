@@ -57,12 +57,6 @@ class MatrelConfig:
     # A block whose density is below this threshold is stored sparse (CSR).
     sparse_threshold: float = 0.05
 
-    # --- SUMMA ---
-    # Number of k-blocks broadcast per SUMMA step (panel depth). 1 panel =
-    # one block column of A / block row of B. Panels are double-buffered so
-    # the broadcast of panel k+1 overlaps the local GEMM of panel k.
-    summa_panel_depth: int = 1
-
     # --- execution ---
     # Use a separate HIP stream for collectives so comm overlaps compute.
     overlap_comm: bool = True

@@ -337,8 +337,11 @@ class Executor:
         return m.to_dense()
 
     def _solve(self, n: Solve) -> BlockMatrix:
-        a_full = self._gather_dense(n.children[0]).to(torch.float64)
-        b_full = self._gather_dense(n.children[1]).to(torch.float64)
+        from ..config import get_config
+        sdt = (torch.float64 if get_config().solve_dtype == "fp64"
+               else torch.float32)
+        a_full = self._gather_dense(n.children[0]).to(sdt)
+        b_full = self._gather_dense(n.children[1]).to(sdt)
         # SPD fast path (X^T X of the OLS pipeline): Cholesky, fall back to
         # LU for general A (solve in fp32/fp64).
         try:
@@ -350,7 +353,10 @@ class Executor:
         return _vector_result(x.float(), ref, tuple(x.shape))
 
     def _inverse(self, n: Inverse) -> BlockMatrix:
-        a_full = self._gather_dense(n.children[0]).to(torch.float64)
+        from ..config import get_config
+        sdt = (torch.float64 if get_config().solve_dtype == "fp64"
+               else torch.float32)
+        a_full = self._gather_dense(n.children[0]).to(sdt)
         x = torch.linalg.inv(a_full)
         ref = self._eval(n.children[0])
         return _vector_result(x.float(), ref, tuple(x.shape))
