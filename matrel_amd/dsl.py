@@ -309,5 +309,9 @@ class Matrix:
         from .io.format import save_matrix
         save_matrix(self._materialized(), path)
 
+    def save_coo(self, path: str) -> None:
+        from .io.format import save_coo_text
+        save_coo_text(self._materialized(), path)
+
     def __repr__(self) -> str:
         return f"Matrix({self.nrows}x{self.ncols}, plan={type(self.plan).__name__})"

@@ -141,3 +141,18 @@ def load_coo_text(path: str, nrows: int, ncols: int,
     for key, t in staged.items():
         out.blocks[key] = t.to(device=out.device, dtype=out.dtype)
     return out
+
+
+def save_coo_text(m: BlockMatrix, path: str) -> None:
+    """Export nonzeros as 'i j v' text (the reference's COO text format).
+    Rank 0 writes; intended for small/sparse matrices."""
+    rt = get_runtime()
+    full = m.to_dense()
+    if rt.rank == 0:
+        nz = full.nonzero(as_tuple=False).cpu()
+        vals = full[full != 0].cpu()
+        with open(path, "w") as f:
+            f.write("# i j v\n")
+            for (i, j), v in zip(nz.tolist(), vals.tolist()):
+                f.write(f"{i} {j} {v}\n")
+    rt.barrier()

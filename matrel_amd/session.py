@@ -40,6 +40,24 @@ class MatrelSession:
     def zeros(self, m: int, n: int, **kw) -> Matrix:
         return Matrix.from_blocks(BlockMatrix.zeros(m, n, **kw))
 
+    def ones(self, m: int, n: int, **kw) -> Matrix:
+        import torch
+        bm = BlockMatrix._filled(m, n, lambda s, g: torch.ones(*s), None,
+                                 **kw)
+        return Matrix.from_blocks(bm)
+
+    def eye(self, n: int, **kw) -> Matrix:
+        """Identity: only diagonal blocks materialized (block-sparse)."""
+        import torch
+        bm = BlockMatrix(n, n, **kw)
+        for (bi, bj) in bm.owned_indices():
+            if bi != bj:
+                continue
+            h, w = bm.block_shape(bi, bj)
+            t = torch.eye(h, w)
+            bm.blocks[(bi, bj)] = t.to(device=bm.device, dtype=bm.dtype)
+        return Matrix.from_blocks(bm)
+
     def from_dense(self, t: torch.Tensor, **kw) -> Matrix:
         return Matrix.from_blocks(BlockMatrix.from_dense(t, **kw))
 

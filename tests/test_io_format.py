@@ -53,3 +53,28 @@ def test_coo_text_import(session, tmp_path):
     assert t[10, 20] == -2.0
     assert t[99, 49] == 3.25
     assert t.count_nonzero() == 3
+
+
+def test_ones_eye_factories(session):
+    import torch
+    A = session.ones(70, 50)
+    assert torch.equal(A.to_tensor(), torch.ones(70, 50))
+    I = session.eye(100)
+    assert torch.equal(I.to_tensor(), torch.eye(100))
+    # identity is block-sparse: off-diagonal blocks absent
+    assert len(I.blocks.blocks) == I.blocks.grid_rows
+    B = session.randn(100, 60, seed=1)
+    out = (I @ B).compute()
+    assert torch.allclose(out.to_tensor(), B.to_tensor(), atol=1e-5)
+
+
+def test_coo_roundtrip(session, tmp_path):
+    import torch
+    t = torch.zeros(60, 40)
+    t[3, 4] = 1.5
+    t[59, 39] = -2.25
+    A = session.from_dense(t)
+    p = str(tmp_path / "out.coo")
+    A.save_coo(p)
+    B = session.load_coo(p, 60, 40)
+    assert torch.equal(B.to_tensor(), t)
