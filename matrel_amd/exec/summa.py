@@ -135,12 +135,94 @@ def _try_batch_local(a: BlockMatrix, b: BlockMatrix,
     return True
 
 
+def _bt_strip(b: BlockMatrix, k: int) -> torch.Tensor:
+    """Concatenated transposed B block-row k: [N_total, bs] bf16 (cached
+    on the matrix; absent blocks materialize as zeros)."""
+    cache = getattr(b, "_bt_strip_cache", None)
+    if cache is None:
+        cache = b._bt_strip_cache = {}
+    s = cache.get(k)
+    if s is None:
+        parts = [_bt_cached(b, k, bj, blockops.densify(
+                     b.get_block_dense(k, bj)))
+                 for bj in range(b.grid_cols)]
+        s = torch.cat(parts, dim=0).contiguous()
+        cache[k] = s
+    return s
+
+
+def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
+                     out: BlockMatrix) -> bool:
+    """Sparse-A strip mode: fuse a present A block's whole row of
+    per-block GEMMs into ONE wide-N GEMM against the concatenated
+    (transposed) B block-row - C(bi, :) += A(bi, k) @ B(k, :). Fine
+    blocks (config 4's 1%-block-density SpMM) then run in the
+    large-grid phased-GEMM regime instead of many under-filled small
+    launches."""
+    if out.device.type != "cuda" or out.dtype != torch.bfloat16:
+        return False
+    if get_config().gemm_precision == "mx8":
+        return False
+    bs = out.block_size
+    if bs % 256 != 0 or bs > 4096 or b.grid_cols < 4:
+        return False
+    if (a.nrows % bs or a.ncols % bs or b.ncols % bs
+            or a.block_size != bs or b.block_size != bs):
+        return False
+    # only pays off when A is block-sparse (dense A prefers the fused
+    # per-block path); require full B rows for the strips
+    total_a = a.grid_rows * a.grid_cols
+    if len(a.blocks) * 2 > total_a:
+        return False
+    for _, t in a.local_items():
+        if blockops.is_sparse(t) or t.dtype != torch.bfloat16:
+            return False
+    n_total = b.ncols
+    my_bis = sorted({bi for (bi, k) in a.blocks})
+    strip_bytes = 4 * bs * n_total
+    if strip_bytes * len(my_bis) > 64 << 30:
+        return False
+    accs, acc_idx = [], {}
+    a_list, a_map = [], {}
+    bt_list, bt_map = [], {}
+    tasks = []
+    for bi in my_bis:
+        ks = sorted(k for (i, k) in a.blocks if i == bi)
+        ai = len(accs)
+        acc_idx[bi] = ai
+        accs.append(torch.empty(bs, n_total, dtype=torch.float32,
+                                device=out.device))
+        for pos, k in enumerate(ks):
+            if (bi, k) not in a_map:
+                a_map[(bi, k)] = len(a_list)
+                a_list.append(a.blocks[(bi, k)].contiguous())
+            if k not in bt_map:
+                bt_map[k] = len(bt_list)
+                bt_list.append(_bt_strip(b, k))
+            tasks.append((ai, a_map[(bi, k)], bt_map[k],
+                          1 if pos == 0 else 0))
+    if not tasks:
+        return True  # nothing to compute; out stays empty (zeros)
+    from ..ops.backend import require_ext
+    require_ext().gemm_bt_batch(accs, a_list, bt_list,
+                                torch.tensor(tasks, dtype=torch.int64))
+    for bi, ai in acc_idx.items():
+        strip = accs[ai]
+        for bj in range(out.grid_cols):
+            if not out.is_local(bi, bj):
+                continue
+            out.blocks[(bi, bj)] =                 strip[:, bj * bs:(bj + 1) * bs].to(out.dtype).contiguous()
+    return True
+
+
 def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     """Pure-local block multiply: every needed A/B block is already on the
     rank that owns the corresponding C block (single process, or
     partition-aware alignment: A row-partitioned + B replicated, etc.).
     Loops (j, k) outer so each B block is transposed once for the B^T MFMA
     kernel and reused across the block-row loop."""
+    if _try_strip_local(a, b, out):
+        return
     if _try_batch_local(a, b, out):
         return
     fused_ok = (out.device.type == "cuda" and out.dtype == torch.bfloat16)

@@ -358,3 +358,18 @@ def test_batch_executor_fine_blocks():
     ref = A.to_tensor() @ B.to_tensor()
     assert (C.to_tensor() - ref).abs().max().item() < 1.5
     reset_runtime()
+
+
+def test_strip_mode_sparse_spmm():
+    """Strip-mode SpMM (sparse A, fused wide-N GEMMs) vs dense ref."""
+    import matrel_amd as ma
+    from matrel_amd.exec.runtime import reset_runtime
+    reset_runtime()
+    ma.set_config(ma.MatrelConfig(block_size=256))
+    s = ma.MatrelSession()
+    A = s.block_sparse_random(4096, 4096, 0.1, seed=70)
+    B = s.randn(4096, 2048, seed=71)
+    C = (A @ B).compute()
+    ref = A.to_tensor() @ B.to_tensor()
+    assert (C.to_tensor() - ref).abs().max().item() < 1.0
+    reset_runtime()
