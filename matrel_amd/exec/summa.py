@@ -170,9 +170,10 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
             or a.block_size != bs or b.block_size != bs):
         return False
     # only pays off when A is block-sparse (dense A prefers the fused
-    # per-block path); require full B rows for the strips
-    total_a = a.grid_rows * a.grid_cols
-    if len(a.blocks) * 2 > total_a:
+    # per-block path). Use the GLOBAL presence fraction - the local block
+    # count under a distributed scheme would make dense matrices look
+    # sparse (1/world of the blocks are local).
+    if a.block_density() > 0.5:
         return False
     for _, t in a.local_items():
         if blockops.is_sparse(t) or t.dtype != torch.bfloat16:
