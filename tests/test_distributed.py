@@ -118,6 +118,26 @@ def _body_save_load(s, tmpdir):
     assert torch.allclose(A.to_tensor(), B.to_tensor(), atol=0)
 
 
+def _body_partition_aware_local(s, tmpdir):
+    """Schemes already aligned (A row-partitioned, B replicated) => the
+    planner must pick the zero-communication 'local' algorithm."""
+    from matrel_amd.blocks.schemes import BCAST, ROW
+    from matrel_amd.dsl import Matrix
+    from matrel_amd.blocks.matrix import BlockMatrix
+    from matrel_amd.plan.optimizer import optimize
+    from matrel_amd.plan.physical import plan_physical
+    A = Matrix.from_blocks(BlockMatrix.randn(96, 64, seed=21, scheme=ROW))
+    B = Matrix.from_blocks(
+        BlockMatrix.randn(64, 4096, seed=22, scheme=BCAST))
+    expr = A @ B
+    opt = optimize(expr.plan)
+    pp = plan_physical(opt)
+    assert pp.get(opt).mm_algo == "local", pp.get(opt).mm_algo
+    C = expr.compute()
+    ref = A.to_tensor() @ B.to_tensor()
+    assert torch.allclose(C.to_tensor(), ref, atol=1e-2)
+
+
 def _body_ols(s, tmpdir):
     X = s.randn(128, 16, seed=11)
     y = s.randn(128, 1, seed=12)
@@ -132,7 +152,7 @@ def _body_ols(s, tmpdir):
 @pytest.mark.parametrize("body", [
     "_body_summa", "_body_bcast_mm", "_body_aggregates", "_body_transpose",
     "_body_repartition", "_body_block_sparse", "_body_sparse_skewed",
-    "_body_save_load", "_body_ols",
+    "_body_save_load", "_body_partition_aware_local", "_body_ols",
 ])
 def test_distributed(body, tmp_path):
     _run(body, tmp_path)
