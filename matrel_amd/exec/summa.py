@@ -135,6 +135,26 @@ def _try_batch_local(a: BlockMatrix, b: BlockMatrix,
     return True
 
 
+def _pad_t(t: torch.Tensor, bs: int) -> torch.Tensor:
+    if t.shape == (bs, bs):
+        return t
+    pad = torch.zeros(bs, bs, dtype=t.dtype, device=t.device)
+    pad[: t.shape[0], : t.shape[1]] = t
+    return pad
+
+
+def _mx8_padded(m: BlockMatrix, key, t: torch.Tensor, bs: int):
+    """Cached MX-fp8 conversion of a (zero-padded) block."""
+    cache = getattr(m, "_mx8_cache", None)
+    if cache is None:
+        cache = m._mx8_cache = {}
+    ent = cache.get(key)
+    if ent is None:
+        ent = blockops.to_mx8(_pad_t(blockops.densify(t), bs))
+        cache[key] = ent
+    return ent
+
+
 def _bt_strip(b: BlockMatrix, k: int) -> torch.Tensor:
     """Concatenated transposed B block-row k: [N_total, bs] bf16 (cached
     on the matrix; absent blocks materialize as zeros)."""
@@ -289,6 +309,12 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     gk = a.grid_cols
     assert gk == b.grid_rows
     pa, pb = _presence_map(a), _presence_map(b)
+    bs0 = a.block_size
+    # MX-fp8 SUMMA: owners convert panels to fp8+scales ONCE and the
+    # broadcasts move HALF the bf16 bytes over xGMI (+ 1/32 for scales).
+    mx8 = (get_config().gemm_precision == "mx8"
+           and a.device.type == "cuda" and a.dtype == torch.bfloat16
+           and b.dtype == torch.bfloat16 and bs0 % 256 == 0)
 
     my_is = [i for i in range(out.grid_rows) if i % pr == p]
     my_js = [j for j in range(out.grid_cols) if j % pc == q]
@@ -319,23 +345,44 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         in ka (kb) - group-uniform by construction.
         """
         a_rows, a_stack, wa = [], None, None
+        a_sc = None
         if k in ka:
             a_src = rt.rank_of(p, k % pc)
             a_rows = [i for i in my_is if pa[i, k]]
-            # empty, not zeros: the broadcast overwrites it on receivers,
-            # and the source zero-pads only ragged edge blocks
-            a_stack = torch.empty(len(a_rows), bs, bs, dtype=a.dtype,
-                                  device=a.device)
-            if rt.rank == a_src:
-                for n, i in enumerate(a_rows):
-                    t = blockops.densify(a.get_block_dense(i, k))
-                    h, w = t.shape
-                    if (h, w) != (bs, bs):
-                        a_stack[n].zero_()
-                    a_stack[n, :h, :w] = t
-            wa = dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
-                                async_op=True)
+            if mx8:
+                a_stack = torch.empty(len(a_rows), bs, bs,
+                                      dtype=torch.uint8, device=a.device)
+                a_sc = torch.empty(len(a_rows), bs // 32, bs,
+                                   dtype=torch.uint8, device=a.device)
+                if rt.rank == a_src:
+                    for n, i in enumerate(a_rows):
+                        p8, sc = _mx8_padded(a, ("s", i, k),
+                                             a.get_block_dense(i, k), bs)
+                        a_stack[n] = p8
+                        a_sc[n] = sc
+                # NCCL executes ops of one communicator in enqueue order,
+                # so waiting the (later) scale broadcast also fences the
+                # payload broadcast.
+                dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
+                               async_op=True)
+                wa = dist.broadcast(a_sc, src=a_src, group=rt.row_group(p),
+                                    async_op=True)
+            else:
+                # empty, not zeros: the broadcast overwrites it on
+                # receivers; the source zero-pads only ragged edge blocks
+                a_stack = torch.empty(len(a_rows), bs, bs, dtype=a.dtype,
+                                      device=a.device)
+                if rt.rank == a_src:
+                    for n, i in enumerate(a_rows):
+                        t = blockops.densify(a.get_block_dense(i, k))
+                        h, w = t.shape
+                        if (h, w) != (bs, bs):
+                            a_stack[n].zero_()
+                        a_stack[n, :h, :w] = t
+                wa = dist.broadcast(a_stack, src=a_src,
+                                    group=rt.row_group(p), async_op=True)
         b_cols, b_stack, wb = [], None, None
+        b_sc = None
         if k in kb:
             # B panel is broadcast PRE-TRANSPOSED: the owner transposes
             # each block once (the B^T GEMM operand), so the other pc-1
@@ -343,24 +390,42 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             # receivers feed panels straight into the MFMA kernel.
             b_src = rt.rank_of(k % pr, q)
             b_cols = [j for j in my_js if pb[k, j]]
-            b_stack = torch.empty(len(b_cols), bs, bs, dtype=b.dtype,
-                                  device=b.device)
-            if rt.rank == b_src:
-                for n, j in enumerate(b_cols):
-                    t = blockops.densify(b.get_block_dense(k, j))
-                    h, w = t.shape
-                    if (h, w) != (bs, bs):
-                        pad = torch.zeros(bs, bs, dtype=t.dtype,
-                                          device=t.device)
-                        pad[:h, :w] = t
-                        t = pad
-                    if t.is_cuda:
-                        b_stack[n] = blockops.transpose(t)
-                    else:
-                        b_stack[n] = t.t()
-            wb = dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
-                                async_op=True)
-        return a_rows, a_stack, b_cols, b_stack, wa, wb
+            if mx8:
+                b_stack = torch.empty(len(b_cols), bs, bs,
+                                      dtype=torch.uint8, device=b.device)
+                b_sc = torch.empty(len(b_cols), bs // 32, bs,
+                                   dtype=torch.uint8, device=b.device)
+                if rt.rank == b_src:
+                    for n, j in enumerate(b_cols):
+                        t = _pad_t(blockops.densify(
+                            b.get_block_dense(k, j)), bs)
+                        p8, sc = _mx8_padded(b, ("st", k, j),
+                                             blockops.transpose(t), bs)
+                        b_stack[n] = p8
+                        b_sc[n] = sc
+                dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
+                               async_op=True)
+                wb = dist.broadcast(b_sc, src=b_src, group=rt.col_group(q),
+                                    async_op=True)
+            else:
+                b_stack = torch.empty(len(b_cols), bs, bs, dtype=b.dtype,
+                                      device=b.device)
+                if rt.rank == b_src:
+                    for n, j in enumerate(b_cols):
+                        t = blockops.densify(b.get_block_dense(k, j))
+                        h, w = t.shape
+                        if (h, w) != (bs, bs):
+                            pad = torch.zeros(bs, bs, dtype=t.dtype,
+                                              device=t.device)
+                            pad[:h, :w] = t
+                            t = pad
+                        if t.is_cuda:
+                            b_stack[n] = blockops.transpose(t)
+                        else:
+                            b_stack[n] = t.t()
+                wb = dist.broadcast(b_stack, src=b_src,
+                                    group=rt.col_group(q), async_op=True)
+        return a_rows, a_stack, a_sc, b_cols, b_stack, b_sc, wa, wb
 
     # last contributing k per owned (i, j) - enables the fused bf16-out
     # epilogue on that call (no fp32 store + no conversion pass).
@@ -380,7 +445,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     cur = issue_bcast(ks[0])
     for step, k in enumerate(ks):
         nxt = issue_bcast(ks[step + 1]) if step + 1 < len(ks) else None
-        a_rows, a_stack, b_cols, b_stack, wa, wb = cur
+        a_rows, a_stack, a_sc, b_cols, b_stack, b_sc, wa, wb = cur
         if wa is not None:
             wa.wait()
         if wb is not None:
@@ -397,6 +462,10 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                     c = torch.empty(bs, bs, dtype=torch.float32,
                                     device=out.device)
                     acc[(i, j)] = c
+                if mx8:
+                    blockops.gemm_bt_mx8_acc(c, a_stack[n_i], a_sc[n_i],
+                                             bt, b_sc[n_j], beta0=first)
+                    continue
                 if fused_ok and last_k.get((i, j)) == k:
                     ob = torch.empty(bs, bs, dtype=out.dtype,
                                      device=out.device)
