@@ -70,9 +70,20 @@ __global__ void mx8_convert_k(unsigned char* __restrict__ out8,
     sc[sidx] = (unsigned char)(e + 127);
   }
   float inv = exp2f((float)-e);
-  unsigned char* dst = out8 + r * cols + b * 32;
+  // pack 32 e4m3 bytes into 8 dwords and store as two 16-B vectors
+  unsigned int words[8];
 #pragma unroll
-  for (int i = 0; i < 32; ++i) dst[i] = f32_to_e4m3(v[i] * inv);
+  for (int w = 0; w < 8; ++w) {
+    unsigned int x = 0;
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      x |= (unsigned int)f32_to_e4m3(v[w * 4 + j] * inv) << (8 * j);
+    words[w] = x;
+  }
+  typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
+  u32x4* dst = (u32x4*)(out8 + r * cols + b * 32);
+  dst[0] = u32x4{words[0], words[1], words[2], words[3]};
+  dst[1] = u32x4{words[4], words[5], words[6], words[7]};
 }
 
 extern "C" void launch_mx8_convert(void* out8, void* sc, const void* in,
