@@ -232,7 +232,8 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
         for bj in range(out.grid_cols):
             if not out.is_local(bi, bj):
                 continue
-            out.blocks[(bi, bj)] =                 strip[:, bj * bs:(bj + 1) * bs].to(out.dtype).contiguous()
+            out.blocks[(bi, bj)] = (
+                strip[:, bj * bs:(bj + 1) * bs].to(out.dtype).contiguous())
     return True
 
 
@@ -309,16 +310,15 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     gk = a.grid_cols
     assert gk == b.grid_rows
     pa, pb = _presence_map(a), _presence_map(b)
-    bs0 = a.block_size
+    bs = a.block_size
     # MX-fp8 SUMMA: owners convert panels to fp8+scales ONCE and the
     # broadcasts move HALF the bf16 bytes over xGMI (+ 1/32 for scales).
     mx8 = (get_config().gemm_precision == "mx8"
            and a.device.type == "cuda" and a.dtype == torch.bfloat16
-           and b.dtype == torch.bfloat16 and bs0 % 256 == 0)
+           and b.dtype == torch.bfloat16 and bs % 256 == 0)
 
     my_is = [i for i in range(out.grid_rows) if i % pr == p]
     my_js = [j for j in range(out.grid_cols) if j % pc == q]
-    bs = a.block_size
 
     # fp32 accumulators for owned C blocks
     acc: Dict[Tuple[int, int], torch.Tensor] = {}
