@@ -328,14 +328,20 @@ class BlockMatrix:
                      torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
                                  device=self.device))
                 dist.broadcast(t, src=src)
+                from ..exec.metrics import add_comm
+                add_comm("repartition_bcast",
+                         t.numel() * t.element_size())
                 out.blocks[(bi, bj)] = t
             elif src == dst:
                 if rt.rank == src:
                     out.blocks[(bi, bj)] = self.blocks[(bi, bj)]
             else:
                 if rt.rank == src:
-                    dist.send(_densify(self.blocks[(bi, bj)]).contiguous(),
-                              dst=dst)
+                    tt = _densify(self.blocks[(bi, bj)]).contiguous()
+                    dist.send(tt, dst=dst)
+                    from ..exec.metrics import add_comm
+                    add_comm("repartition_send",
+                             tt.numel() * tt.element_size())
                 elif rt.rank == dst:
                     t = torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
                                     device=self.device)

@@ -19,7 +19,17 @@ from typing import Any, Dict
 _ops: Dict[str, Dict[str, float]] = defaultdict(
     lambda: {"count": 0, "wall_s": 0.0, "est_flops": 0.0,
              "est_comm_bytes": 0.0})
+_comm: Dict[str, Dict[str, float]] = defaultdict(
+    lambda: {"count": 0, "bytes": 0.0})
 _run_start = time.time()
+
+
+def add_comm(kind: str, nbytes: int) -> None:
+    """Record ACTUAL bytes handed to a collective (broadcast / all_reduce
+    / send; called by summa.py and BlockMatrix.repartition)."""
+    e = _comm[kind]
+    e["count"] += 1
+    e["bytes"] += float(nbytes)
 
 
 @contextmanager
@@ -45,6 +55,8 @@ def report() -> Dict[str, Any]:
         "op_wall_s": total_wall,
         "avg_tflops": (total_flops / total_wall / 1e12) if total_wall else 0,
         "ops": {k: dict(v) for k, v in sorted(_ops.items())},
+        "collectives": {k: dict(v) for k, v in sorted(_comm.items())},
+        "total_comm_gb": sum(v["bytes"] for v in _comm.values()) / 1e9,
     }
     path = os.environ.get("MATREL_METRICS_PATH")
     if path:
@@ -56,4 +68,5 @@ def report() -> Dict[str, Any]:
 def reset() -> None:
     global _run_start
     _ops.clear()
+    _comm.clear()
     _run_start = time.time()

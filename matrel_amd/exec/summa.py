@@ -26,6 +26,7 @@ import torch.distributed as dist
 from ..blocks.matrix import BlockMatrix
 from ..config import get_config
 from ..ops import blockops
+from .metrics import add_comm
 from .runtime import get_runtime
 
 
@@ -381,6 +382,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                         a_stack[n, :h, :w] = t
                 wa = dist.broadcast(a_stack, src=a_src,
                                     group=rt.row_group(p), async_op=True)
+            add_comm("summa_a_bcast", a_stack.numel()
+                     * a_stack.element_size())
         b_cols, b_stack, wb = [], None, None
         b_sc = None
         if k in kb:
@@ -425,6 +428,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                             b_stack[n] = t.t()
                 wb = dist.broadcast(b_stack, src=b_src,
                                     group=rt.col_group(q), async_op=True)
+            add_comm("summa_b_bcast", b_stack.numel()
+                     * b_stack.element_size())
         return a_rows, a_stack, a_sc, b_cols, b_stack, b_sc, wa, wb
 
     # last contributing k per owned (i, j) - enables the fused bf16-out
