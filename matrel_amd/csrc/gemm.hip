@@ -369,8 +369,19 @@ gemm_bt_bf16_acc_256_k(float* __restrict__ C, const bf16* __restrict__ A,
 // flight across every barrier). Quadrant snake order (0,0)->(0,1)->
 // (1,1)->(1,0) reuses one operand's fragments at each transition.
 
+// DEEPRING: stage ALL FOUR halves of tile t+1 during phases 1-2 of tile t
+// (legal: buffer nxt's last reader is tile t-1, whose final read of any
+// half region is >=1 barrier before t.p1). Every vmcnt then proves a load
+// issued >=4 phases earlier, so the waits never actually park the wave -
+// counted ring per the 8-phase template's "vmcnt never 0 in the main
+// loop" rule. Steady-state waits: p1-end vmcnt(6) proves B1(cur),
+// p2-end vmcnt(8) proves A1(cur), p4-end vmcnt(4) proves A0',B0'.
+// SCHED: split each phase's reads+MFMAs by kk half and pin the shape with
+// sched_group_barrier so the kk=0 MFMA cluster starts after only its own
+// 6 LDS reads land (the compiler's conservative lgkmcnt(0) then covers
+// exactly those 6); the kk=1 reads' latency hides under the kk=0 MFMAs.
 template <bool MIDBAR, bool STATICPRIO = false, bool BETA0 = false,
-          bool BF16OUT = false>
+          bool BF16OUT = false, bool DEEPRING = false, bool SCHED = false>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
@@ -449,6 +460,35 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
         bfr[kk][n] = *(const bf16x8*)(smem + off + byte);
       }
   };
+  // SCHED: per-kk fragment reads, so each phase can be hard-split with
+  // sched_barrier(0) fences into {kk0 reads + stage | kk0 MFMA cluster |
+  // kk1 reads | kk1 MFMA cluster}: the compiler's conservative lgkmcnt(0)
+  // before the kk0 cluster then covers only the 2-6 kk0 reads (instead of
+  // all 12), and the kk1 reads' LDS latency hides under the kk0 MFMAs.
+  auto read_a_kk = [&](u32 off, int kk) {
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      u32 row = arow_base + m * 16;
+      u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      af[kk][m] = *(const bf16x8*)(smem + off + byte);
+    }
+  };
+  auto read_b_kk = [&](u32 off, int kk) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      u32 row = brow_base + n * 16;
+      u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      bfr[kk][n] = *(const bf16x8*)(smem + off + byte);
+    }
+  };
+
+#define MFMA_KK(q, kk)                                                     \
+  _Pragma("unroll") for (int m = 0; m < 2; ++m)                            \
+      _Pragma("unroll") for (int n = 0; n < 4; ++n)                        \
+          acc[q][m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
+              af[kk][m], bfr[kk][n], acc[q][m][n], 0, 0, 0);
 
 #define MFMA_QUAD(q)                                                       \
   if (!STATICPRIO) __builtin_amdgcn_s_setprio(1);                          \
@@ -479,38 +519,100 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   for (int t = 0; t < NT; ++t) {
     const u32 cur = (u32)(t & 1) * BUFSZ, nxt = cur ^ BUFSZ;
     const bool pre = (t + 1 < NT);
-    // ---- phase 1: Q(0,0) - reads A0,B0; stages A0 of t+1
-    read_a(cur + HOFF(0, 0));
-    read_b(cur + HOFF(0, 2));
-    if (pre) stage_half(nxt, 0, t + 1);
-    if (MIDBAR) __builtin_amdgcn_s_barrier();
-    MFMA_QUAD(0)
+    // ---- phase 1: Q(0,0) - reads A0,B0; stages A0 of t+1 (+B0' if DEEP)
+    if constexpr (SCHED) {
+      read_a_kk(cur + HOFF(0, 0), 0);
+      read_b_kk(cur + HOFF(0, 2), 0);
+      if (pre) stage_half(nxt, 0, t + 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(0, 0)
+      __builtin_amdgcn_sched_barrier(0);
+      read_a_kk(cur + HOFF(0, 0), 1);
+      read_b_kk(cur + HOFF(0, 2), 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(0, 1)
+    } else {
+      read_a(cur + HOFF(0, 0));
+      read_b(cur + HOFF(0, 2));
+      if (pre) stage_half(nxt, 0, t + 1);
+      if (DEEPRING && pre) stage_half(nxt, 2, t + 1);
+      if (MIDBAR) __builtin_amdgcn_s_barrier();
+      MFMA_QUAD(0)
+    }
     // drain B1(t); leave {A1(t), A0'} (or {A1} when not staging)
-    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    if (DEEPRING) {
+      if (pre) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    } else {
+      if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
     // ---- phase 2: Q(0,1) - reads B1 (A frags reused); stages B0'
-    read_b(cur + HOFF(0, 3));
-    if (pre) stage_half(nxt, 2, t + 1);
-    if (MIDBAR) __builtin_amdgcn_s_barrier();
-    MFMA_QUAD(1)
+    // (DEEP: B1'+A1' - tile t+1 fully issued by end of this phase)
+    if constexpr (SCHED) {
+      read_b_kk(cur + HOFF(0, 3), 0);
+      if (pre) stage_half(nxt, 2, t + 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(1, 0)
+      __builtin_amdgcn_sched_barrier(0);
+      read_b_kk(cur + HOFF(0, 3), 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(1, 1)
+    } else {
+      read_b(cur + HOFF(0, 3));
+      if (DEEPRING) {
+        if (pre) { stage_half(nxt, 3, t + 1); stage_half(nxt, 1, t + 1); }
+      } else {
+        if (pre) stage_half(nxt, 2, t + 1);
+      }
+      if (MIDBAR) __builtin_amdgcn_s_barrier();
+      MFMA_QUAD(1)
+    }
     // drain A1(t)
-    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (DEEPRING) {
+      if (pre) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    } else {
+      if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
     // ---- phase 3: Q(1,1) - reads A1 (B frags reused); stages B1'
-    read_a(cur + HOFF(0, 1));
-    if (pre) stage_half(nxt, 3, t + 1);
-    if (MIDBAR) __builtin_amdgcn_s_barrier();
-    MFMA_QUAD(2)
+    if constexpr (SCHED) {
+      read_a_kk(cur + HOFF(0, 1), 0);
+      if (pre) stage_half(nxt, 3, t + 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(2, 0)
+      __builtin_amdgcn_sched_barrier(0);
+      read_a_kk(cur + HOFF(0, 1), 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(2, 1)
+    } else {
+      read_a(cur + HOFF(0, 1));
+      if (!DEEPRING && pre) stage_half(nxt, 3, t + 1);
+      if (MIDBAR) __builtin_amdgcn_s_barrier();
+      MFMA_QUAD(2)
+    }
     // no barrier here: phase 4 reads B0(cur) - landed and barriered at
     // the tile boundary - and stages into the OTHER buffer.
     if (MIDBAR) __builtin_amdgcn_s_barrier();
     // ---- phase 4: Q(1,0) - re-reads B0 (landed long ago); stages A1'
-    read_b(cur + HOFF(0, 2));
-    if (pre) stage_half(nxt, 1, t + 1);
-    if (MIDBAR) __builtin_amdgcn_s_barrier();
-    MFMA_QUAD(3)
+    if constexpr (SCHED) {
+      read_b_kk(cur + HOFF(0, 2), 0);
+      if (pre) stage_half(nxt, 1, t + 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(3, 0)
+      __builtin_amdgcn_sched_barrier(0);
+      read_b_kk(cur + HOFF(0, 2), 1);
+      __builtin_amdgcn_sched_barrier(0);
+      MFMA_KK(3, 1)
+    } else {
+      read_b(cur + HOFF(0, 2));
+      if (!DEEPRING && pre) stage_half(nxt, 1, t + 1);
+      if (MIDBAR) __builtin_amdgcn_s_barrier();
+      MFMA_QUAD(3)
+    }
     // end of tile: drain up to {B1', A1'} for next P1
     if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -557,6 +659,7 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
   }
 }
 #undef MFMA_QUAD
+#undef MFMA_KK
 #undef HOFF
 
 // --------------------------- bf16, 256^2 tile, phased, 32x32x16 MFMA ----
@@ -795,6 +898,21 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<true, true>), grid,
                          dim3(512), lds, stream, C, (const bf16*)A,
                          (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
+    else if (v == 8)
+      hipLaunchKernelGGL(
+          (gemm_bt_bf16_acc_256p_k<false, true, false, false, true>), grid,
+          dim3(512), lds, stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+          K, g_st_m, g_st_n);
+    else if (v == 9)
+      hipLaunchKernelGGL(
+          (gemm_bt_bf16_acc_256p_k<false, false, false, false, true>), grid,
+          dim3(512), lds, stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+          K, g_st_m, g_st_n);
+    else if (v == 10)
+      hipLaunchKernelGGL(
+          (gemm_bt_bf16_acc_256p_k<false, true, false, false, false, true>),
+          grid, dim3(512), lds, stream, C, (const bf16*)A, (const bf16*)Bt,
+          M, N, K, g_st_m, g_st_n);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
