@@ -37,7 +37,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--n-size", type=int, default=0,
                     help="override matrix side N")
-    ap.add_argument("--block-size", type=int, default=16384)
+    ap.add_argument("--block-size", type=int, default=0,
+                    help="0 = auto (single block at 1 GPU, 16384 for SUMMA)")
     ap.add_argument("--precision", choices=["bf16", "mx8"], default="bf16",
                     help="GEMM compute precision (mx8 = MX-scaled fp8; "
                          "NOT the headline config - BASELINE is bf16)")
@@ -62,6 +63,9 @@ def main():
 
     A = sess.randn(n, n, seed=1)
     B = sess.randn(n, n, seed=2)
+
+    resolved_block = A.plan.data.block_size if hasattr(A.plan, "data") \
+        else block
 
     def step():
         C = (A @ B).compute()
@@ -113,7 +117,7 @@ def main():
             "config": {
                 "model": "NxN distributed dense matmul (BASELINE configs 2/3)",
                 "n": n,
-                "block_size": block,
+                "block_size": resolved_block,
                 "global_batch": 1,
                 "seq_len": n,
                 "parallelism": f"summa_grid{rt.grid[0]}x{rt.grid[1]}",

@@ -56,7 +56,18 @@ class BlockMatrix:
         rt = get_runtime()
         self.nrows = int(nrows)
         self.ncols = int(ncols)
-        self.block_size = int(block_size or cfg.block_size)
+        bs = int(block_size or cfg.block_size)
+        if bs <= 0:
+            # auto block size: single-process dense work wants ONE block per
+            # matrix (the raw MFMA GEMM, no block orchestration overhead) up
+            # to 65536 (8 GiB bf16 - comfortable in 288 GB HBM3E); multi-GPU
+            # keeps 16384 so SUMMA has k-panel granularity to overlap
+            # broadcasts with compute.
+            if rt.world_size == 1:
+                bs = max(1, min(65536, max(int(nrows), int(ncols))))
+            else:
+                bs = 16384
+        self.block_size = bs
         self.scheme = scheme
         self.device = device if device is not None else rt.device
         if dtype is None:
