@@ -102,8 +102,17 @@ class Leaf(Plan):
     def density(self) -> float:
         if self._density is not None:
             return self._density
+        # GLOBAL block-presence fraction (one cached all_reduce), NOT the
+        # local block count: the planner runs replicated on every rank and
+        # every cost-model input must be rank-uniform, or ranks pick
+        # different matmul algorithms and the collectives deadlock (seen
+        # at world 4 with ragged shapes: summa vs bcast_b cost inversion).
         total = self.data.grid_rows * self.data.grid_cols
-        return max(len(self.data.blocks), 1) / total if total else 1.0
+        if total == 0:
+            return 1.0
+        d = self.data.block_density()
+        self._density = max(d, 1.0 / total)
+        return self._density
 
     def key(self) -> tuple:
         return ("leaf", id(self.data))
