@@ -138,6 +138,7 @@ class BlockMatrix:
             f"block {(bi, bj)}: {tuple(t.shape)} != {self.block_shape(bi, bj)}"
         self.blocks[(bi, bj)] = t.to(device=self.device, dtype=self.dtype)
         self._presence = None  # invalidate cached presence map
+        self._nnzmap = None
 
     def coll_device(self) -> torch.device:
         """Device for small bookkeeping collectives: RCCL needs GPU
@@ -166,6 +167,30 @@ class BlockMatrix:
 
     def local_items(self) -> Iterator[Tuple[BlockIndex, torch.Tensor]]:
         return iter(self.blocks.items())
+
+    def nnz_map(self) -> torch.Tensor:
+        """Global (grid_rows x grid_cols) int64 map of STORED nnz per
+        block: CSR blocks report their nnz, dense blocks report h*w
+        (no scan - storage state is the owner's sparsify decision), absent
+        blocks 0. One all_reduce, cached; every rank agrees, which lets
+        SUMMA decide dense-vs-CSR wire format group-uniformly and size
+        the variable-length CSR broadcast buffers without a handshake."""
+        cached = getattr(self, "_nnzmap", None)
+        if cached is not None:
+            return cached
+        rt = get_runtime()
+        nm = torch.zeros(self.grid_rows, self.grid_cols, dtype=torch.int64)
+        for (bi, bj), t in self.blocks.items():
+            if t.layout == torch.sparse_csr:
+                nm[bi, bj] = t._nnz()
+            else:
+                nm[bi, bj] = t.shape[0] * t.shape[1]
+        if rt.is_distributed and not self.scheme.is_replicated():
+            dev_nm = nm.to(self.coll_device())
+            dist.all_reduce(dev_nm, op=dist.ReduceOp.SUM)
+            nm = dev_nm.cpu()
+        self._nnzmap = nm
+        return nm
 
     # --- stats (feed the optimizer's sparsity-aware cost model) ---------
 
@@ -294,7 +319,10 @@ class BlockMatrix:
             else:
                 shape = self.block_shape(bi, bj)
                 if rt.rank == owner:
-                    t = self.get_block_dense(bi, bj).float()
+                    # contiguous is mandatory: gloo/RCCL collectives on a
+                    # strided view (e.g. a ragged c[:h, :w] crop) send the
+                    # wrong bytes silently
+                    t = self.get_block_dense(bi, bj).float().contiguous()
                 else:
                     t = torch.zeros(shape, dtype=torch.float32,
                                     device=self.device)

@@ -311,6 +311,12 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     gk = a.grid_cols
     assert gk == b.grid_rows
     pa, pb = _presence_map(a), _presence_map(b)
+    # Global stored-nnz map: drives the group-uniform dense-vs-CSR wire
+    # decision for A panels (element-sparse blocks broadcast their CSR
+    # arrays - nnz-proportional xGMI bytes - instead of a densified bs^2
+    # payload). Must be called by every rank (collective, cached).
+    nma = a.nnz_map()
+    sp_thr = get_config().sparse_threshold
     bs = a.block_size
     # MX-fp8 SUMMA: owners convert panels to fp8+scales ONCE and the
     # broadcasts move HALF the bf16 bytes over xGMI (+ 1/32 for scales).
@@ -347,9 +353,13 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         """
         a_rows, a_stack, wa = [], None, None
         a_sc = None
+        a_spinfo: Dict[int, tuple] = {}
+        a_dn_index: Dict[int, int] = {}
+        waits = []
         if k in ka:
             a_src = rt.rank_of(p, k % pc)
             a_rows = [i for i in my_is if pa[i, k]]
+            a_dn_index = {i: n for n, i in enumerate(a_rows)}
             if mx8:
                 a_stack = torch.empty(len(a_rows), bs, bs,
                                       dtype=torch.uint8, device=a.device)
@@ -364,27 +374,87 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 # NCCL executes ops of one communicator in enqueue order,
                 # so waiting the (later) scale broadcast also fences the
                 # payload broadcast.
-                dist.broadcast(a_stack, src=a_src, group=rt.row_group(p),
-                               async_op=True)
-                wa = dist.broadcast(a_sc, src=a_src, group=rt.row_group(p),
-                                    async_op=True)
+                waits.append(dist.broadcast(a_stack, src=a_src,
+                                            group=rt.row_group(p),
+                                            async_op=True))
+                waits.append(dist.broadcast(a_sc, src=a_src,
+                                            group=rt.row_group(p),
+                                            async_op=True))
             else:
+                # dense-vs-CSR wire split, uniform across the row group
+                # (derived from the globally-synced nnz map + geometry)
+                sp_rows = [i for i in a_rows
+                           if int(nma[i, k]) < sp_thr
+                           * a.block_shape(i, k)[0] * a.block_shape(i, k)[1]]
+                spset = set(sp_rows)
+                dn_rows = [i for i in a_rows if i not in spset]
+                a_dn_index = {i: n for n, i in enumerate(dn_rows)}
                 # empty, not zeros: the broadcast overwrites it on
                 # receivers; the source zero-pads only ragged edge blocks
-                a_stack = torch.empty(len(a_rows), bs, bs, dtype=a.dtype,
+                a_stack = torch.empty(len(dn_rows), bs, bs, dtype=a.dtype,
                                       device=a.device)
                 if rt.rank == a_src:
-                    for n, i in enumerate(a_rows):
+                    for n, i in enumerate(dn_rows):
                         t = blockops.densify(a.get_block_dense(i, k))
                         h, w = t.shape
                         if (h, w) != (bs, bs):
                             a_stack[n].zero_()
                         a_stack[n, :h, :w] = t
-                wa = dist.broadcast(a_stack, src=a_src,
-                                    group=rt.row_group(p), async_op=True)
-            add_comm("summa_a_bcast", a_stack.numel()
-                     * a_stack.element_size())
-        b_cols, b_stack, wb = [], None, None
+                if len(dn_rows):
+                    waits.append(dist.broadcast(a_stack, src=a_src,
+                                                group=rt.row_group(p),
+                                                async_op=True))
+                    add_comm("summa_a_bcast", a_stack.numel()
+                             * a_stack.element_size())
+                if sp_rows:
+                    # concatenated CSR arrays, exact sizes known to every
+                    # rank: per block [crow (h+1) | cols (nnz)] int32,
+                    # values in the operand dtype.
+                    shapes = {i: a.block_shape(i, k) for i in sp_rows}
+                    nnzs = {i: int(nma[i, k]) for i in sp_rows}
+                    idx_len = sum(shapes[i][0] + 1 + nnzs[i]
+                                  for i in sp_rows)
+                    val_len = sum(nnzs[i] for i in sp_rows)
+                    idx_buf = torch.empty(idx_len, dtype=torch.int32,
+                                          device=a.device)
+                    val_buf = torch.empty(val_len, dtype=a.dtype,
+                                          device=a.device)
+                    if rt.rank == a_src:
+                        io = vo = 0
+                        for i in sp_rows:
+                            t = a.blocks[(i, k)]
+                            if not blockops.is_sparse(t):
+                                t = t.to_sparse_csr()
+                            h = shapes[i][0]
+                            nz = nnzs[i]
+                            idx_buf[io:io + h + 1] = \
+                                t.crow_indices().to(torch.int32)
+                            idx_buf[io + h + 1:io + h + 1 + nz] = \
+                                t.col_indices().to(torch.int32)
+                            val_buf[vo:vo + nz] = t.values()
+                            io += h + 1 + nz
+                            vo += nz
+                    waits.append(dist.broadcast(idx_buf, src=a_src,
+                                                group=rt.row_group(p),
+                                                async_op=True))
+                    waits.append(dist.broadcast(val_buf, src=a_src,
+                                                group=rt.row_group(p),
+                                                async_op=True))
+                    add_comm("summa_a_bcast_csr",
+                             idx_buf.numel() * 4
+                             + val_buf.numel() * val_buf.element_size())
+                    # per-block views into the (in-flight) buffers; valid
+                    # to read after the waits drain
+                    io = vo = 0
+                    for i in sp_rows:
+                        h, w = shapes[i]
+                        nz = nnzs[i]
+                        a_spinfo[i] = (idx_buf[io:io + h + 1],
+                                       idx_buf[io + h + 1:io + h + 1 + nz],
+                                       val_buf[vo:vo + nz], h, w)
+                        io += h + 1 + nz
+                        vo += nz
+        b_cols, b_stack = [], None
         b_sc = None
         if k in kb:
             # B panel is broadcast PRE-TRANSPOSED: the owner transposes
@@ -406,10 +476,12 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                                              blockops.transpose(t), bs)
                         b_stack[n] = p8
                         b_sc[n] = sc
-                dist.broadcast(b_stack, src=b_src, group=rt.col_group(q),
-                               async_op=True)
-                wb = dist.broadcast(b_sc, src=b_src, group=rt.col_group(q),
-                                    async_op=True)
+                waits.append(dist.broadcast(b_stack, src=b_src,
+                                            group=rt.col_group(q),
+                                            async_op=True))
+                waits.append(dist.broadcast(b_sc, src=b_src,
+                                            group=rt.col_group(q),
+                                            async_op=True))
             else:
                 b_stack = torch.empty(len(b_cols), bs, bs, dtype=b.dtype,
                                       device=b.device)
@@ -426,11 +498,15 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                             b_stack[n] = blockops.transpose(t)
                         else:
                             b_stack[n] = t.t()
-                wb = dist.broadcast(b_stack, src=b_src,
-                                    group=rt.col_group(q), async_op=True)
+                waits.append(dist.broadcast(b_stack, src=b_src,
+                                            group=rt.col_group(q),
+                                            async_op=True))
             add_comm("summa_b_bcast", b_stack.numel()
                      * b_stack.element_size())
-        return a_rows, a_stack, a_sc, b_cols, b_stack, b_sc, wa, wb
+        return {"a_rows": a_rows, "a_stack": a_stack, "a_sc": a_sc,
+                "a_spinfo": a_spinfo, "a_dn": a_dn_index,
+                "b_cols": b_cols, "b_stack": b_stack, "b_sc": b_sc,
+                "waits": waits}
 
     # last contributing k per owned (i, j) - enables the fused bf16-out
     # epilogue on that call (no fp32 store + no conversion pass).
@@ -450,23 +526,40 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     cur = issue_bcast(ks[0])
     for step, k in enumerate(ks):
         nxt = issue_bcast(ks[step + 1]) if step + 1 < len(ks) else None
-        a_rows, a_stack, a_sc, b_cols, b_stack, b_sc, wa, wb = cur
-        if wa is not None:
-            wa.wait()
-        if wb is not None:
-            wb.wait()
+        a_rows, a_stack = cur["a_rows"], cur["a_stack"]
+        a_sc, a_spinfo, a_dn = cur["a_sc"], cur["a_spinfo"], cur["a_dn"]
+        b_cols, b_stack, b_sc = cur["b_cols"], cur["b_stack"], cur["b_sc"]
+        for w_ in cur["waits"]:
+            w_.wait()
         if not (k in ka and k in kb):
             cur = nxt
             continue
         for n_j, j in enumerate(b_cols):
             bt = b_stack[n_j]  # already transposed by the panel owner
-            for n_i, i in enumerate(a_rows):
+            b_unt = None  # lazily un-transposed B for CSR panel blocks
+            for i in a_rows:
                 c = acc.get((i, j))
                 first = c is None
                 if first:
                     c = torch.empty(bs, bs, dtype=torch.float32,
                                     device=out.device)
                     acc[(i, j)] = c
+                sp = a_spinfo.get(i)
+                if sp is not None:
+                    # CSR-wire A block: spmm against the un-transposed
+                    # dense B block (one transpose-back per (j, step),
+                    # only on steps that actually carry sparse rows)
+                    if first:
+                        c.zero_()
+                    if b_unt is None:
+                        b_unt = (blockops.transpose(bt) if bt.is_cuda
+                                 else bt.t())
+                    crow, cols, vals, h, w = sp
+                    a_csr = torch.sparse_csr_tensor(
+                        crow, cols, vals, size=(h, w))
+                    blockops.spmm_acc(c[:h], a_csr, b_unt[:w])
+                    continue
+                n_i = a_dn[i]
                 if mx8:
                     blockops.gemm_bt_mx8_acc(c, a_stack[n_i], a_sc[n_i],
                                              bt, b_sc[n_j], beta0=first)
@@ -488,7 +581,10 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             out.blocks[(i, j)] = (ob if (h, w) == (bs, bs)
                                   else ob[:h, :w].contiguous())
         else:
-            out.blocks[(i, j)] = c[:h, :w].to(out.dtype)
+            # contiguous: on CPU .to() with an unchanged dtype is a no-op
+            # and a ragged crop stays a strided view, which a later
+            # collective (to_dense gather, repartition) must not see
+            out.blocks[(i, j)] = c[:h, :w].to(out.dtype).contiguous()
 
 
 def bcast_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix,
