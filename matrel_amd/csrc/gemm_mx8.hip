@@ -316,17 +316,200 @@ __global__ void gemm_bt_mx8_acc_k128p(float* __restrict__,
                                       int, int, int, int);
 #define MXP_BUF (66u * 1024u)
 
+// ------------------- BK=128, 2-phase counted-vmcnt (kk-split) -------------
+// The simple k128 kernel drains vmcnt to 0 at every __syncthreads (full
+// double-buffer flush per K-tile). This variant splits each K-tile into
+// the two kk byte-halves: phase kk0 consumes bytes [0,64) of every A/B
+// row and stages {scales', byte-half0'} of tile t+1; phase kk1 consumes
+// bytes [64,128) and stages byte-half1'. Raw s_barriers with COUNTED
+// waits - the glds queue never drains inside the loop (cdna_hip
+// _programming.md T3/T4 applied to the mx8 kernel). Scales go through
+// LDS (2 KiB/tile via glds4) so ALL operand traffic is proven by the
+// same vmcnt+barrier flow and no plain VMEM loads pollute the counted
+// queue. Register plan stays the simple kernel's (per-kk transient
+// fragments) - the parked k128p's failure was quadrant re-reads blowing
+// past 256 VGPRs, which this variant avoids.
+// LDS per buffer: A 32K | B 32K | Asc 1K | Bsc 1K = 66 KiB (x2 = 132 K).
+
+#define MXCV_BUF (66u * 1024u)
+#define MXCV_BOFF (32u * 1024u)
+#define MXCV_SOFF (64u * 1024u)
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx8_acc_k128cv(float* __restrict__ C,
+                       const unsigned char* __restrict__ A,
+                       const unsigned char* __restrict__ Asc,
+                       const unsigned char* __restrict__ Bt,
+                       const unsigned char* __restrict__ Bsc, int M, int N,
+                       int K, int beta0) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;
+
+  const u32 ntm = (u32)(M / MXBM), ntn = (u32)(N / MXBN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 4 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 4, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * MXBM, n0 = (long)tn * MXBN;
+
+  // byte-half staging: dst slots {4h..4h+3} of each 128-B row, source
+  // slot XOR-swizzled over the FULL row (same LDS image as the simple
+  // k128 kernel, so the fragment reads below are unchanged).
+  const u32 srow = tid >> 2;   // 0..127 (+128 second pass)
+  const u32 sslot4 = tid & 3;  // slot within the half
+
+  auto stage_half = [&](u32 bufoff, int h, int t) {
+    const long k0 = (long)t * 128;
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      u32 row = srow + pass * 128;
+      u32 s = (u32)h * 4 + sslot4;       // dst 16-B slot in the row
+      u32 ss = s ^ ((row >> 1) & 7);     // swizzled source slot
+      glds16(A + (m0 + row) * (long)K + k0 + ss * 16,
+             (unsigned char*)smem + bufoff + row * 128u + s * 16u);
+      glds16(Bt + (n0 + row) * (long)K + k0 + ss * 16,
+             (unsigned char*)smem + bufoff + MXCV_BOFF + row * 128u
+                 + s * 16u);
+    }
+  };
+  // 2-KiB scale region: A blocks [4t,4t+4) then B; layout blk*256 + row.
+  auto stage_scales = [&](u32 bufoff, int t) {
+    u32 i = tid * 4;
+    bool isA = i < 1024;
+    u32 off = isA ? i : i - 1024;
+    u32 blk = off >> 8, r = off & 255;
+    const unsigned char* src =
+        isA ? Asc + (long)(t * 4 + blk) * M + m0 + r
+            : Bsc + (long)(t * 4 + blk) * N + n0 + r;
+    glds4(src, (unsigned char*)smem + bufoff + MXCV_SOFF + i);
+  };
+
+  f32x16 acc[4][2] = {};
+  const u32 arow = wm * 128 + (lane & 31);
+  const u32 brow = wn * 64 + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  // one kk half: 12 ds_read_b128 frags + 6 ds_read_u8 scales + 8 MFMA
+#define MXCV_KK(kk)                                                        \
+  {                                                                        \
+    const unsigned char* a_base = (const unsigned char*)smem + cur;        \
+    const unsigned char* b_base =                                          \
+        (const unsigned char*)smem + cur + MXCV_BOFF;                      \
+    const unsigned char* s_base =                                          \
+        (const unsigned char*)smem + cur + MXCV_SOFF;                      \
+    v8i af[4], bf[2];                                                      \
+    int sa[4], sb[2];                                                      \
+    _Pragma("unroll") for (int m = 0; m < 4; ++m) {                        \
+      u32 row = arow + m * 32;                                             \
+      u32 byte = row * 128u + (u32)(kk)*64u + kh * 16u;                    \
+      byte ^= ((row >> 1) & 7) << 4;                                       \
+      af[m] = v8_from((const char*)a_base + byte,                          \
+                      (const char*)a_base + (byte ^ 32u));                 \
+      sa[m] = s_base[((u32)(kk)*2u + kh) * 256u + row];                    \
+    }                                                                      \
+    _Pragma("unroll") for (int n = 0; n < 2; ++n) {                        \
+      u32 row = brow + n * 32;                                             \
+      u32 byte = row * 128u + (u32)(kk)*64u + kh * 16u;                    \
+      byte ^= ((row >> 1) & 7) << 4;                                       \
+      bf[n] = v8_from((const char*)b_base + byte,                          \
+                      (const char*)b_base + (byte ^ 32u));                 \
+      sb[n] = s_base[1024u + ((u32)(kk)*2u + kh) * 256u + row];            \
+    }                                                                      \
+    _Pragma("unroll") for (int m = 0; m < 4; ++m)                          \
+        _Pragma("unroll") for (int n = 0; n < 2; ++n)                      \
+            acc[m][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(   \
+                af[m], bf[n], acc[m][n], 0, 0, 0, sa[m], 0, sb[n]);        \
+  }
+
+  // prologue: tile 0 fully staged; prove {scales, half0} before phase 1
+  stage_scales(0u, 0);
+  stage_half(0u, 0, 0);
+  stage_half(0u, 1, 0);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // sc0 + h0(0) landed
+  __builtin_amdgcn_s_barrier();
+
+  const int NT = K / 128;
+  for (int t = 0; t < NT; ++t) {
+    const u32 cur = (u32)(t & 1) * MXCV_BUF, nxt = cur ^ MXCV_BUF;
+    const bool pre = (t + 1 < NT);
+    // phase kk0: consume {sc(t), h0(t)}; stage {sc(t+1), h0(t+1)}
+    if (pre) {
+      stage_scales(nxt, t + 1);
+      stage_half(nxt, 0, t + 1);
+    }
+    MXCV_KK(0)
+    // prove h1(t): queue [h1(t) 4, sc+h0(t+1) 5] -> 5; tail tile: [h1] -> 0
+    if (pre) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // phase kk1: consume h1(t); stage h1(t+1)
+    if (pre) stage_half(nxt, 1, t + 1);
+    MXCV_KK(1)
+    // prove {sc, h0}(t+1): queue [sc+h0(t+1) 5, h1(t+1) 4] -> 4
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+#undef MXCV_KK
+
+  const long ccol = n0 + wn * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      float cv[16];
+      if (!beta0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                     8 * (r >> 2) + 4 * (lane >> 5);
+          cv[r] = C[row * (long)N + ccol + n * 32];
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                   8 * (r >> 2) + 4 * (lane >> 5);
+        C[row * (long)N + ccol + n * 32] =
+            (beta0 ? 0.0f : cv[r]) + acc[m][n][r];
+      }
+    }
+}
+
+static int g_mx8_variant = 0;  // 0 = auto (simple k128), 2 = counted 2-phase
+extern "C" void set_mx8_variant(int v) { g_mx8_variant = v; }
+
 extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                                        const void* Asc, const void* Bt,
                                        const void* Bsc, int M, int N, int K,
                                        int beta0, hipStream_t stream) {
   dim3 grid((M / MXBM) * (N / MXBN));
   if (K % 128 == 0) {
-    // NOTE: the phased variant (gemm_bt_mx8_acc_k128p) is parked: the
-    // 8-register mfma_scale operand tuples fragment the allocator at
-    // 2 waves/SIMD and the spill code's scratch VMEM defeats its counted
-    // vmcnt (171-321 TF measured). The simple double-buffered structure
-    // below measures 1574-1696 TF.
+    if (g_mx8_variant == 2) {
+      size_t lds = 2u * MXCV_BUF;  // 132 KiB
+      hipLaunchKernelGGL(gemm_bt_mx8_acc_k128cv, grid, dim3(512), lds,
+                         stream, C, (const unsigned char*)A,
+                         (const unsigned char*)Asc,
+                         (const unsigned char*)Bt,
+                         (const unsigned char*)Bsc, M, N, K, beta0);
+      return;
+    }
+    // NOTE: the quadrant-phased variant (gemm_bt_mx8_acc_k128p) is
+    // parked: the 8-register mfma_scale operand tuples fragment the
+    // allocator at 2 waves/SIMD and the spill code's scratch VMEM
+    // defeats its counted vmcnt (171-321 TF measured). The simple
+    // double-buffered structure below measures 1574-1696 TF; the
+    // kk-split counted variant above is the A/B candidate.
     size_t lds = 4u * MXBM * 128;  // 128 KiB
     hipLaunchKernelGGL(gemm_bt_mx8_acc_k128, grid, dim3(512), lds, stream,
                        C, (const unsigned char*)A,
