@@ -15,7 +15,7 @@ N = 64
 
 def _gen_spec(rng, depth):
     if depth == 0 or rng.random() < 0.25:
-        return ("leaf", rng.randrange(3))
+        return ("leaf", rng.randrange(4))
     op = rng.choice(["add", "sub", "mul", "mm", "t", "scale", "sel",
                      "min", "max"])
     if op in ("add", "sub", "mul", "mm", "min", "max"):
@@ -72,6 +72,9 @@ def _worker(rank, world, port):
     ma.set_config(ma.MatrelConfig(block_size=24))
     s = ma.MatrelSession()
     leaves = [(s.randn(N, N, seed=k), None) for k in (1, 2, 3)]
+    # element-sparse (CSR-stored) leaf: random trees route it through
+    # ewise densify, SUMMA CSR-wire panels, select masks, transpose
+    leaves.append((s.sparse_random(N, N, 0.02, seed=4), None))
     leaves = [(m, m.to_tensor()) for m, _ in leaves]
     rng = random.Random(77)
     for i in range(25):

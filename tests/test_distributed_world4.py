@@ -31,7 +31,7 @@ def _worker(rank: int, world: int, port: int, fn_name: str, tmpdir: str):
 
 
 def _run(fn_name: str, tmp_path) -> None:
-    port = 29810 + abs(hash(fn_name)) % 150
+    port = 29965 + abs(hash(fn_name)) % 30
     mp.spawn(_worker, args=(WORLD, port, fn_name, str(tmp_path)),
              nprocs=WORLD, join=True)
 
