@@ -138,6 +138,30 @@ def _body_elem_sparse_summa(s, tmpdir):
     assert torch.allclose(out2.to_dense(), ref2, atol=1e-3)
 
 
+def _body_relational(s, tmpdir):
+    """Distributed relational verbs (SURVEY.md §2.1 joins/aggregates over
+    matrix data): row-index join (gram), cross-product join + aggregate,
+    value semi-join - all over real collectives."""
+    A = s.randn(96, 48, seed=31)
+    B = s.randn(96, 48, seed=32)
+    at, bt = A.to_tensor(), B.to_tensor()
+    g = A.join_rows(B).compute()
+    assert torch.allclose(g.to_tensor(), at.t() @ bt, atol=1e-2)
+    u = s.randn(48, 1, seed=33)
+    v = s.randn(1, 64, seed=34)
+    total = u.cross_join(v, "mul").sum().item()
+    ref = (u.to_tensor() @ v.to_tensor()).sum().item()
+    assert abs(total - ref) < 1e-1
+    torch.manual_seed(7)  # same tensor on every rank
+    t = torch.randint(0, 5, (96, 96)).float()
+    C = s.from_dense(t)
+    D = s.from_dense(torch.tensor([[1.0, 3.0]]))
+    out = C.semi_join_values(D).compute()
+    ct = C.to_tensor()
+    ref = torch.where((ct == 1) | (ct == 3), ct, torch.zeros(()))
+    assert torch.allclose(out.to_tensor(), ref, atol=0)
+
+
 def _body_save_load(s, tmpdir):
     A = s.randn(96, 64, seed=10)
     path = os.path.join(tmpdir, "dist_mat")
@@ -180,7 +204,7 @@ def _body_ols(s, tmpdir):
 @pytest.mark.parametrize("body", [
     "_body_summa", "_body_bcast_mm", "_body_aggregates", "_body_transpose",
     "_body_repartition", "_body_block_sparse", "_body_sparse_skewed",
-    "_body_elem_sparse_summa",
+    "_body_elem_sparse_summa", "_body_relational",
     "_body_save_load", "_body_partition_aware_local", "_body_ols",
 ])
 def test_distributed(body, tmp_path):
