@@ -437,7 +437,8 @@ class CrossJoin(Plan):
     the rank-structured case realized as outer products. We expose the
     common case used by the reference workloads: join A (m x 1) with
     B (1 x n) producing C[i,j] = f(A[i], B[j]) for f in {mul, add}.
-    (General relational joins on indices: exec-level, see exec/relational.)
+    (General index-equality joins are the join_elem/join_rows/join_cols
+    verbs, executed block-wise in the executor.)
     """
 
     def __init__(self, l: Plan, r: Plan, op: str = "mul") -> None:
