@@ -61,7 +61,7 @@ def _build(spec, leaves):
     return am.select_rows(lo, hi), ref
 
 
-def _worker(rank, world, port):
+def _worker(rank, world, port, iters=25):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
                       MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
     sys.path.insert(0, os.path.dirname(os.path.dirname(
@@ -76,8 +76,8 @@ def _worker(rank, world, port):
     # ewise densify, SUMMA CSR-wire panels, select masks, transpose
     leaves.append((s.sparse_random(N, N, 0.02, seed=4), None))
     leaves = [(m, m.to_tensor()) for m, _ in leaves]
-    rng = random.Random(77)
-    for i in range(25):
+    rng = random.Random(77 + world)
+    for i in range(iters):
         spec = _gen_spec(rng, 4)
         m, ref = _build(spec, leaves)
         out = m.compute().to_tensor()
@@ -88,3 +88,9 @@ def _worker(rank, world, port):
 
 def test_distributed_fuzz():
     mp.spawn(_worker, args=(2, 29810), nprocs=2, join=True)
+
+
+def test_distributed_fuzz_world4():
+    """Same random trees on the 2x2 grid (both sub-group dimensions
+    non-trivial - the driver's 4/8-GPU shape class)."""
+    mp.spawn(_worker, args=(4, 29812, 12), nprocs=4, join=True)
