@@ -373,3 +373,46 @@ def test_strip_mode_sparse_spmm():
     ref = A.to_tensor() @ B.to_tensor()
     assert (C.to_tensor() - ref).abs().max().item() < 1.0
     reset_runtime()
+
+
+# ------------------------------------------------- CSR-wire spmm shape --
+def test_spmm_wire_views(ext):
+    """The SUMMA CSR-wire path reconstructs per-block CSR from narrow
+    views into concatenated int32/value broadcast buffers and accumulates
+    into a row-prefix of a bs^2 accumulator - exercise exactly that
+    calling pattern on GPU (single rank: buffers built locally)."""
+    from matrel_amd.ops import blockops
+    torch.manual_seed(11)
+    bs = 512
+    blocks = []
+    for h, w, density in ((bs, bs, 0.02), (bs, 384, 0.03), (128, bs, 0.05)):
+        d = torch.rand(h, w).cuda()
+        d = torch.where(d < density, torch.randn(h, w).cuda(), 
+                        torch.zeros((), device="cuda"))
+        blocks.append(d.to_sparse_csr())
+    # concatenated wire buffers, per block [crow | cols] int32 + values
+    idx_parts, val_parts, meta = [], [], []
+    for t in blocks:
+        h, w = t.shape
+        nz = t._nnz()
+        idx_parts.append(t.crow_indices().to(torch.int32))
+        idx_parts.append(t.col_indices().to(torch.int32))
+        val_parts.append(t.values().float())
+        meta.append((h, w, nz))
+    idx_buf = torch.cat(idx_parts).cuda()
+    val_buf = torch.cat(val_parts).cuda()
+    b = torch.randn(bs, bs).cuda()
+    io = vo = 0
+    for (h, w, nz), t in zip(meta, blocks):
+        crow = idx_buf[io:io + h + 1]
+        cols = idx_buf[io + h + 1:io + h + 1 + nz]
+        vals = val_buf[vo:vo + nz]
+        io += h + 1 + nz
+        vo += nz
+        a_csr = torch.sparse_csr_tensor(crow, cols, vals, size=(h, w))
+        c = torch.zeros(bs, bs, dtype=torch.float32, device="cuda")
+        blockops.spmm_acc(c[:h], a_csr, b[:w].contiguous())
+        ref = t.to_dense().float() @ b[:w].float()
+        err = (c[:h] - ref).abs().max().item()
+        assert err < 1e-3, (h, w, err)
+        assert c[h:].abs().max().item() == 0.0
