@@ -415,4 +415,5 @@ def test_spmm_wire_views(ext):
         ref = t.to_dense().float() @ b[:w].float()
         err = (c[:h] - ref).abs().max().item()
         assert err < 1e-3, (h, w, err)
-        assert c[h:].abs().max().item() == 0.0
+        if h < bs:
+            assert c[h:].abs().max().item() == 0.0
