@@ -60,6 +60,11 @@ def _body_sparse_2x2(s, tmpdir):
     C = (A @ B).compute()
     ref = A.to_tensor() @ B.to_tensor()
     assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
+    # element-sparse (CSR) operand + transpose through the full path
+    E = s.sparse_random(128, 96, 0.03, seed=17)
+    D = (E.T @ A).compute()
+    refd = E.to_tensor().t() @ A.to_tensor()
+    assert torch.allclose(D.to_tensor(), refd, atol=1e-3)
 
 
 def _body_agg_2x2(s, tmpdir):
