@@ -11,7 +11,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import matrel_amd as ma  # noqa
 
 
-def timeit(fn, iters=3, warmup=1):
+def timeit(fn, iters=5, warmup=3):
     for _ in range(warmup):
         fn()
     torch.cuda.synchronize()
