@@ -811,6 +811,136 @@ gemm_bt_bf16_acc_256p32_k(float* __restrict__ C, const bf16* __restrict__ A,
 #undef MFMA_QUAD32
 #undef HOFF32
 
+// ----------------------- bf16, 256^2 tile, 4 FAT waves (512 VGPR) --------
+// Geometry reverse-engineered from hipBLASLt's winning gfx950 kernel for
+// this shape class (Tensile name: MT256x256x64 MI16x16 MIWT8_8 WG32_8_1
+// WS64 - profiles/README.md "Library comparison"): 256 threads = 4 waves
+// at ONE wave per SIMD, each wave owning a 128x128 quadrant as an 8x8
+// fragment grid held in a ~350-VGPR register tile (the unified 512-VGPR
+// file is only addressable at occupancy 1). Payoff vs the 8-wave kernel:
+// 16 ds_reads feed 64 MFMAs per kk (0.25 reads/MFMA vs 0.75) and there is
+// no 2-wave/SIMD arbitration; latency is hidden by the wave's own ILP
+// (64 independent MFMA chains) instead of a co-resident wave.
+
+__global__ void
+__launch_bounds__(256, 1)
+__attribute__((amdgpu_waves_per_eu(1, 1)))
+gemm_bt_bf16_acc_256w4_k(float* __restrict__ C, const bf16* __restrict__ A,
+                         const bf16* __restrict__ Bt, int M, int N, int K,
+                         u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                        // [2][256*64]
+  bf16* Bs = (bf16*)(smem + 2 * BM2 * BKH * 2);  // [2][256*64]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wr = (wave >> 1) * 128, wc = (wave & 1) * 128;
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  // staging: 256 rows x 8 slots per operand = 2048 chunks; 256 threads
+  // -> 8 passes of 32 rows; same XOR-16 source swizzle as the 8-wave
+  // kernels (identical LDS image, identical fragment reads).
+  const u32 srow = tid >> 3, sslot = tid & 7;  // rows 0..31 per pass
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  // rolled (unroll 2): a fully unrolled 16-glds stage keeps 16 64-bit
+  // addresses live and pushes the 512-VGPR allocation into in-loop
+  // scratch spills; incremental pointers keep the register high-water
+  // under the cap (VALU cost hides under the MFMA clusters).
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKH;
+    const bf16* ga = A + (m0 + srow) * (long)K + k0 + src_slot * 8;
+    const bf16* gb = Bt + (n0 + srow) * (long)K + k0 + src_slot * 8;
+    bf16* la = As + buf * BM2 * BKH + (srow * 8u + sslot) * 8u;
+    bf16* lb = Bs + buf * BM2 * BKH + (srow * 8u + sslot) * 8u;
+#pragma unroll 2
+    for (int it = 0; it < 8; ++it) {
+      glds16(ga, la);
+      glds16(gb, lb);
+      ga += 32 * (long)K;
+      gb += 32 * (long)K;
+      la += 32 * 64;
+      lb += 32 * 64;
+    }
+  };
+
+  f32x4 acc[8][8] = {};  // 256 VGPRs: the 128x128 register tile
+  const u32 arow_base = wr + (lane & 15);
+  const u32 brow_base = wc + (lane & 15);
+  const u32 kgrp = lane >> 4;
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);
+    const char* a_base = (const char*)(As + cur * BM2 * BKH);
+    const char* b_base = (const char*)(Bs + cur * BM2 * BKH);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[8], bf[8];
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+        u32 row = arow_base + m * 16;
+        u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        af[m] = *(const bf16x8*)(a_base + byte);
+      }
+#pragma unroll
+      for (int n = 0; n < 8; ++n) {
+        u32 row = brow_base + n * 16;
+        u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+        byte ^= (row & 7) << 4;
+        bf[n] = *(const bf16x8*)(b_base + byte);
+      }
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 8; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[m], bf[n], acc[m][n], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: batched two-pass per fragment row
+  const u32 crow0 = wr + (lane >> 4) * 4;
+  const u32 ccol0 = wc + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+    float cv[8][4];
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      const float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cv[n][r] = cp[(long)r * N];
+    }
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        cp[(long)r * N] = cv[n][r] + acc[m][n][r];
+    }
+  }
+}
+
 // ------------------------------------------------------------- launchers --
 
 static int g_gemm_variant = 0;  // 0=auto, 1=simple 256, 2=phased 256, 3=128
@@ -913,6 +1043,10 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
           (gemm_bt_bf16_acc_256p_k<false, true, false, false, false, true>),
           grid, dim3(512), lds, stream, C, (const bf16*)A, (const bf16*)Bt,
           M, N, K, g_st_m, g_st_n);
+    else if (v == 12)
+      hipLaunchKernelGGL(gemm_bt_bf16_acc_256w4_k, grid, dim3(256), lds,
+                         stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+                         K, g_st_m, g_st_n);
     else
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256_k, grid, dim3(512), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
