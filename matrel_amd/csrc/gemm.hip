@@ -941,6 +941,147 @@ gemm_bt_bf16_acc_256w4_k(float* __restrict__ C, const bf16* __restrict__ A,
   }
 }
 
+// -------------------- fat-wave, software-pipelined (variant 13) ----------
+// v12's deficit is the post-barrier fragment-read latency: at 1 wave/SIMD
+// there is no co-resident wave to execute while this wave waits on its
+// kk0 ds_reads. v13 pipelines WITHIN the wave: each tile's kk0 fragments
+// are prefetched into registers under the PREVIOUS tile's kk1 MFMA
+// cluster, so every MFMA cluster starts on proven-resident operands.
+// Per tile: stage(t+1) | MFMA kk0 (prefetched) | reads kk1 | vmcnt+bar |
+// MFMA kk1 | prefetch kk0(t+1) | bar. Both barriers land where the waits
+// they imply are already satisfied. sched_barrier(0) fences pin the
+// section order against scheduler hoisting (cf. the kk-split experiment).
+
+__global__ void
+__launch_bounds__(256, 1)
+__attribute__((amdgpu_waves_per_eu(1, 1)))
+gemm_bt_bf16_acc_256w4p_k(float* __restrict__ C,
+                          const bf16* __restrict__ A,
+                          const bf16* __restrict__ Bt, int M, int N, int K,
+                          u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                        // [2][256*64]
+  bf16* Bs = (bf16*)(smem + 2 * BM2 * BKH * 2);  // [2][256*64]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wr = (wave >> 1) * 128, wc = (wave & 1) * 128;
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  const u32 srow = tid >> 3, sslot = tid & 7;
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKH;
+    const bf16* ga = A + (m0 + srow) * (long)K + k0 + src_slot * 8;
+    const bf16* gb = Bt + (n0 + srow) * (long)K + k0 + src_slot * 8;
+    bf16* la = As + buf * BM2 * BKH + (srow * 8u + sslot) * 8u;
+    bf16* lb = Bs + buf * BM2 * BKH + (srow * 8u + sslot) * 8u;
+#pragma unroll 2
+    for (int it = 0; it < 8; ++it) {
+      glds16(ga, la);
+      glds16(gb, lb);
+      ga += 32 * (long)K;
+      gb += 32 * (long)K;
+      la += 32 * 64;
+      lb += 32 * 64;
+    }
+  };
+
+  f32x4 acc[8][8] = {};
+  bf16x8 af0[8], bf0[8];  // kk0 fragments (prefetched a tile ahead)
+  bf16x8 af1[8], bf1[8];  // kk1 fragments
+  const u32 arow_base = wr + (lane & 15);
+  const u32 brow_base = wc + (lane & 15);
+  const u32 kgrp = lane >> 4;
+
+  auto read_kk = [&](int buf, int kk, bf16x8* afp, bf16x8* bfp) {
+    const char* a_base = (const char*)(As + buf * BM2 * BKH);
+    const char* b_base = (const char*)(Bs + buf * BM2 * BKH);
+#pragma unroll
+    for (int m = 0; m < 8; ++m) {
+      u32 row = arow_base + m * 16;
+      u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      afp[m] = *(const bf16x8*)(a_base + byte);
+    }
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      u32 row = brow_base + n * 16;
+      u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      bfp[n] = *(const bf16x8*)(b_base + byte);
+    }
+  };
+
+#define MFMA64(afp, bfp)                                                   \
+  _Pragma("unroll") for (int m = 0; m < 8; ++m)                            \
+      _Pragma("unroll") for (int n = 0; n < 8; ++n)                        \
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(             \
+              afp[m], bfp[n], acc[m][n], 0, 0, 0);
+
+  // prologue: stage tile 0, prove it, prefetch its kk0 fragments
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  read_kk(0, 0, af0, bf0);
+
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    const bool pre = (t + 1 < NT);
+    if (pre) stage(cur ^ 1, t + 1);             // 16 glds
+    __builtin_amdgcn_sched_barrier(0);
+    MFMA64(af0, bf0)                            // kk0, prefetched regs
+    __builtin_amdgcn_sched_barrier(0);
+    read_kk(cur, 1, af1, bf1);                  // kk1 (lands under kk0)
+    __builtin_amdgcn_sched_barrier(0);
+    if (pre) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // stage in
+    __builtin_amdgcn_s_barrier();               // nxt buffer readable
+    MFMA64(af1, bf1)                            // kk1
+    __builtin_amdgcn_sched_barrier(0);
+    if (pre) read_kk(cur ^ 1, 0, af0, bf0);     // prefetch t+1 kk0
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_barrier();               // cur safe to restage
+    cur ^= 1;
+  }
+#undef MFMA64
+
+  const u32 crow0 = wr + (lane >> 4) * 4;
+  const u32 ccol0 = wc + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+    float cv[8][4];
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      const float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cv[n][r] = cp[(long)r * N];
+    }
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        cp[(long)r * N] = cv[n][r] + acc[m][n][r];
+    }
+  }
+}
+
 // ------------------------------------------------------------- launchers --
 
 static int g_gemm_variant = 0;  // 0=auto, 1=simple 256, 2=phased 256, 3=128
@@ -1045,6 +1186,10 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
           M, N, K, g_st_m, g_st_n);
     else if (v == 12)
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256w4_k, grid, dim3(256), lds,
+                         stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+                         K, g_st_m, g_st_n);
+    else if (v == 13)
+      hipLaunchKernelGGL(gemm_bt_bf16_acc_256w4p_k, grid, dim3(256), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
                          K, g_st_m, g_st_n);
     else

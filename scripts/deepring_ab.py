@@ -31,7 +31,7 @@ def screen(n, runs):
     ref = gemm(6, a, bt, n)
     tol = 1e-3 * n ** 0.5
     ok = True
-    for v in (12,):
+    for v in (12, 13):
         for r in range(runs):
             c = gemm(v, a, bt, n)
             err = (c - ref).abs().max().item()
@@ -50,9 +50,9 @@ def perf(n, it, reps):
     a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
     bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
     c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
-    res = {6: [], 12: []}
+    res = {6: [], 12: [], 13: []}
     for rep in range(reps):
-        for v in (6, 12):
+        for v in (6, 12, 13):
             ext.set_gemm_variant(v)
             for _ in range(2):
                 ext.gemm_bt_acc(c, a, bt)
@@ -63,7 +63,7 @@ def perf(n, it, reps):
             torch.cuda.synchronize()
             dt = (time.perf_counter() - t0) / it
             res[v].append(2 * n ** 3 / dt / 1e12)
-    for v in (6, 12):
+    for v in (6, 12, 13):
         xs = res[v]
         print(f"n={n} v{v}: " + " ".join(f"{x:.0f}" for x in xs) +
               f"  best={max(xs):.0f} TF")
