@@ -1037,24 +1037,35 @@ gemm_bt_bf16_acc_256w4p_k(float* __restrict__ C,
   __builtin_amdgcn_s_barrier();
   read_kk(0, 0, af0, bf0);
 
+  // fine interleave directives: alternate {8 MFMA | 2 DS reads | 2 VMEM}
+  // inside each half-tile region so the in-order wave keeps feeding the
+  // MFMA pipe while its loads issue (single-wave: nothing else hides
+  // issue bursts).
+#define SCHED_MIX()                                                        \
+  _Pragma("unroll") for (int g = 0; g < 8; ++g) {                          \
+    __builtin_amdgcn_sched_group_barrier(0x008, 8, 0);                     \
+    __builtin_amdgcn_sched_group_barrier(0x100, 2, 0);                     \
+    __builtin_amdgcn_sched_group_barrier(0x020, 2, 0);                     \
+  }
+
   int cur = 0;
   for (int t = 0; t < NT; ++t) {
     const bool pre = (t + 1 < NT);
     if (pre) stage(cur ^ 1, t + 1);             // 16 glds
-    __builtin_amdgcn_sched_barrier(0);
     MFMA64(af0, bf0)                            // kk0, prefetched regs
-    __builtin_amdgcn_sched_barrier(0);
     read_kk(cur, 1, af1, bf1);                  // kk1 (lands under kk0)
+    SCHED_MIX()
     __builtin_amdgcn_sched_barrier(0);
     if (pre) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // stage in
     __builtin_amdgcn_s_barrier();               // nxt buffer readable
     MFMA64(af1, bf1)                            // kk1
-    __builtin_amdgcn_sched_barrier(0);
     if (pre) read_kk(cur ^ 1, 0, af0, bf0);     // prefetch t+1 kk0
+    SCHED_MIX()
     __builtin_amdgcn_sched_barrier(0);
     __builtin_amdgcn_s_barrier();               // cur safe to restage
     cur ^= 1;
   }
+#undef SCHED_MIX
 #undef MFMA64
 
   const u32 crow0 = wr + (lane >> 4) * 4;
