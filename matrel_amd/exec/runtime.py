@@ -7,7 +7,9 @@ ROCm) over xGMI, one rank per MI355X, plus "gloo" for CPU-only test runs.
 The Runtime owns:
   * the global process group and the row/column sub-groups of the logical
     (Pr x Pc) device grid (SUMMA broadcasts ride on these),
-  * the compute / comm HIP streams of this rank,
+  * the compute stream of this rank (comm/compute overlap rides on
+    ProcessGroupNCCL's internal RCCL stream: async_op collectives enqueue
+    there and wait() fences the compute stream - see exec/summa.py),
   * device placement for this rank.
 
 Single-process use (no init_process_group) is fully supported: world size 1,
@@ -54,7 +56,6 @@ class Runtime:
         self._row_groups: List[Optional[dist.ProcessGroup]] = []
         self._col_groups: List[Optional[dist.ProcessGroup]] = []
         self.compute_stream: Optional[torch.cuda.Stream] = None
-        self.comm_stream: Optional[torch.cuda.Stream] = None
         self._initialized = False
 
     # --- init -----------------------------------------------------------
@@ -90,8 +91,6 @@ class Runtime:
             torch.cuda.set_device(local_rank % torch.cuda.device_count())
             self.device = torch.device("cuda", torch.cuda.current_device())
             self.compute_stream = torch.cuda.current_stream()
-            self.comm_stream = (torch.cuda.Stream()
-                                if cfg.overlap_comm else self.compute_stream)
         else:
             self.device = torch.device("cpu")
 

@@ -523,15 +523,23 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 and bs % 128 == 0)
     fused: Dict[Tuple[int, int], torch.Tensor] = {}
 
+    # overlap_comm=True (default): panels for step k+1 are issued BEFORE
+    # step k's GEMMs launch - RCCL's stream runs them under the compute.
+    # False (deterministic/debug mode): each step's broadcasts are issued
+    # only after the previous step's compute is enqueued (serialized).
+    prefetch = get_config().overlap_comm
     cur = issue_bcast(ks[0])
     for step, k in enumerate(ks):
-        nxt = issue_bcast(ks[step + 1]) if step + 1 < len(ks) else None
+        nxt = (issue_bcast(ks[step + 1])
+               if (step + 1 < len(ks) and prefetch) else None)
         a_rows, a_stack = cur["a_rows"], cur["a_stack"]
         a_sc, a_spinfo, a_dn = cur["a_sc"], cur["a_spinfo"], cur["a_dn"]
         b_cols, b_stack, b_sc = cur["b_cols"], cur["b_stack"], cur["b_sc"]
         for w_ in cur["waits"]:
             w_.wait()
         if not (k in ka and k in kb):
+            if nxt is None and step + 1 < len(ks):
+                nxt = issue_bcast(ks[step + 1])
             cur = nxt
             continue
         for n_j, j in enumerate(b_cols):
@@ -572,6 +580,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                         fused[(i, j)] = ob
                         continue
                 blockops.gemm_bt_acc(c, a_stack[n_i], bt, beta0=first)
+        if nxt is None and step + 1 < len(ks):
+            nxt = issue_bcast(ks[step + 1])
         cur = nxt
 
     for (i, j), c in acc.items():
