@@ -22,6 +22,8 @@
 #define BN 128
 #define BKH 64  // bf16 K-step
 #define BKF 32  // f32  K-step
+#define BM2 256
+#define BN2 256
 
 // ---------------------------------------------------------------- bf16 ----
 // LDS tiles: As[128][64] bf16 (16 KiB), Bs[128][64] bf16, x2 buffers = 64 KiB
@@ -241,14 +243,126 @@ gemm_bt_f32_acc_k(float* __restrict__ C, const float* __restrict__ A,
   }
 }
 
+// ----------------------------------------------------- f32, 256^2 tile ---
+// Exact-f32 on the 256^2/8-wave geometry (the 128^2 f32 kernel runs 2
+// blocks/CU with per-K-tile barriers; the bigger tile halves the barrier
+// rate per MFMA and doubles operand reuse). BK=32: LDS 2 buf x (A+B) x
+// 256x32 f32 = 128 KiB; per-wave output 128x64 (8x4 16x16 frags) on
+// v_mfma_f32_16x16x4_f32 (8 k4-steps per K-tile).
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_f32_acc_256_k(float* __restrict__ C, const float* __restrict__ A,
+                      const float* __restrict__ Bt, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* As = (float*)smem;                        // [2][256*32]
+  float* Bs = (float*)(smem + 2 * BM2 * BKF * 4);  // [2][256*32]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;  // 2 x 4 wave grid; 128x64 out
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 8 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 8, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  // staging: tile row = 32 f32 = 128 B = 8 slots; 256 rows x 8 slots =
+  // 2048 chunks per operand; 512 threads x 4 passes x 2 operands.
+  const u32 srow = tid >> 3, sslot = tid & 7;  // rows 0..63 per pass
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKF;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKF;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      u32 row = srow + it * 64;
+      glds16(A + (m0 + row) * (long)K + k0 + src_slot * 4,
+             As + buf * BM2 * BKF + (row * 8u + sslot) * 4u);
+      glds16(Bt + (n0 + row) * (long)K + k0 + src_slot * 4,
+             Bs + buf * BM2 * BKF + (row * 8u + sslot) * 4u);
+    }
+  };
+
+  f32x4 acc[8][4] = {};
+  const u32 arow_base = wm * 128 + (lane & 15);
+  const u32 brow_base = wn * 64 + (lane & 15);
+  const u32 kgrp = lane >> 4;  // k = k4*4 + kgrp
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);
+    const char* a_base = (const char*)(As + cur * BM2 * BKF);
+    const char* b_base = (const char*)(Bs + cur * BM2 * BKF);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int k4 = 0; k4 < BKF / 4; ++k4) {
+      float a_s[8], b_s[4];
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+        u32 row = arow_base + m * 16;
+        u32 byte = row * 128u + (k4 * 4 + kgrp) * 4u;
+        byte ^= (row & 7) << 4;
+        a_s[m] = *(const float*)(a_base + byte);
+      }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        u32 row = brow_base + n * 16;
+        u32 byte = row * 128u + (k4 * 4 + kgrp) * 4u;
+        byte ^= (row & 7) << 4;
+        b_s[n] = *(const float*)(b_base + byte);
+      }
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_s[m], b_s[n], acc[m][n], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const u32 crow0 = wm * 128 + (lane >> 4) * 4;
+  const u32 ccol0 = wn * 64 + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+    float cv[4][4];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      const float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) cv[n][r] = cp[(long)r * N];
+    }
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long row = m0 + crow0 + m * 16;
+      long col = n0 + ccol0 + n * 16;
+      float* cp = C + row * (long)N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        cp[(long)r * N] = cv[n][r] + acc[m][n][r];
+    }
+  }
+}
+
 // --------------------------------------------------- bf16, 256^2 tile ----
 // 8 waves (2M x 4N), per-wave 128x64 output (8x4 16x16 frags), BK=64,
 // 2 LDS buffers x (A 256x64 + B 256x64) = 128 KiB -> 1 block/CU.
 // The cdna_hip_programming.md §5 table row: "glds, 2 LDS buffers, BK=64,
 // vmcnt(0) + plain __syncthreads()" - the top BK=64 tier for this tile.
-
-#define BM2 256
-#define BN2 256
 
 __global__ void
 __launch_bounds__(512)
@@ -1218,6 +1332,15 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
 extern "C" void launch_gemm_bt_f32_acc(float* C, const float* A,
                                        const float* Bt, int M, int N, int K,
                                        hipStream_t stream) {
+  // 256^2 tile when the shape allows and the grid still fills the chip;
+  // fall back to the 128^2 kernel for small/ragged-tile problems.
+  int nb = (M / BM2) * (N / BN2);
+  if (M % BM2 == 0 && N % BN2 == 0 && nb >= 192) {
+    size_t lds = 2u * BM2 * BKF * 4 * 2;  // 128 KiB
+    hipLaunchKernelGGL(gemm_bt_f32_acc_256_k, dim3((unsigned)nb), dim3(512),
+                       lds, stream, C, A, Bt, M, N, K);
+    return;
+  }
   dim3 grid((M / BM) * (N / BN));
   size_t lds = 2u * BM * BKF * 4 * 2;
   hipLaunchKernelGGL(gemm_bt_f32_acc_k, grid, dim3(256), lds, stream, C, A,
