@@ -16,15 +16,15 @@ def run(n, it=4):
     a = torch.randn(n, n, device="cuda")
     bt = torch.randn(n, n, device="cuda")
     c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
-    ext.gemm_bt_f32_acc(c, a, bt)
+    ext.gemm_bt_acc(c, a, bt)
     ref = (a.double() @ bt.double().t())
     rel = ((c.double() - ref).abs().max() / ref.abs().max()).item()
     for _ in range(2):
-        ext.gemm_bt_f32_acc(c, a, bt)
+        ext.gemm_bt_acc(c, a, bt)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(it):
-        ext.gemm_bt_f32_acc(c, a, bt)
+        ext.gemm_bt_acc(c, a, bt)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / it
     print(f"n={n}: {2*n**3/dt/1e12:.1f} TF  relerr={rel:.2e}")
