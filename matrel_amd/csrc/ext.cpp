@@ -21,6 +21,7 @@ void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
 void set_gemm_variant(int);
 void set_mx8_variant(int);
+void set_f32_variant(int);
 void set_supertile(int, int);
 void launch_ewise_bf16(int, void*, const void*, const void*, long,
                        hipStream_t);
@@ -437,6 +438,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         },
         "last-k fused: bf16 out straight from the accumulate epilogue; "
         "returns false when geometry unsupported (caller falls back)");
+  m.def("set_f32_variant", [](int64_t v) { set_f32_variant((int)v); },
+        "f32 GEMM variant (0 = 32x32x2 default, 1 = 16x16x4 256-tile)");
   m.def("set_mx8_variant", [](int64_t v) { set_mx8_variant((int)v); },
         "mx8 GEMM variant (0 auto, 2 counted 2-phase)");
   m.def("set_gemm_variant", [](int64_t v) { set_gemm_variant((int)v); },

@@ -24,6 +24,7 @@
 #define BKF 32  // f32  K-step
 #define BM2 256
 #define BN2 256
+typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 // ---------------------------------------------------------------- bf16 ----
 // LDS tiles: As[128][64] bf16 (16 KiB), Bs[128][64] bf16, x2 buffers = 64 KiB
@@ -356,6 +357,119 @@ gemm_bt_f32_acc_256_k(float* __restrict__ C, const float* __restrict__ A,
         cp[(long)r * N] = cv[n][r] + acc[m][n][r];
     }
   }
+}
+
+// ------------------------------------ f32, 256^2 tile, 32x32x2 MFMA ------
+// The 16x16x4 f32 MFMA is only ~8 pipe-cycles per instruction - at the
+// 4-cycle wave issue cadence plus fragment reads the f32 kernels above
+// are ISSUE-bound (~73% of the 157 TF f32 rate). v_mfma_f32_32x32x2_f32
+// does 2x the flops per instruction (4096), halving the MFMA instruction
+// count per tile: per-wave 128x64 output = 4x2 32x32 fragments,
+// 16 k2-steps per BK=32 tile, 6 b32 fragment reads per step.
+// C/D map (32x32): col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_f32_acc_256x32_k(float* __restrict__ C, const float* __restrict__ A,
+                         const float* __restrict__ Bt, int M, int N,
+                         int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* As = (float*)smem;                        // [2][256*32]
+  float* Bs = (float*)(smem + 2 * BM2 * BKF * 4);  // [2][256*32]
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;  // 2 x 4 grid; 128x64 per wave
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 8 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 8, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  const u32 srow = tid >> 3, sslot = tid & 7;
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKF;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * BKF;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      u32 row = srow + it * 64;
+      glds16(A + (m0 + row) * (long)K + k0 + src_slot * 4,
+             As + buf * BM2 * BKF + (row * 8u + sslot) * 4u);
+      glds16(Bt + (n0 + row) * (long)K + k0 + src_slot * 4,
+             Bs + buf * BM2 * BKF + (row * 8u + sslot) * 4u);
+    }
+  };
+
+  f32x16 acc[4][2] = {};  // [m-frag 32][n-frag 32]
+  const u32 arow_base = wm * 128 + (lane & 31);
+  const u32 brow_base = wn * 64 + (lane & 31);
+  const u32 kh = lane >> 5;  // k = k2*2 + kh
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);
+    const char* a_base = (const char*)(As + cur * BM2 * BKF);
+    const char* b_base = (const char*)(Bs + cur * BM2 * BKF);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int k2 = 0; k2 < BKF / 2; ++k2) {
+      float a_s[4], b_s[2];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        u32 row = arow_base + m * 32;
+        u32 byte = row * 128u + (k2 * 2 + kh) * 4u;
+        byte ^= (row & 7) << 4;
+        a_s[m] = *(const float*)(a_base + byte);
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        u32 row = brow_base + n * 32;
+        u32 byte = row * 128u + (k2 * 2 + kh) * 4u;
+        byte ^= (row & 7) << 4;
+        b_s[n] = *(const float*)(b_base + byte);
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+              a_s[m], b_s[n], acc[m][n], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: 32x32 C/D map, batched two-pass
+  const long ccol = n0 + wn * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      float cv[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + wm * 128 + m * 32 + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        cv[r] = C[row * (long)N + ccol + n * 32];
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + wm * 128 + m * 32 + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        C[row * (long)N + ccol + n * 32] = cv[r] + acc[m][n][r];
+      }
+    }
 }
 
 // --------------------------------------------------- bf16, 256^2 tile ----
@@ -784,7 +898,6 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
 // (4 kk steps over BK=64). C/D map: col = lane&31,
 // row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
 
-typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 __global__ void
 __launch_bounds__(512)
@@ -1329,6 +1442,9 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
                      (const bf16*)A, (const bf16*)Bt, M, N, K);
 }
 
+static int g_f32_variant = 0;  // 0 = 32x32x2 (default), 1 = 16x16x4 256-tile
+extern "C" void set_f32_variant(int v) { g_f32_variant = v; }
+
 extern "C" void launch_gemm_bt_f32_acc(float* C, const float* A,
                                        const float* Bt, int M, int N, int K,
                                        hipStream_t stream) {
@@ -1337,8 +1453,12 @@ extern "C" void launch_gemm_bt_f32_acc(float* C, const float* A,
   int nb = (M / BM2) * (N / BN2);
   if (M % BM2 == 0 && N % BN2 == 0 && nb >= 192) {
     size_t lds = 2u * BM2 * BKF * 4 * 2;  // 128 KiB
-    hipLaunchKernelGGL(gemm_bt_f32_acc_256_k, dim3((unsigned)nb), dim3(512),
-                       lds, stream, C, A, Bt, M, N, K);
+    if (g_f32_variant == 1)
+      hipLaunchKernelGGL(gemm_bt_f32_acc_256_k, dim3((unsigned)nb),
+                         dim3(512), lds, stream, C, A, Bt, M, N, K);
+    else
+      hipLaunchKernelGGL(gemm_bt_f32_acc_256x32_k, dim3((unsigned)nb),
+                         dim3(512), lds, stream, C, A, Bt, M, N, K);
     return;
   }
   dim3 grid((M / BM) * (N / BN));

@@ -12,7 +12,8 @@ from matrel_amd.ops.backend import require_ext
 ext = require_ext()
 
 
-def run(n, it=4):
+def run(n, it=4, v=0):
+    ext.set_f32_variant(v)
     a = torch.randn(n, n, device="cuda")
     bt = torch.randn(n, n, device="cuda")
     c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
@@ -27,10 +28,12 @@ def run(n, it=4):
         ext.gemm_bt_acc(c, a, bt)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / it
-    print(f"n={n}: {2*n**3/dt/1e12:.1f} TF  relerr={rel:.2e}")
+    print(f"n={n} v{v}: {2*n**3/dt/1e12:.1f} TF  relerr={rel:.2e}")
     del a, bt, c, ref
 
 
 for n in (512, 1024, 4096, 8192, 16384):
-    run(n, it=6 if n <= 8192 else 3)
+    for v in (1, 0):
+        run(n, it=6 if n <= 8192 else 3, v=v)
+ext.set_f32_variant(0)
 print("done")
