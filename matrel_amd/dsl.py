@@ -115,6 +115,11 @@ class Matrix:
     def __truediv__(self, o):
         return self._binary(o, "div")
 
+    def __rtruediv__(self, o):
+        # s / A == s * A**(-1) elementwise
+        return Matrix(L.ScalarOp(
+            L.ScalarOp(self.plan, "pow", -1.0), "mul", float(o)))
+
     def __neg__(self):
         return Matrix(L.ScalarOp(self.plan, "mul", -1.0))
 

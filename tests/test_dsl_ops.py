@@ -195,3 +195,14 @@ def test_block_sparse(session):
     # sum skips absent blocks but matches dense
     assert math.isclose(A.sum().item(), dense(A).sum().item(),
                         rel_tol=1e-3, abs_tol=1e-3)
+
+
+def test_reflected_scalar_ops(session):
+    A = (session.rand(64, 64, seed=9) + 0.5).compute()
+    t = A.to_tensor()
+    for expr, ref in [
+        (2.0 - A, 2.0 - t), (2.0 + A, 2.0 + t), (2.0 * A, 2.0 * t),
+        (-A, -t), (A / 2.0, t / 2.0), (2.0 / A, 2.0 / t),
+        (A ** 2, t ** 2),
+    ]:
+        assert torch.allclose(expr.to_tensor(), ref, atol=1e-4)
