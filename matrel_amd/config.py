@@ -78,6 +78,18 @@ class MatrelConfig:
         cfg = MatrelConfig()
         cfg.block_size = _env_int("MATREL_BLOCK_SIZE", cfg.block_size)
         cfg.dtype = _env_str("MATREL_DTYPE", cfg.dtype)
+        cfg.solve_dtype = _env_str("MATREL_SOLVE_DTYPE", cfg.solve_dtype)
+        cfg.gemm_precision = _env_str("MATREL_GEMM_PRECISION",
+                                      cfg.gemm_precision)
+        st = os.environ.get("MATREL_SPARSE_THRESHOLD")
+        if st:
+            cfg.sparse_threshold = float(st)
+        cfg.overlap_comm = _env_int("MATREL_OVERLAP_COMM",
+                                    int(cfg.overlap_comm)) != 0
+        cfg.deterministic = _env_int("MATREL_DETERMINISTIC",
+                                     int(cfg.deterministic)) != 0
+        cfg.trace = _env_int("MATREL_TRACE", int(cfg.trace)) != 0
+        cfg.trace_path = _env_str("MATREL_TRACE_PATH", cfg.trace_path)
         g = os.environ.get("MATREL_GRID")
         if g:
             pr, pc = g.lower().split("x")

@@ -69,3 +69,20 @@ def test_debug_env(session):
     serialize_kernels(False)
     assert "AMD_SERIALIZE_KERNEL" not in os.environ
     assert isinstance(debug_env_report(), dict)
+
+
+def test_config_from_env(monkeypatch):
+    from matrel_amd.config import MatrelConfig
+    monkeypatch.setenv("MATREL_BLOCK_SIZE", "2048")
+    monkeypatch.setenv("MATREL_GEMM_PRECISION", "mx8")
+    monkeypatch.setenv("MATREL_SPARSE_THRESHOLD", "0.1")
+    monkeypatch.setenv("MATREL_OVERLAP_COMM", "0")
+    monkeypatch.setenv("MATREL_DETERMINISTIC", "1")
+    monkeypatch.setenv("MATREL_GRID", "4x2")
+    cfg = MatrelConfig.from_env()
+    assert cfg.block_size == 2048
+    assert cfg.gemm_precision == "mx8"
+    assert cfg.sparse_threshold == 0.1
+    assert cfg.overlap_comm is False
+    assert cfg.deterministic is True
+    assert cfg.grid == (4, 2)
