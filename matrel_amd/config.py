@@ -72,6 +72,7 @@ class MatrelConfig:
     enable_cse: bool = True
     enable_scheme_selection: bool = True
     enable_scalar_folding: bool = True
+    enable_agg_pushdown: bool = True
 
     @staticmethod
     def from_env() -> "MatrelConfig":

@@ -18,8 +18,8 @@ from __future__ import annotations
 from typing import Dict, List, Tuple
 
 from ..config import get_config
-from .logical import (Ewise, MatMul, MatPower, Plan, ScalarOp, Select,
-                      SelectPredicate, Transpose)
+from .logical import (Agg, Ewise, MatMul, MatPower, Plan, ScalarOp,
+                      Select, SelectPredicate, Transpose)
 
 
 # --------------------------------------------------------------------------
@@ -200,6 +200,53 @@ def push_select(p: Plan) -> Plan:
 
 
 # --------------------------------------------------------------------------
+# pass: aggregate push-down through matmul
+# --------------------------------------------------------------------------
+
+def push_agg(p: Plan) -> Plan:
+    """Rewrite linear aggregations of matrix products so the product is
+    never materialized - MatRel's joint LA/relational optimization
+    (SURVEY.md §2.1 aggregations optimized with the algebra; the classic
+    SystemML/MatRel rewrite family). O(m*k*n) -> O(m*k + k*n):
+
+      sum(A @ B)      = colsums(A) @ rowsums(B)      (1x1)
+      avg(A @ B)      = sum(A @ B) / (m*n)
+      trace(A @ B)    = sum(A * B^T)                 (elementwise)
+      rowsums(A @ B)  = A @ rowsums(B)               (matvec)
+      colsums(A @ B)  = colsums(A) @ B
+      row/col avgs    = the sums scaled by 1/k-extent
+
+    max/min/count/nnz/norm_fro do NOT commute with @ and stay in place.
+    """
+    def rule(n: Plan) -> Plan:
+        if not isinstance(n, Agg) or not isinstance(n.children[0], MatMul):
+            return n
+        mm = n.children[0]
+        a, b = mm.children
+        if n.kind == "sum" and n.axis is None:
+            return Agg(MatMul(Agg(a, "sum", 0), Agg(b, "sum", 1)),
+                       "sum", None)
+        if n.kind == "avg" and n.axis is None:
+            inner = rule(Agg(mm, "sum", None))
+            return ScalarOp(inner, "mul", 1.0 / (mm.nrows * mm.ncols))
+        if n.kind == "trace":
+            # A (m x k) @ B (k x m): tr = sum over A .* B^T
+            return Agg(Ewise(a, Transpose(b), "mul"), "sum", None)
+        if n.kind == "sum" and n.axis == 1:
+            return MatMul(a, Agg(b, "sum", 1))
+        if n.kind == "sum" and n.axis == 0:
+            return MatMul(Agg(a, "sum", 0), b)
+        if n.kind == "avg" and n.axis == 1:
+            return ScalarOp(MatMul(a, Agg(b, "sum", 1)), "mul",
+                            1.0 / mm.ncols)
+        if n.kind == "avg" and n.axis == 0:
+            return ScalarOp(MatMul(Agg(a, "sum", 0), b), "mul",
+                            1.0 / mm.nrows)
+        return n
+    return _rewrite_bottom_up(p, rule)
+
+
+# --------------------------------------------------------------------------
 # pass: matrix-chain reordering (sparsity-aware DP)
 # --------------------------------------------------------------------------
 
@@ -283,6 +330,8 @@ def optimize(p: Plan) -> Plan:
     if cfg.enable_scalar_folding:
         p = fold_scalars(p)
     p = push_select(p)
+    if cfg.enable_agg_pushdown:
+        p = push_agg(p)
     if cfg.enable_chain_reorder:
         p = reorder_chains(p)
     if cfg.enable_scalar_folding:

@@ -174,3 +174,59 @@ def test_supertile_map_bijective():
         seen = {smap(i, ntm, ntn) for i in range(ntm * ntn)}
         assert len(seen) == ntm * ntn
         assert all(0 <= tm < ntm and 0 <= tn < ntn for tm, tn in seen)
+
+
+def test_agg_pushdown_sum_avoids_matmul(session):
+    """sum(A@B) must plan as colsums(A) @ rowsums(B) - the O(n^3) product
+    is never materialized (MatRel's joint LA/relational rewrite family)."""
+    from matrel_amd.plan import logical as L
+    from matrel_amd.plan.optimizer import optimize
+    A = session.randn(96, 64, seed=1)
+    B = session.randn(64, 80, seed=2)
+    opt = optimize((A @ B).sum().plan)
+
+    def mms(n, acc):
+        if isinstance(n, L.MatMul):
+            acc.append((n.children[0].shape, n.children[1].shape))
+        for c in n.children:
+            mms(c, acc)
+        return acc
+    shapes = mms(opt, [])
+    assert all(s[0][0] == 1 or s[1][1] == 1 for s in shapes), shapes
+
+
+def test_agg_pushdown_numeric(session):
+    A = session.randn(96, 64, seed=3)
+    B = session.randn(64, 80, seed=4)
+    C = session.randn(64, 96, seed=5)
+    ta, tb, tc = A.to_tensor(), B.to_tensor(), C.to_tensor()
+    p = ta @ tb
+    assert abs((A @ B).sum().item() - p.sum().item()) <= \
+        1e-3 * abs(p.sum().item()) + 2e-2
+    assert abs((A @ B).avg().item() - p.mean().item()) < 1e-3
+    assert abs((A @ C).trace().item() - (ta @ tc).trace().item()) < 2e-2
+    assert torch.allclose((A @ B).row_sums().to_tensor().reshape(-1),
+                          p.sum(dim=1), atol=1e-2, rtol=1e-3)
+    assert torch.allclose((A @ B).col_sums().to_tensor().reshape(-1),
+                          p.sum(dim=0), atol=1e-2, rtol=1e-3)
+    assert torch.allclose((A @ B).row_avgs().to_tensor().reshape(-1),
+                          p.mean(dim=1), atol=1e-3)
+    assert torch.allclose((A @ B).col_avgs().to_tensor().reshape(-1),
+                          p.mean(dim=0), atol=1e-3)
+    # non-linear aggregates must NOT be rewritten (max of product != any
+    # aggregate of the factors)
+    assert abs((A @ B).max().item() - p.max().item()) < 1e-2
+
+
+def test_agg_pushdown_ablation(session):
+    import matrel_amd as ma
+    A = session.randn(64, 64, seed=6)
+    B = session.randn(64, 64, seed=7)
+    ref = (A @ B).sum().item()
+    cfg = ma.get_config()
+    cfg.enable_agg_pushdown = False
+    try:
+        off = (A @ B).sum().item()
+    finally:
+        cfg.enable_agg_pushdown = True
+    assert abs(ref - off) <= 1e-3 * abs(off) + 2e-2
