@@ -216,12 +216,48 @@ def push_agg(p: Plan) -> Plan:
       colsums(A @ B)  = colsums(A) @ B
       row/col avgs    = the sums scaled by 1/k-extent
 
+    plus the linear-aggregate distributions that skip materializing the
+    operand: agg(X^T) (axis swap / invariance), sum(s*X) = s*sum(X),
+    sum(X+s) = sum(X) + s*extent, sum/avg/trace(A +- B) distribute.
+
     max/min/count/nnz/norm_fro do NOT commute with @ and stay in place.
     """
     def rule(n: Plan) -> Plan:
-        if not isinstance(n, Agg) or not isinstance(n.children[0], MatMul):
+        if not isinstance(n, Agg):
             return n
-        mm = n.children[0]
+        c = n.children[0]
+        # ---- through transpose: full reductions are transpose-invariant;
+        # axis reductions swap axis (rowsums(X^T) = colsums(X)^T)
+        if isinstance(c, Transpose):
+            x = c.children[0]
+            if n.axis is None:
+                return rule(Agg(x, n.kind, None))
+            if n.kind in ("sum", "avg", "min", "max", "count", "nnz"):
+                inner = rule(Agg(x, n.kind, 1 - n.axis))
+                return Transpose(inner)
+        # ---- through scalar ops: sum(s*X) = s*sum(X);
+        # sum(X + s) = sum(X) + s*extent (the shifted matrix is never
+        # materialized); avg analogous
+        if isinstance(c, ScalarOp) and n.kind in ("sum", "avg", "trace"):
+            x = c.children[0]
+            if c.op == "mul":
+                return ScalarOp(rule(Agg(x, n.kind, n.axis)), "mul", c.s)
+            if c.op == "add" and n.kind in ("sum", "avg"):
+                if n.kind == "avg":
+                    return ScalarOp(rule(Agg(x, "avg", n.axis)), "add", c.s)
+                extent = (x.nrows * x.ncols if n.axis is None
+                          else (x.ncols if n.axis == 1 else x.nrows))
+                return ScalarOp(rule(Agg(x, "sum", n.axis)), "add",
+                                c.s * extent)
+        # ---- through elementwise add/sub: linear aggregates distribute
+        if isinstance(c, Ewise) and c.op in ("add", "sub") \
+                and n.kind in ("sum", "avg", "trace"):
+            l = rule(Agg(c.children[0], n.kind, n.axis))
+            r = rule(Agg(c.children[1], n.kind, n.axis))
+            return Ewise(l, r, c.op)
+        if not isinstance(c, MatMul):
+            return n
+        mm = c
         a, b = mm.children
         if n.kind == "sum" and n.axis is None:
             return Agg(MatMul(Agg(a, "sum", 0), Agg(b, "sum", 1)),

@@ -230,3 +230,28 @@ def test_agg_pushdown_ablation(session):
     finally:
         cfg.enable_agg_pushdown = True
     assert abs(ref - off) <= 1e-3 * abs(off) + 2e-2
+
+
+def test_agg_linear_distributions(session):
+    """agg(X^T) axis swap, sum(s*X), sum(X+s), sum/trace over A+-B - the
+    operand is never materialized; numerics must match torch."""
+    A = session.randn(96, 64, seed=11)
+    B = session.randn(96, 64, seed=12)
+    ta, tb = A.to_tensor(), B.to_tensor()
+    assert abs(A.T.sum().item() - ta.sum().item()) < 2e-2
+    assert torch.allclose(A.T.row_sums().to_tensor().reshape(-1),
+                          ta.sum(dim=0), atol=1e-2)
+    assert torch.allclose(A.T.col_maxs().to_tensor().reshape(-1),
+                          ta.max(dim=1).values, atol=1e-4)
+    assert abs((3.0 * A).sum().item() - 3 * ta.sum().item()) < 5e-2
+    assert abs((A + 2.0).sum().item() - (ta + 2).sum().item()) < 5e-1
+    assert torch.allclose((A + 2.0).row_sums().to_tensor().reshape(-1),
+                          (ta + 2).sum(dim=1), atol=1e-1)
+    assert abs((A + 2.0).avg().item() - (ta + 2).mean().item()) < 1e-3
+    assert abs((A + B).sum().item() - (ta + tb).sum().item()) < 5e-2
+    assert torch.allclose((A - B).col_sums().to_tensor().reshape(-1),
+                          (ta - tb).sum(dim=0), atol=2e-2)
+    C = session.randn(64, 64, seed=13)
+    D = session.randn(64, 64, seed=14)
+    assert abs((C - D).trace().item()
+               - (C.to_tensor() - D.to_tensor()).trace().item()) < 2e-2
