@@ -83,6 +83,15 @@ def _worker(rank, world, port, iters=25):
         out = m.compute().to_tensor()
         assert torch.allclose(out, ref, atol=2e-3, rtol=2e-3), \
             f"iter {i}: {spec}"
+        # aggregate of the same random tree: exercises the push_agg
+        # rewrite family (incl. through matmul/transpose/scalar/ewise)
+        got = m.sum().item()
+        want = ref.sum().item()
+        assert abs(got - want) <= 2e-3 * max(1.0, abs(want)) + 2e-2, \
+            f"iter {i} sum: {got} vs {want}: {spec}"
+        rs = m.row_sums().to_tensor().reshape(-1)
+        assert torch.allclose(rs, ref.sum(dim=1), atol=2e-2, rtol=2e-3), \
+            f"iter {i} rowsums: {spec}"
     dist.destroy_process_group()
 
 
