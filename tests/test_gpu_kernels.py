@@ -417,3 +417,20 @@ def test_spmm_wire_views(ext):
         assert err < 1e-3, (h, w, err)
         if h < bs:
             assert c[h:].abs().max().item() == 0.0
+
+
+def test_dsl_agg_pushdown_gpu(gpu_session):
+    """sum/trace/rowsums of A@B on GPU: the push_agg rewrites must agree
+    with the materialized product."""
+    s = gpu_session
+    A = s.randn(1024, 768, seed=31)
+    B = s.randn(768, 512, seed=32)
+    p = (A.to_tensor().float() @ B.to_tensor().float())
+    got = (A @ B).sum().item()
+    assert abs(got - p.sum().item()) <= 1e-3 * abs(p.sum().item()) + 0.5
+    rs = (A @ B).row_sums().to_tensor().reshape(-1).cpu()
+    assert torch.allclose(rs, p.sum(dim=1).cpu(), atol=0.5, rtol=1e-3)
+    C = s.randn(768, 1024, seed=33)
+    tr = (A @ C).trace().item()
+    ref_tr = (A.to_tensor().float() @ C.to_tensor().float()).trace().item()
+    assert abs(tr - ref_tr) <= 1e-3 * abs(ref_tr) + 0.5
