@@ -77,7 +77,19 @@ def _body_agg_2x2(s, tmpdir):
 
 @pytest.mark.parametrize("body", [
     "_body_summa_2x2", "_body_chain_2x2", "_body_sparse_2x2",
-    "_body_agg_2x2",
+    "_body_agg_2x2", "_body_ols_2x2",
 ])
 def test_distributed_world4(body, tmp_path):
     _run(body, tmp_path)
+
+
+def _body_ols_2x2(s, tmpdir):
+    """Config-5 shape class on the 2x2 grid: chain DP + CSE + schemes +
+    gather-solve end-to-end."""
+    X = s.randn(256, 24, seed=51)
+    y = s.randn(256, 1, seed=52)
+    w = ((X.T @ X).inverse() @ (X.T @ y)).compute()
+    import torch as _t
+    xx, yy = X.to_tensor().double(), y.to_tensor().double()
+    ref = _t.linalg.lstsq(xx, yy).solution
+    assert _t.allclose(w.to_tensor().double(), ref, atol=1e-2)
