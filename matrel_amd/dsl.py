@@ -290,6 +290,45 @@ class Matrix:
         result = Executor().execute(opt)
         return Matrix(L.Leaf(result))
 
+    def explain(self, verbose: bool = True) -> str:
+        """Plan inspection (the reference's Spark `explain()` analog):
+        the logical tree, the optimized tree, and per-node physical
+        annotations (partition scheme, matmul algorithm, estimated flops
+        and xGMI bytes). Prints when verbose; always returns the text."""
+        from .plan.physical import plan_physical
+        opt = optimize(self.plan)
+        pp = plan_physical(opt)
+        lines = ["== logical plan ==", self.plan.pretty().rstrip(),
+                 "== optimized plan ==", opt.pretty().rstrip(),
+                 "== physical annotations =="]
+
+        seen = set()
+
+        def walk(n: L.Plan, indent: int) -> None:
+            pi = pp.get(n)
+            extra = f" algo={pi.mm_algo}" if pi.mm_algo else ""
+            stats = ""
+            if pi.est_flops:
+                stats += f" ~{pi.est_flops / 1e9:.2f} GF"
+            if pi.est_comm_bytes:
+                stats += f" ~{pi.est_comm_bytes / 1e6:.2f} MB moved"
+            if id(n) in seen and n.children:
+                # CSE-shared subtree: executes once, memoized
+                lines.append("  " * indent + f"{type(n).__name__} "
+                             f"{n.shape} (cse-shared, see above)")
+                return
+            seen.add(id(n))
+            lines.append("  " * indent + f"{type(n).__name__} "
+                         f"{n.shape} scheme={pi.scheme}{extra}{stats}")
+            for c in n.children:
+                walk(c, indent + 1)
+
+        walk(opt, 0)
+        text = "\n".join(lines)
+        if verbose:
+            print(text)
+        return text
+
     # --- materialized accessors ----------------------------------------
 
     def _materialized(self) -> BlockMatrix:

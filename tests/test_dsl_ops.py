@@ -206,3 +206,13 @@ def test_reflected_scalar_ops(session):
         (A ** 2, t ** 2),
     ]:
         assert torch.allclose(expr.to_tensor(), ref, atol=1e-4)
+
+
+def test_explain(session):
+    X = session.randn(128, 16, seed=21)
+    y = session.randn(128, 1, seed=22)
+    txt = ((X.T @ X).inverse() @ (X.T @ y)).explain(verbose=False)
+    assert "== logical plan ==" in txt
+    assert "== optimized plan ==" in txt
+    assert "algo=" in txt
+    assert "cse-shared" in txt  # the shared X^T must be marked
