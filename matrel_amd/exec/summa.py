@@ -204,12 +204,38 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
     strip_bytes = 4 * bs * n_total
     if strip_bytes * len(my_bis) > 64 << 30:
         return False
+    # Split rows by contribution count. At 1% block density most rows
+    # have exactly ONE contributing A block: those run the fused
+    # beta0+bf16-out GEMM straight into a bf16 strip (no fp32 strip, no
+    # conversion pass - the rocprof profile showed the f32->bf16 slice
+    # conversions costing ~20% of config-4 time). Multi-k rows keep the
+    # fp32-accumulating native batch executor.
+    row_ks = {bi: sorted(k for (i, k) in a.blocks if i == bi)
+              for bi in my_bis}
+    single = [bi for bi in my_bis if len(row_ks[bi]) == 1]
+    multi = [bi for bi in my_bis if len(row_ks[bi]) > 1]
+
+    strips: Dict[int, torch.Tensor] = {}
+    dummy = None
+    for bi in single:
+        k = row_ks[bi][0]
+        ab = a.blocks[(bi, k)].contiguous()
+        bt = _bt_strip(b, k)
+        if dummy is None:
+            dummy = torch.empty(bs, n_total, dtype=torch.float32,
+                                device=out.device)  # never touched (beta0)
+        ob = torch.empty(bs, n_total, dtype=out.dtype, device=out.device)
+        if blockops.gemm_bt_final(dummy, ob, ab, bt, beta0=True):
+            strips[bi] = ob
+        else:
+            multi.append(bi)  # fused geometry unsupported: f32 path
+
     accs, acc_idx = [], {}
     a_list, a_map = [], {}
     bt_list, bt_map = [], {}
     tasks = []
-    for bi in my_bis:
-        ks = sorted(k for (i, k) in a.blocks if i == bi)
+    for bi in multi:
+        ks = row_ks[bi]
         ai = len(accs)
         acc_idx[bi] = ai
         accs.append(torch.empty(bs, n_total, dtype=torch.float32,
@@ -223,18 +249,22 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
                 bt_list.append(_bt_strip(b, k))
             tasks.append((ai, a_map[(bi, k)], bt_map[k],
                           1 if pos == 0 else 0))
-    if not tasks:
-        return True  # nothing to compute; out stays empty (zeros)
-    from ..ops.backend import require_ext
-    require_ext().gemm_bt_batch(accs, a_list, bt_list,
-                                torch.tensor(tasks, dtype=torch.int64))
-    for bi, ai in acc_idx.items():
-        strip = accs[ai]
+    if tasks:
+        from ..ops.backend import require_ext
+        require_ext().gemm_bt_batch(accs, a_list, bt_list,
+                                    torch.tensor(tasks, dtype=torch.int64))
+    for bi in my_bis:
+        fstrip = strips.get(bi)
+        strip = fstrip if fstrip is not None else (
+            accs[acc_idx[bi]] if bi in acc_idx else None)
+        if strip is None:
+            continue
         for bj in range(out.grid_cols):
             if not out.is_local(bi, bj):
                 continue
-            out.blocks[(bi, bj)] = (
-                strip[:, bj * bs:(bj + 1) * bs].to(out.dtype).contiguous())
+            sl = strip[:, bj * bs:(bj + 1) * bs]
+            out.blocks[(bi, bj)] = (sl.contiguous() if fstrip is not None
+                                    else sl.to(out.dtype).contiguous())
     return True
 
 
