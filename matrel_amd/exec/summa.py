@@ -41,11 +41,23 @@ def _pre_t(b: torch.Tensor) -> torch.Tensor:
     return blockops.transpose(b) if b.is_cuda else b.t()
 
 
+def _cache_has_headroom(nbytes: int) -> bool:
+    """Only grow a convenience cache while the device keeps at least 4x
+    the entry size PLUS 16 GiB free - caches are a speedup, not worth an
+    OOM on near-capacity jobs (288 GB fills up fast at 65536^2 blocks)."""
+    try:
+        free, _total = torch.cuda.mem_get_info()
+    except Exception:
+        return True
+    return free > 4 * nbytes + (16 << 30)
+
+
 def _bt_cached(b: BlockMatrix, k: int, bj: int,
                bbd: torch.Tensor) -> torch.Tensor:
     """Transposed-block cache on the B matrix (GPU only): iterative
     workloads (bench steps, power iterations) reuse B across calls, so
-    the transpose is paid once per block ever, not once per multiply."""
+    the transpose is paid once per block ever, not once per multiply.
+    Under memory pressure the transpose is recomputed instead of cached."""
     if not bbd.is_cuda:
         return bbd.t()
     cache = getattr(b, "_bt_cache", None)
@@ -54,7 +66,8 @@ def _bt_cached(b: BlockMatrix, k: int, bj: int,
     bt = cache.get((k, bj))
     if bt is None:
         bt = blockops.transpose(bbd)
-        cache[(k, bj)] = bt
+        if _cache_has_headroom(bt.numel() * bt.element_size()):
+            cache[(k, bj)] = bt
     return bt
 
 
@@ -67,7 +80,9 @@ def _mx8_cached(m: BlockMatrix, key, tensor_bf16: torch.Tensor):
     ent = cache.get(key)
     if ent is None:
         ent = blockops.to_mx8(tensor_bf16)
-        cache[key] = ent
+        if _cache_has_headroom(sum(t.numel() * t.element_size()
+                                   for t in ent)):
+            cache[key] = ent
     return ent
 
 
@@ -152,7 +167,9 @@ def _mx8_padded(m: BlockMatrix, key, t: torch.Tensor, bs: int):
     ent = cache.get(key)
     if ent is None:
         ent = blockops.to_mx8(_pad_t(blockops.densify(t), bs))
-        cache[key] = ent
+        if _cache_has_headroom(sum(x.numel() * x.element_size()
+                                   for x in ent)):
+            cache[key] = ent
     return ent
 
 
@@ -168,7 +185,9 @@ def _bt_strip(b: BlockMatrix, k: int) -> torch.Tensor:
                      b.get_block_dense(k, bj)))
                  for bj in range(b.grid_cols)]
         s = torch.cat(parts, dim=0).contiguous()
-        cache[k] = s
+        if not s.is_cuda or _cache_has_headroom(
+                s.numel() * s.element_size()):
+            cache[k] = s
     return s
 
 
