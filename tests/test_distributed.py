@@ -126,6 +126,16 @@ def _body_elem_sparse_summa(s, tmpdir):
     assert torch.allclose(out.to_dense(), ref, atol=1e-3)
     rep = metrics.report()
     assert "summa_a_bcast_csr" in rep["collectives"]
+    # nnz-proportional: the CSR wire must move far fewer bytes than the
+    # densified panels would (2% dense blocks -> ~10x+ reduction incl.
+    # index overhead)
+    csr_bytes = rep["collectives"]["summa_a_bcast_csr"]["bytes"]
+    n_sparse_blocks = len(A.blocks)
+    dense_equiv = 0
+    for (bi, bk) in A.blocks:
+        h, w = A.block_shape(bi, bk)
+        dense_equiv += h * w * A.blocks[(bi, bk)].element_size()
+    assert csr_bytes < 0.5 * dense_equiv, (csr_bytes, dense_equiv)
     # mixed dense/sparse rows + ragged edges
     A2 = BlockMatrix.sparse_random(100, 88, 0.03, seed=23)
     d = torch.randn(A2.block_shape(0, 0))
