@@ -57,12 +57,29 @@ def save_matrix(m: BlockMatrix, path: str) -> None:
     for (bi, bj), t in m.local_items():
         if m.owner(bi, bj) == -1 and rt.rank != 0:
             continue  # replicated blocks written once
+        fn = _block_file(bi, bj)
         if t.layout == torch.sparse_csr:
-            t = t.to_dense()
+            # CSR on disk: [crow int32 (h+1) | col int32 (nnz) | values]
+            # in one file - nnz-proportional bytes (the reference's
+            # SparseMatrix serialization analog); offsets derive from
+            # shape+nnz, so the manifest needs no extra fields.
+            h, w = t.shape
+            nnz = int(t._nnz())
+            with open(os.path.join(path, fn), "wb") as f:
+                t.crow_indices().cpu().to(torch.int32).numpy().tofile(f)
+                t.col_indices().cpu().to(torch.int32).numpy().tofile(f)
+                vals = t.values().detach().cpu().contiguous()
+                raw = (vals.view(torch.uint16)
+                       if vals.dtype == torch.bfloat16 else vals)
+                raw.numpy().tofile(f)
+            entries.append({
+                "bi": bi, "bj": bj, "file": fn, "layout": "csr",
+                "shape": [int(h), int(w)], "nnz": nnz,
+            })
+            continue
         cpu = t.detach().cpu().contiguous()
         raw = cpu.view(torch.uint16) if cpu.dtype == torch.bfloat16 else cpu
         arr = raw.numpy()
-        fn = _block_file(bi, bj)
         arr.tofile(os.path.join(path, fn))
         entries.append({
             "bi": bi, "bj": bj, "file": fn,
@@ -104,10 +121,27 @@ def load_matrix(path: str, scheme: Optional[Scheme] = None,
         bi, bj = e["bi"], e["bj"]
         if not out.is_local(bi, bj):
             continue
+        fp = os.path.join(path, e["file"])
         np_dt = _NP_DTYPE[dtype]
-        arr = np.memmap(os.path.join(path, e["file"]), dtype=np_dt,
-                        mode="r") if mmap else \
-            np.fromfile(os.path.join(path, e["file"]), dtype=np_dt)
+        if e.get("layout") == "csr":
+            h, w = e["shape"]
+            nnz = e["nnz"]
+            raw = np.fromfile(fp, dtype=np.uint8)
+            crow = torch.from_numpy(
+                raw[:4 * (h + 1)].view(np.int32).copy())
+            cols = torch.from_numpy(
+                raw[4 * (h + 1):4 * (h + 1 + nnz)].view(np.int32).copy())
+            vb = raw[4 * (h + 1 + nnz):].view(np_dt).copy()
+            vals = torch.from_numpy(vb)
+            if dtype == torch.bfloat16:
+                vals = vals.view(torch.bfloat16)
+            t = torch.sparse_csr_tensor(
+                crow.to(out.device), cols.to(out.device),
+                vals.to(device=out.device, dtype=dtype), size=(h, w))
+            out.blocks[(bi, bj)] = t
+            continue
+        arr = np.memmap(fp, dtype=np_dt, mode="r") if mmap else \
+            np.fromfile(fp, dtype=np_dt)
         t = torch.from_numpy(np.ascontiguousarray(arr)).reshape(e["shape"])
         if dtype == torch.bfloat16:
             t = t.view(torch.bfloat16)

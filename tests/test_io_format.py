@@ -78,3 +78,26 @@ def test_coo_roundtrip(session, tmp_path):
     A.save_coo(p)
     B = session.load_coo(p, 60, 40)
     assert torch.equal(B.to_tensor(), t)
+
+
+def test_save_load_elem_sparse_csr(session, tmp_path):
+    """CSR-stored blocks go to disk in CSR layout (nnz-proportional
+    bytes) and come back CSR-stored with exact values."""
+    import os
+    A = session.sparse_random(300, 260, 0.02, seed=8)
+    bm = A.blocks
+    n_csr = sum(1 for _, t in bm.local_items()
+                if t.layout == torch.sparse_csr)
+    assert n_csr > 0
+    path = str(tmp_path / "csr_mat")
+    A.save(path)
+    # on-disk bytes must be far below the dense footprint
+    disk = sum(os.path.getsize(os.path.join(path, "blocks", f))
+               for f in os.listdir(os.path.join(path, "blocks")))
+    dense_bytes = 300 * 260 * 4
+    assert disk < 0.4 * dense_bytes, (disk, dense_bytes)
+    B = session.load(path)
+    n_csr2 = sum(1 for _, t in B.blocks.local_items()
+                 if t.layout == torch.sparse_csr)
+    assert n_csr2 == n_csr
+    assert torch.allclose(A.to_tensor(), B.to_tensor(), atol=0)
