@@ -372,6 +372,12 @@ def test_strip_mode_sparse_spmm():
     C = (A @ B).compute()
     ref = A.to_tensor() @ B.to_tensor()
     assert (C.to_tensor() - ref).abs().max().item() < 1.0
+    # low density: most rows have ONE contributing block -> the fused
+    # beta0+bf16-out single-row strip path
+    A2 = s.block_sparse_random(4096, 4096, 0.02, seed=72)
+    C2 = (A2 @ B).compute()
+    ref2 = A2.to_tensor() @ B.to_tensor()
+    assert (C2.to_tensor() - ref2).abs().max().item() < 1.0
     reset_runtime()
 
 
