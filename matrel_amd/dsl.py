@@ -211,6 +211,14 @@ class Matrix:
     def norm_fro(self) -> "Matrix":
         return self._agg("norm_fro", None)
 
+    def row_norms(self) -> "Matrix":
+        """Per-row L2 norms (m x 1): sqrt(rowsums(A*A))."""
+        return ((self * self).row_sums()) ** 0.5
+
+    def col_norms(self) -> "Matrix":
+        """Per-column L2 norms (1 x n)."""
+        return ((self * self).col_sums()) ** 0.5
+
     def row_sums(self) -> "Matrix":
         return self._agg("sum", 1)
 

@@ -216,3 +216,12 @@ def test_explain(session):
     assert "== optimized plan ==" in txt
     assert "algo=" in txt
     assert "cse-shared" in txt  # the shared X^T must be marked
+
+
+def test_row_col_norms(session):
+    A = session.randn(96, 64, seed=31)
+    t = A.to_tensor()
+    rn = A.row_norms().to_tensor().reshape(-1)
+    assert torch.allclose(rn, t.norm(dim=1), atol=1e-3)
+    cn = A.col_norms().to_tensor().reshape(-1)
+    assert torch.allclose(cn, t.norm(dim=0), atol=1e-3)
