@@ -15,9 +15,10 @@ import matrel_amd as ma  # noqa
 from matrel_amd.ops.backend import require_ext  # noqa
 
 require_ext()
-N = 600  # with block 256: 3x3 grid, ragged edges (600 = 2*256 + 88)
+N = int(os.environ.get("GPU_FUZZ_N", "600"))
+BS = int(os.environ.get("GPU_FUZZ_BS", "256"))
 PREC = os.environ.get("GPU_FUZZ_PRECISION", "bf16")
-ma.set_config(ma.MatrelConfig(block_size=256, gemm_precision=PREC))
+ma.set_config(ma.MatrelConfig(block_size=BS, gemm_precision=PREC))
 s = ma.MatrelSession()
 # fp8 accumulates larger relative error through deep trees
 ATOL, RTOL = (0.15, 0.05) if PREC == "bf16" else (1.5, 0.2)
