@@ -34,13 +34,6 @@ def _presence_map(m: BlockMatrix) -> torch.Tensor:
     return m.presence()
 
 
-def _pre_t(b: torch.Tensor) -> torch.Tensor:
-    """B operand for the B^T GEMM kernel: materialized transpose on GPU
-    (the bandwidth-bound HIP transpose, amortized across every A-row block
-    that reuses it), a free .t() view on CPU."""
-    return blockops.transpose(b) if b.is_cuda else b.t()
-
-
 def _cache_has_headroom(nbytes: int) -> bool:
     """Only grow a convenience cache while the device keeps at least 4x
     the entry size PLUS 16 GiB free - caches are a speedup, not worth an
