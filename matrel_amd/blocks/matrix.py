@@ -194,9 +194,6 @@ class BlockMatrix:
 
     # --- stats (feed the optimizer's sparsity-aware cost model) ---------
 
-    def local_nnz_blocks(self) -> int:
-        return len(self.blocks)
-
     def block_density(self) -> float:
         """Fraction of blocks present, across all ranks."""
         total = self.grid_rows * self.grid_cols
