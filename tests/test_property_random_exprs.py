@@ -102,7 +102,7 @@ def _build(spec, leaves):
     raise ValueError(kind)
 
 
-@settings(max_examples=150, deadline=None)
+@settings(max_examples=150, deadline=None, derandomize=True)
 @given(spec=_expr(4))
 def test_random_expression_matches_torch(spec, _session):
     leaves = _leaves(_session)
@@ -113,7 +113,7 @@ def test_random_expression_matches_torch(spec, _session):
         f"{(out - ref).abs().max().item():.3e}"
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60, deadline=None, derandomize=True)
 @given(spec=_expr(3),
        agg=st.sampled_from(["sum", "max", "min", "row_sums", "col_sums"]))
 def test_random_expression_aggregates(spec, agg, _session):
