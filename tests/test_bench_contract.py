@@ -45,3 +45,21 @@ def test_bench_main_world2(tmp_path):
     assert d["metric"] == "dense_matmul_tflops"
     assert d["data"] == "synthetic"
     assert d["config"]["parallelism"].startswith("summa_grid")
+
+
+def test_bench_single_process(tmp_path):
+    import subprocess
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+                        "MASTER_PORT")}
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--steps", "1",
+         "--warmup", "0", "--n-size", "256"],
+        capture_output=True, text=True, cwd=root, env=env, timeout=300)
+    assert out.returncode == 0, out.stderr[-800:]
+    lines = [ln for ln in out.stdout.strip().splitlines()
+             if ln.startswith("{")]
+    assert len(lines) == 1
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 1 and d["metric"] == "dense_matmul_tflops"
