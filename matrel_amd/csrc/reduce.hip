@@ -135,7 +135,24 @@ __global__ void reduce_cols_stage1_k(float* __restrict__ partials,
 #pragma unroll
   for (int j = 0; j < V; ++j) acc[j] = rk_init(KIND);
   if (c0 + V <= cols) {
-    for (long r = r0; r < r1; ++r) {
+    // 4-deep row unroll: keeps 4 independent 16-B loads in flight per
+    // thread (single-load chains leave the column reduction latency-
+    // bound at ~half of HBM bandwidth)
+    long r = r0;
+    for (; r + 4 <= r1; r += 4) {
+      vecT v0 = *(const vecT*)(a + (r + 0) * cols + c0);
+      vecT v1 = *(const vecT*)(a + (r + 1) * cols + c0);
+      vecT v2 = *(const vecT*)(a + (r + 2) * cols + c0);
+      vecT v3 = *(const vecT*)(a + (r + 3) * cols + c0);
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        acc[j] = rk_comb(KIND, acc[j], rk_map(KIND, (float)v0[j]));
+        acc[j] = rk_comb(KIND, acc[j], rk_map(KIND, (float)v1[j]));
+        acc[j] = rk_comb(KIND, acc[j], rk_map(KIND, (float)v2[j]));
+        acc[j] = rk_comb(KIND, acc[j], rk_map(KIND, (float)v3[j]));
+      }
+    }
+    for (; r < r1; ++r) {
       vecT v = *(const vecT*)(a + r * cols + c0);
 #pragma unroll
       for (int j = 0; j < V; ++j)
