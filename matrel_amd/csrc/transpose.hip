@@ -32,10 +32,10 @@ __global__ void transpose_k(T* __restrict__ out, const T* __restrict__ in,
     long c = c0 + tx * 8;
     if (r < rows) {
       if (c + 8 <= cols) {
+        // vector LDS store: the dst (row*TPAD + tx*8) elements are
+        // contiguous and 16-B aligned (TPAD*sizeof(T) is a 16-multiple)
         V8 v = *(const V8*)(in + r * cols + c);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          tile[(ty + h * 32) * TPAD + tx * 8 + j] = ((const T*)&v)[j];
+        *(V8*)(tile + (ty + h * 32) * TPAD + tx * 8) = v;
       } else {
         for (int j = 0; j < 8 && c + j < cols; ++j)
           tile[(ty + h * 32) * TPAD + tx * 8 + j] = in[r * cols + c + j];
