@@ -11,18 +11,11 @@
 #include "common.h"
 
 #define TT 64
-// LDS row stride in elements, per dtype: the transpose-side reads walk
-// LDS at TPAD-element stride, so TPAD*sizeof(T) must be an ODD number of
-// dwords to spread the 64 lanes over all 64 banks (72 elems = 36 dwords
-// for bf16 hit 2-banks-in-4 patterns). bf16: 66 elems = 33 dwords; f32:
-// 65 elems = 65 dwords.
-template <typename T>
-constexpr int tpad() { return sizeof(T) == 2 ? 66 : 65; }
+#define TPAD 72  // lds row stride in elements
 
 template <typename T, typename V8>
 __global__ void transpose_k(T* __restrict__ out, const T* __restrict__ in,
                             long rows, long cols) {
-  constexpr int TPAD = tpad<T>();
   __shared__ T tile[TT * TPAD];
   // block covers input tile [r0, r0+64) x [c0, c0+64)
   long tr = blockIdx.x / ((cols + TT - 1) / TT);
@@ -39,10 +32,10 @@ __global__ void transpose_k(T* __restrict__ out, const T* __restrict__ in,
     long c = c0 + tx * 8;
     if (r < rows) {
       if (c + 8 <= cols) {
+        // vector LDS store: the dst (row*TPAD + tx*8) elements are
+        // contiguous and 16-B aligned (TPAD*sizeof(T) is a 16-multiple)
         V8 v = *(const V8*)(in + r * cols + c);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          tile[(ty + h * 32) * TPAD + tx * 8 + j] = ((const T*)&v)[j];
+        *(V8*)(tile + (ty + h * 32) * TPAD + tx * 8) = v;
       } else {
         for (int j = 0; j < 8 && c + j < cols; ++j)
           tile[(ty + h * 32) * TPAD + tx * 8 + j] = in[r * cols + c + j];
