@@ -1,0 +1,40 @@
+"""World-size-8 gloo test: the 4x2 grid is EXACTLY the shape the driver's
+8-GPU SCALE run uses (one rank per MI355X over RCCL). One spawn, one
+combined body, to keep suite time bounded."""
+
+import os
+import sys
+
+import torch
+import torch.multiprocessing as mp
+
+
+def _worker(rank: int, world: int, port: int) -> None:
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      HSA_ENABLE_IPC_MODE_LEGACY="0")
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import matrel_amd as ma
+    ma.set_config(ma.MatrelConfig(block_size=32))
+    s = ma.MatrelSession()
+    assert s.runtime.grid == (4, 2), s.runtime.grid
+    # dense SUMMA on the 4x2 grid (bench shape class, ragged)
+    A = s.randn(160, 144, seed=81)
+    B = s.randn(144, 96, seed=82)
+    C = (A @ B).compute()
+    ref = A.to_tensor() @ B.to_tensor()
+    assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
+    # element-sparse + transpose + aggregate through the same grid
+    E = s.sparse_random(144, 160, 0.03, seed=83)
+    D = (E.T @ A).compute()
+    refd = E.to_tensor().t() @ A.to_tensor()
+    assert torch.allclose(D.to_tensor(), refd, atol=1e-3)
+    assert abs((A @ B).sum().item() - ref.sum().item()) < 1e-1
+    dist.destroy_process_group()
+
+
+def test_distributed_world8():
+    mp.spawn(_worker, args=(8, 29931), nprocs=8, join=True)
