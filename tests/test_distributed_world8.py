@@ -28,7 +28,7 @@ def _worker(rank: int, world: int, port: int) -> None:
     ref = A.to_tensor() @ B.to_tensor()
     assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
     # element-sparse + transpose + aggregate through the same grid
-    E = s.sparse_random(144, 160, 0.03, seed=83)
+    E = s.sparse_random(160, 128, 0.03, seed=83)
     D = (E.T @ A).compute()
     refd = E.to_tensor().t() @ A.to_tensor()
     assert torch.allclose(D.to_tensor(), refd, atol=1e-3)
