@@ -357,7 +357,13 @@ class Executor:
         sdt = (torch.float64 if get_config().solve_dtype == "fp64"
                else torch.float32)
         a_full = self._gather_dense(n.children[0]).to(sdt)
-        x = torch.linalg.inv(a_full)
+        # SPD fast path (X^T X of the OLS pipeline): Cholesky inverse -
+        # rocprof showed LU (rocsolver getf2) at ~34% of config-5 time.
+        try:
+            ch = torch.linalg.cholesky(a_full)
+            x = torch.cholesky_inverse(ch)
+        except Exception:
+            x = torch.linalg.inv(a_full)
         ref = self._eval(n.children[0])
         return _vector_result(x.float(), ref, tuple(x.shape))
 
