@@ -37,6 +37,57 @@ def _densify(t: torch.Tensor) -> torch.Tensor:
 BlockIndex = Tuple[int, int]
 
 
+class _BlockStore(dict):
+    """Per-matrix block dict that invalidates the matrix's cached global
+    stats (presence / nnz maps) and derived-panel caches on ANY mutation.
+
+    Much of the engine writes `m.blocks[idx] = t` directly; before this
+    existed only `set_block` invalidated, so a stale nnz map could make the
+    SUMMA CSR-wire encoder size broadcast buffers from outdated values
+    (ADVICE.md round 1). Now every mutation path is safe.
+    """
+
+    __slots__ = ("_owner",)
+
+    def __init__(self, owner: "BlockMatrix", *args) -> None:
+        super().__init__(*args)
+        self._owner = owner
+
+    def _inv(self) -> None:
+        self._owner._invalidate_stats()
+
+    def __setitem__(self, k, v) -> None:
+        super().__setitem__(k, v)
+        self._inv()
+
+    def __delitem__(self, k) -> None:
+        super().__delitem__(k)
+        self._inv()
+
+    def pop(self, *args):
+        r = super().pop(*args)
+        self._inv()
+        return r
+
+    def popitem(self):
+        r = super().popitem()
+        self._inv()
+        return r
+
+    def update(self, *args, **kw) -> None:
+        super().update(*args, **kw)
+        self._inv()
+
+    def clear(self) -> None:
+        super().clear()
+        self._inv()
+
+    def setdefault(self, k, default=None):
+        r = super().setdefault(k, default)
+        self._inv()
+        return r
+
+
 def _torch_dtype(name: str) -> torch.dtype:
     return {
         "bf16": torch.bfloat16,
@@ -74,8 +125,19 @@ class BlockMatrix:
             dtype = _torch_dtype(cfg.dtype) if self.device.type == "cuda" \
                 else torch.float32
         self.dtype = dtype
-        # Local (this-rank) blocks only.
-        self.blocks: Dict[BlockIndex, torch.Tensor] = {}
+        # Local (this-rank) blocks only. _BlockStore invalidates the
+        # cached presence/nnz maps on every mutation.
+        self._presence = None
+        self._nnzmap = None
+        self.blocks: Dict[BlockIndex, torch.Tensor] = _BlockStore(self)
+
+    def _invalidate_stats(self) -> None:
+        self._presence = None
+        self._nnzmap = None
+        # derived-panel caches keyed by block index are stale too
+        for attr in ("_bt_cache", "_bt_strip_cache", "_mx8_cache"):
+            if getattr(self, attr, None):
+                setattr(self, attr, {})
 
     # --- geometry -------------------------------------------------------
 
@@ -137,8 +199,6 @@ class BlockMatrix:
         assert t.shape == self.block_shape(bi, bj), \
             f"block {(bi, bj)}: {tuple(t.shape)} != {self.block_shape(bi, bj)}"
         self.blocks[(bi, bj)] = t.to(device=self.device, dtype=self.dtype)
-        self._presence = None  # invalidate cached presence map
-        self._nnzmap = None
 
     def coll_device(self) -> torch.device:
         """Device for small bookkeeping collectives: RCCL needs GPU
@@ -343,7 +403,7 @@ class BlockMatrix:
         out = BlockMatrix(self.nrows, self.ncols, self.block_size, scheme,
                           self.dtype, self.device)
         if not rt.is_distributed:
-            out.blocks = dict(self.blocks)
+            out.blocks.update(self.blocks)
             return out
         pm = self.presence()
         for (bi, bj) in self.block_indices():

@@ -48,6 +48,9 @@ class MatrelConfig:
     dtype: str = "bf16"
     # dtype used for reductions / solves that need extra precision.
     solve_dtype: str = "fp32"
+    # Solve/Inverse gather their operand DENSE to every rank (the OLS k x k
+    # Gram matrix). Refuse silently-catastrophic gathers above this side.
+    solve_max_side: int = 32768
     # GEMM compute precision: "bf16" (default) or "mx8" (MX-scaled fp8
     # e4m3 with per-32-element e8m0 block scales - ~2x the bf16 MFMA rate
     # at ~2-3 decimal digits; opt-in).
@@ -80,6 +83,8 @@ class MatrelConfig:
         cfg.block_size = _env_int("MATREL_BLOCK_SIZE", cfg.block_size)
         cfg.dtype = _env_str("MATREL_DTYPE", cfg.dtype)
         cfg.solve_dtype = _env_str("MATREL_SOLVE_DTYPE", cfg.solve_dtype)
+        cfg.solve_max_side = _env_int("MATREL_SOLVE_MAX_SIDE",
+                                      cfg.solve_max_side)
         cfg.gemm_precision = _env_str("MATREL_GEMM_PRECISION",
                                       cfg.gemm_precision)
         st = os.environ.get("MATREL_SPARSE_THRESHOLD")

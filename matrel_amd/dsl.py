@@ -21,7 +21,6 @@ import torch
 from .blocks.matrix import BlockMatrix
 from .plan import logical as L
 from .plan.optimizer import optimize
-from .plan.physical import plan_physical
 
 Number = Union[int, float]
 
@@ -262,27 +261,6 @@ class Matrix:
 
     def optimized_plan(self) -> L.Plan:
         return optimize(self.plan)
-
-    def explain(self) -> str:
-        """Optimizer's plan for this expression (like Spark explain())."""
-        opt = self.optimized_plan()
-        pp = plan_physical(opt)
-        lines = ["== optimized logical/physical plan =="]
-
-        def walk(n: L.Plan, d: int, seen):
-            pi = pp.get(n)
-            mark = " (shared)" if id(n) in seen else ""
-            seen.add(id(n))
-            extra = getattr(n, "_pretty_extra", lambda: "")()
-            algo = f" algo={pi.mm_algo}" if pi.mm_algo else ""
-            lines.append("  " * d + f"{type(n).__name__}{extra} {n.shape} "
-                         f"scheme={pi.scheme.kind}{algo}"
-                         f" d={n.density:.3g}{mark}")
-            if not mark:
-                for c in n.children:
-                    walk(c, d + 1, seen)
-        walk(opt, 0, set())
-        return "\n".join(lines)
 
     def compile(self):
         """Capture this expression's kernel sequence in a hipGraph and

@@ -151,7 +151,14 @@ class Executor:
         if a.block_size != b.block_size:
             raise ValueError("elementwise op needs equal block sizes")
         if a.scheme != b.scheme:
-            b = b.repartition(a.scheme)
+            # Follow the planner's move decision (physical.py Ewise rule:
+            # align to the BIGGER operand's scheme so the smaller/sparser
+            # side is the one that crosses xGMI).
+            tgt = self.physical.get(n).scheme
+            if a.scheme != tgt:
+                a = a.repartition(tgt)
+            if b.scheme != tgt:
+                b = b.repartition(tgt)
         out = BlockMatrix(a.nrows, a.ncols, a.block_size, a.scheme,
                           a.dtype, a.device)
         op = {"add": blockops.EW_ADD, "sub": blockops.EW_SUB,
@@ -294,10 +301,15 @@ class Executor:
                         else torch.minimum(init, v))
             if kind in ("max", "min"):
                 if init is None:
-                    # all-zero / empty local part: identity is 0 only if a
-                    # zero entry exists; absent blocks are zeros, so 0 is a
-                    # valid candidate whenever any block is absent.
-                    init = torch.zeros(1, dtype=torch.float32, device=dev)
+                    # This rank owns NO block indices at all (more grid
+                    # rows/cols than block rows/cols under this scheme).
+                    # The identity must be -/+inf: 0.0 would fabricate a
+                    # candidate and win on an all-negative (max) /
+                    # all-positive (min) matrix. Absent-but-owned blocks
+                    # are already densified to zeros by items().
+                    ident = -math.inf if kind == "max" else math.inf
+                    init = torch.full((1,), ident, dtype=torch.float32,
+                                      device=dev)
                 partial = init.reshape(1)
             if rt.is_distributed and not a.scheme.is_replicated():
                 op = {"max": dist.ReduceOp.MAX,
@@ -332,7 +344,15 @@ class Executor:
         shape = (a.nrows, 1) if n.axis == 1 else (1, a.ncols)
         return _vector_result(vec.reshape(shape), a, shape)
 
-    def _gather_dense(self, n: Plan) -> torch.Tensor:
+    def _gather_dense(self, n: Plan, what: str = "operand") -> torch.Tensor:
+        from ..config import get_config
+        limit = get_config().solve_max_side
+        if max(n.shape) > limit:
+            raise ValueError(
+                f"{what} gathers the full {n.shape[0]}x{n.shape[1]} matrix "
+                f"dense to every rank; the configured limit is "
+                f"{limit}x{limit} (MATREL_SOLVE_MAX_SIDE). Solve/Inverse "
+                f"are meant for small (e.g. Gram-matrix) operands.")
         m = self._eval(n)
         return m.to_dense()
 
@@ -340,14 +360,20 @@ class Executor:
         from ..config import get_config
         sdt = (torch.float64 if get_config().solve_dtype == "fp64"
                else torch.float32)
-        a_full = self._gather_dense(n.children[0]).to(sdt)
-        b_full = self._gather_dense(n.children[1]).to(sdt)
+        a_full = self._gather_dense(n.children[0], "solve").to(sdt)
+        b_full = self._gather_dense(n.children[1], "solve rhs").to(sdt)
         # SPD fast path (X^T X of the OLS pipeline): Cholesky, fall back to
-        # LU for general A (solve in fp32/fp64).
-        try:
-            ch = torch.linalg.cholesky(a_full)
-            x = torch.cholesky_solve(b_full, ch)
-        except Exception:
+        # LU for general A (solve in fp32/fp64). cholesky
+        # reads only the lower triangle, so gate it on exact symmetry -
+        # a non-symmetric A with an SPD lower triangle would otherwise
+        # silently solve the wrong system.
+        if _is_symmetric(a_full):
+            try:
+                ch = torch.linalg.cholesky(a_full)
+                x = torch.cholesky_solve(b_full, ch)
+            except Exception:
+                x = torch.linalg.solve(a_full, b_full)
+        else:
             x = torch.linalg.solve(a_full, b_full)
         ref = self._eval(n.children[0])
         return _vector_result(x.float(), ref, tuple(x.shape))
@@ -356,13 +382,17 @@ class Executor:
         from ..config import get_config
         sdt = (torch.float64 if get_config().solve_dtype == "fp64"
                else torch.float32)
-        a_full = self._gather_dense(n.children[0]).to(sdt)
+        a_full = self._gather_dense(n.children[0], "inverse").to(sdt)
         # SPD fast path (X^T X of the OLS pipeline): Cholesky inverse -
         # rocprof showed LU (rocsolver getf2) at ~34% of config-5 time.
-        try:
-            ch = torch.linalg.cholesky(a_full)
-            x = torch.cholesky_inverse(ch)
-        except Exception:
+        # Symmetry-gated for the same reason as _solve.
+        if _is_symmetric(a_full):
+            try:
+                ch = torch.linalg.cholesky(a_full)
+                x = torch.cholesky_inverse(ch)
+            except Exception:
+                x = torch.linalg.inv(a_full)
+        else:
             x = torch.linalg.inv(a_full)
         ref = self._eval(n.children[0])
         return _vector_result(x.float(), ref, tuple(x.shape))
@@ -401,6 +431,11 @@ class Executor:
 
 
 # --------------------------------------------------------------------------
+
+def _is_symmetric(a: torch.Tensor) -> bool:
+    return (a.dim() == 2 and a.shape[0] == a.shape[1]
+            and torch.equal(a, a.mT))
+
 
 def _axis_identity(kind: str, length: int, dev) -> torch.Tensor:
     if kind == "max":

@@ -236,6 +236,54 @@ def _body_partition_aware_local(s, tmpdir):
     assert torch.allclose(C.to_tensor(), ref, atol=1e-2)
 
 
+def _body_maxmin_empty_rank(s, tmpdir):
+    """Regression: full max/min on a matrix with
+    fewer block rows than grid rows - the rank owning ZERO blocks must
+    contribute the -/+inf identity, not 0.0, or an all-negative matrix
+    reports max=0."""
+    # block_size=32, world=2 -> grid (2,1); a 32x64 matrix has 1 block row,
+    # so grid-row-1's rank owns nothing under GRID.
+    neg = s.from_dense(-torch.rand(32, 64) - 1.0)  # all in [-2, -1]
+    m = neg.max().item()
+    assert -2.0 <= m <= -1.0, f"max of all-negative matrix was {m}"
+    pos = s.from_dense(torch.rand(32, 64) + 1.0)   # all in [1, 2]
+    mn = pos.min().item()
+    assert 1.0 <= mn <= 2.0, f"min of all-positive matrix was {mn}"
+
+
+def _body_ewise_move_smaller(s, tmpdir):
+    """Regression: when ewise operands differ in
+    scheme, the SMALLER operand must be the one repartitioned (planner
+    decision), not unconditionally b -> a's scheme."""
+    from matrel_amd.blocks.matrix import BlockMatrix
+    from matrel_amd.blocks.schemes import COL, ROW
+    from matrel_amd.dsl import Matrix
+    from matrel_amd.exec import metrics
+    from matrel_amd.plan.optimizer import optimize
+    from matrel_amd.plan.physical import plan_physical
+    # A block-sparse (small side) in ROW scheme, B dense in COL scheme.
+    A = Matrix.from_blocks(
+        BlockMatrix.block_sparse_random(128, 128, 0.25, seed=71,
+                                        scheme=ROW))
+    B = Matrix.from_blocks(BlockMatrix.randn(128, 128, seed=72,
+                                             scheme=COL))
+    expr = A + B
+    opt = optimize(expr.plan)
+    pp = plan_physical(opt)
+    # plan shape: target scheme is the BIGGER (dense) operand's
+    assert pp.get(opt).scheme == COL, pp.get(opt).scheme
+    metrics.reset()
+    C = expr.compute()
+    ref = A.to_tensor() + B.to_tensor()
+    assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
+    rep = metrics.report()
+    moved = sum(v["bytes"] for k, v in rep["collectives"].items()
+                if k.startswith("repartition"))
+    # only A's ~25% present blocks move; B (dense, 128x128 fp32) stays.
+    dense_bytes = 128 * 128 * 4
+    assert moved < 0.75 * dense_bytes, (moved, dense_bytes)
+
+
 def _body_ols(s, tmpdir):
     X = s.randn(128, 16, seed=11)
     y = s.randn(128, 1, seed=12)
@@ -253,6 +301,7 @@ def _body_ols(s, tmpdir):
     "_body_elem_sparse_summa", "_body_relational",
     "_body_deterministic_serialized", "_body_mx8_config",
     "_body_save_load", "_body_partition_aware_local", "_body_ols",
+    "_body_maxmin_empty_rank", "_body_ewise_move_smaller",
 ])
 def test_distributed(body, tmp_path):
     _run(body, tmp_path)
