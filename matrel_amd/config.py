@@ -63,6 +63,18 @@ class MatrelConfig:
     # --- execution ---
     # Use a separate HIP stream for collectives so comm overlaps compute.
     overlap_comm: bool = True
+    # SUMMA prefetch depth: how many k-step panel broadcasts are kept in
+    # flight AHEAD of the step being computed (RCCL runs them under the
+    # compute). 1 = classic double buffering; >1 pays more panel memory
+    # for deeper overlap on slow links; ignored when overlap_comm=False
+    # (serialized debug mode). Panel SIZE is block_size; grid shape is
+    # MATREL_GRID - together these are the driver-sweepable SUMMA knobs.
+    summa_prefetch: int = 1
+    # Memory cap for replicating an operand in broadcast-MM. This is ONLY
+    # a memory guard (replication must fit next to the shards in 288 GB
+    # HBM3E); the bcast-vs-SUMMA choice itself is by estimated xGMI bytes
+    # in plan/physical.py.
+    max_bcast_bytes: int = 16 << 30
     # Emit a chrome-trace of the executed DAG (exec/tracing.py).
     trace: bool = False
     trace_path: str = "matrel_trace.json"
@@ -92,6 +104,10 @@ class MatrelConfig:
             cfg.sparse_threshold = float(st)
         cfg.overlap_comm = _env_int("MATREL_OVERLAP_COMM",
                                     int(cfg.overlap_comm)) != 0
+        cfg.summa_prefetch = _env_int("MATREL_SUMMA_PREFETCH",
+                                      cfg.summa_prefetch)
+        cfg.max_bcast_bytes = _env_int("MATREL_MAX_BCAST_BYTES",
+                                       cfg.max_bcast_bytes)
         cfg.deterministic = _env_int("MATREL_DETERMINISTIC",
                                      int(cfg.deterministic)) != 0
         cfg.trace = _env_int("MATREL_TRACE", int(cfg.trace)) != 0

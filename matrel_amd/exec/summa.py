@@ -565,24 +565,40 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 and bs % 128 == 0)
     fused: Dict[Tuple[int, int], torch.Tensor] = {}
 
-    # overlap_comm=True (default): panels for step k+1 are issued BEFORE
-    # step k's GEMMs launch - RCCL's stream runs them under the compute.
-    # False (deterministic/debug mode): each step's broadcasts are issued
-    # only after the previous step's compute is enqueued (serialized).
-    prefetch = get_config().overlap_comm
-    cur = issue_bcast(ks[0])
-    for step, k in enumerate(ks):
-        nxt = (issue_bcast(ks[step + 1])
-               if (step + 1 < len(ks) and prefetch) else None)
+    # overlap_comm=True (default): panel broadcasts for the next
+    # `summa_prefetch` k-steps are kept in flight while the current step
+    # computes (RCCL's stream runs them under the compute; depth 1 =
+    # classic double buffering, deeper costs depth x panel bytes of
+    # memory). overlap_comm=False (deterministic/debug mode): each step's
+    # broadcasts are issued only after the previous step's compute is
+    # enqueued (serialized). Every rank issues in the same global order
+    # (row-group then col-group, by ascending k), which keeps the
+    # cross-communicator enqueue order NCCL-consistent at any depth.
+    from collections import deque
+    cfg_ = get_config()
+    depth = max(1, cfg_.summa_prefetch) if cfg_.overlap_comm else 0
+    pending: deque = deque()
+    issued = 0
+
+    def _fill(target: int) -> None:
+        nonlocal issued
+        while issued < len(ks) and len(pending) < target:
+            pending.append((ks[issued], issue_bcast(ks[issued])))
+            issued += 1
+
+    _fill(1 + depth)
+    for _step in range(len(ks)):
+        k, cur = pending.popleft()
+        if depth > 0:
+            _fill(depth)  # keep `depth` panels in flight beyond cur
         a_rows, a_stack = cur["a_rows"], cur["a_stack"]
         a_sc, a_spinfo, a_dn = cur["a_sc"], cur["a_spinfo"], cur["a_dn"]
         b_cols, b_stack, b_sc = cur["b_cols"], cur["b_stack"], cur["b_sc"]
         for w_ in cur["waits"]:
             w_.wait()
         if not (k in ka and k in kb):
-            if nxt is None and step + 1 < len(ks):
-                nxt = issue_bcast(ks[step + 1])
-            cur = nxt
+            if depth == 0:
+                _fill(1)
             continue
         for n_j, j in enumerate(b_cols):
             bt = b_stack[n_j]  # already transposed by the panel owner
@@ -622,9 +638,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                         fused[(i, j)] = ob
                         continue
                 blockops.gemm_bt_acc(c, a_stack[n_i], bt, beta0=first)
-        if nxt is None and step + 1 < len(ks):
-            nxt = issue_bcast(ks[step + 1])
-        cur = nxt
+        if depth == 0:
+            _fill(1)  # serialized mode: issue only after compute enqueued
 
     for (i, j), c in acc.items():
         h, w = out.block_shape(i, j)

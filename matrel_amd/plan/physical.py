@@ -29,10 +29,11 @@ from ..exec.runtime import get_runtime
 from .logical import (Agg, CrossJoin, Ewise, Inverse, Leaf, MatMul, Plan,
                       ScalarOp, Select, SemiJoinValue, Solve, Transpose)
 
-# Operands at or below this many bytes are candidates for replication
-# (broadcast-MM). 288 GB HBM3E per GPU means replication is cheap in memory;
-# the bound is really about broadcast time vs saved SUMMA steps.
-BCAST_BYTES_THRESHOLD = 1 << 30  # 1 GiB
+# Replication candidacy is a MEMORY guard only (the replicated operand must
+# fit next to the shards in 288 GB HBM3E): configurable via
+# MATREL_MAX_BCAST_BYTES. The bcast-vs-SUMMA decision itself is purely the
+# estimated-xGMI-bytes comparison in _plan_matmul - broadcast-MM wins
+# whenever rb < lb/pr + rb/pc (+ repartition moves), however large rb is.
 
 
 @dataclass
@@ -122,10 +123,12 @@ def plan_physical(root: Plan) -> PhysicalPlan:
             cands.append(("local", 0.0, ROW))
         if ch[0].scheme == BCAST and ch[1].scheme == COL:
             cands.append(("local", 0.0, COL))
-        if rb <= BCAST_BYTES_THRESHOLD:
+        from ..config import get_config
+        mem_cap = get_config().max_bcast_bytes
+        if rb <= mem_cap:
             move_a = 0.0 if ch[0].scheme == ROW else lb / w
             cands.append(("bcast_b", rb + move_a, ROW))
-        if lb <= BCAST_BYTES_THRESHOLD:
+        if lb <= mem_cap:
             move_b = 0.0 if ch[1].scheme == COL else rb / w
             cands.append(("bcast_a", lb + move_b, COL))
         # SUMMA: every GPU receives A-panels (N*K/pr of A rows it shares)
