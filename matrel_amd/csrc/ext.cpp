@@ -67,6 +67,12 @@ void launch_spmm_csr_acc_bf16(float*, const int*, const int*, const void*,
                               const void*, long, long, hipStream_t);
 void launch_spmm_csr_acc_f32(float*, const int*, const int*, const float*,
                              const float*, long, long, hipStream_t);
+int launch_spgemm_csr_csr_bf16(float*, const int*, const int*, const void*,
+                               const int*, const int*, const void*, long,
+                               long, hipStream_t);
+int launch_spgemm_csr_csr_f32(float*, const int*, const int*, const float*,
+                              const int*, const int*, const float*, long,
+                              long, hipStream_t);
 void launch_transpose_f32(float*, const float*, long, long, hipStream_t);
 }
 
@@ -310,6 +316,43 @@ void spmm_csr_acc(torch::Tensor C, torch::Tensor crow, torch::Tensor cols,
   }
 }
 
+// C(f32, MxN) += A_csr(M x K) @ B_csr(K x N) - sparse x sparse block
+// product (LDS ds_add_f32 accumulator row; spgemm.hip). Returns false
+// when N exceeds the 64 KiB LDS row (caller falls back to densified B).
+bool spgemm_csr_csr_acc(torch::Tensor C, torch::Tensor crowA,
+                        torch::Tensor colsA, torch::Tensor valsA,
+                        torch::Tensor crowB, torch::Tensor colsB,
+                        torch::Tensor valsB) {
+  check_2d(C, "C");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32, "C must be fp32");
+  TORCH_CHECK(crowA.scalar_type() == torch::kInt32 &&
+              colsA.scalar_type() == torch::kInt32 &&
+              crowB.scalar_type() == torch::kInt32 &&
+              colsB.scalar_type() == torch::kInt32, "indices must be int32");
+  TORCH_CHECK(valsA.scalar_type() == valsB.scalar_type(),
+              "operand value dtypes must match");
+  long M = C.size(0), N = C.size(1);
+  TORCH_CHECK(crowA.numel() == M + 1, "crowA length");
+  auto crA = crowA.contiguous(), clA = colsA.contiguous();
+  auto vlA = valsA.contiguous();
+  auto crB = crowB.contiguous(), clB = colsB.contiguous();
+  auto vlB = valsB.contiguous();
+  int ok;
+  if (valsA.scalar_type() == torch::kBFloat16) {
+    ok = launch_spgemm_csr_csr_bf16(
+        C.data_ptr<float>(), crA.data_ptr<int>(), clA.data_ptr<int>(),
+        vlA.data_ptr(), crB.data_ptr<int>(), clB.data_ptr<int>(),
+        vlB.data_ptr(), M, N, cur_stream());
+  } else {
+    TORCH_CHECK(valsA.scalar_type() == torch::kFloat32, "bf16/fp32 only");
+    ok = launch_spgemm_csr_csr_f32(
+        C.data_ptr<float>(), crA.data_ptr<int>(), clA.data_ptr<int>(),
+        vlA.data_ptr<float>(), crB.data_ptr<int>(), clB.data_ptr<int>(),
+        vlB.data_ptr<float>(), M, N, cur_stream());
+  }
+  return ok != 0;
+}
+
 // bf16 -> (e4m3 bytes, e8m0 scales [rows][cols/32])
 std::vector<torch::Tensor> mx8_convert(torch::Tensor a) {
   check_2d(a, "a");
@@ -460,6 +503,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "execute a block-GEMM task list natively (no Python per launch)");
   m.def("transpose_batch", &transpose_batch, "batched transpose");
   m.def("spmm_csr_acc", &spmm_csr_acc, "C(f32) += CSR @ B (within-block sparse)");
+  m.def("spgemm_csr_csr_acc", &spgemm_csr_csr_acc,
+        "C(f32) += A_csr @ B_csr (sparse x sparse block product)");
   m.def("mx8_convert", &mx8_convert, "bf16 -> (e4m3, e8m0 block scales)");
   m.def("gemm_bt_mx8_acc", &gemm_bt_mx8_acc,
         "C(f32) += A8 @ Bt8^T (MX-fp8 scaled MFMA, ~2x bf16 rate)");

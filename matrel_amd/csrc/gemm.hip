@@ -890,6 +890,217 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
 #undef MFMA_KK
 #undef HOFF
 
+// ----------------- bf16, 256^2 tile, 2-barrier pipelined ring (v14) ------
+// Round-2 restructure of the phased kernel, targeting the 25.2% parked
+// fraction rocprof showed on variant 6 (profiles/README.md bf16v6_16k_pmc):
+// the 4-phase kernel pays 4 barriers per K-tile and starts EVERY phase
+// with a 4-12 ds_read burst + lgkmcnt drain during which the SIMD's MFMA
+// pipe idles. Two structural changes, same register plan (48 fragment
+// VGPRs, B0/B1 re-read in the second half exactly like variant 6):
+//
+//  1. TWO barriers per tile instead of four. All 8 staging glds of tile
+//     t+1 issue during the first half of tile t; the mid-tile barrier M
+//     (own-wave vmcnt(0) first) proves the whole of tile t+1 landed, so
+//     the second half may read ANY of it; the end barrier B is a bare
+//     WAR fence (nobody restages buffer `cur` until every wave passed B).
+//  2. Every 8-MFMA cluster runs on fragments read ONE cluster earlier
+//     (software pipelining at cluster granularity, sched_barrier(0)
+//     fenced): cluster j's lgkmcnt is satisfied by the time cluster j-1's
+//     512+ cycles of MFMA pipe time have drained, so no cluster head ever
+//     parks on LDS. PF0 additionally prefetches the next TILE's head
+//     fragments (A0',B0' kk0) between M and B - legal exactly because M
+//     proved the nxt buffer - so even the tile boundary starts on MFMAs.
+//
+// Cluster stream per tile (Q0=C(0,0) Q1=C(0,1) Q2=C(1,1) Q3=C(1,0);
+// af[kk][2] A-frags, bfr[kk][4] B-frags, rotating in place):
+//   C0:Q0kk0(a0,b0) | stage A0'B0' | R:a1,b1 | C1:Q0kk1 | stage B1'A1' |
+//   R:b2->bfr0 | C2:Q1kk0 | R:b3->bfr1 | C3:Q1kk1 | R:a2->af0 |
+//   vmcnt(0) BARRIER-M | C4:Q2kk0 | R:a3->af1 | C5:Q2kk1 | R:b0->bfr0 |
+//   C6:Q3kk0 | R:b1->bfr1 | C7:Q3kk1 | PF0: R:a0',b0' | BARRIER-B
+
+template <bool BETA0 = false, bool BF16OUT = false, bool PF0 = true>
+__global__ void
+__launch_bounds__(512)
+gemm_bt_bf16_acc_256r_k(float* __restrict__ C, const bf16* __restrict__ A,
+                        const bf16* __restrict__ Bt, int M, int N, int K,
+                        u32 stm_, u32 stn_, bf16* __restrict__ Co = nullptr) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#define HOFFR(b, h) (((b) * 4u + (h)) * (128u * BKH * 2u))
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 qr = (wave & 3) * 32;
+  const u32 qc = (wave >> 2) * 64;
+
+  const u32 ntm = (u32)(M / BM2), ntn = (u32)(N / BN2);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * BM2, n0 = (long)tn * BN2;
+
+  const u32 srow = tid >> 3, sslot = tid & 7;
+  const u32 src_slot = sslot ^ (srow & 7);
+  const int NT = K / BKH;
+
+  auto stage_half = [&](u32 bufoff, int h, int t) {
+    const long k0 = (long)t * BKH;
+    const bf16* src = (h < 2)
+        ? A + (m0 + (h & 1) * 128) * (long)K
+        : Bt + (n0 + (h & 1) * 128) * (long)K;
+    bf16* dst = (bf16*)(smem + bufoff + (u32)h * (128u * BKH * 2u));
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      u32 row = srow + it * 64;
+      glds16(src + row * (long)K + k0 + src_slot * 8,
+             dst + (row * 8u + sslot) * 8u);
+    }
+  };
+
+  f32x4 acc[4][2][4] = {};
+  bf16x8 af[2][2], bfr[2][4];
+
+  const u32 arow_base = qr + (lane & 15);
+  const u32 brow_base = qc + (lane & 15);
+  const u32 kgrp = lane >> 4;
+
+  // slot: which half of the rotating register set to write (0/1);
+  // kk: which 32-wide k sub-slice of the LDS half to read.
+  auto read_a_s = [&](u32 off, int slot, int kk) {
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      u32 row = arow_base + m * 16;
+      u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      af[slot][m] = *(const bf16x8*)(smem + off + byte);
+    }
+  };
+  auto read_b_s = [&](u32 off, int slot, int kk) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      u32 row = brow_base + n * 16;
+      u32 byte = row * 128u + (u32)kk * 64u + kgrp * 16u;
+      byte ^= (row & 7) << 4;
+      bfr[slot][n] = *(const bf16x8*)(smem + off + byte);
+    }
+  };
+
+#define MFMA_RS(q, s)                                                      \
+  _Pragma("unroll") for (int m = 0; m < 2; ++m)                            \
+      _Pragma("unroll") for (int n = 0; n < 4; ++n)                        \
+          acc[q][m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
+              af[s][m], bfr[s][n], acc[q][m][n], 0, 0, 0);
+#define SFENCE() __builtin_amdgcn_sched_barrier(0)
+
+  // static priority for the second-dispatched half (T5 static form)
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  // prologue: stage tile 0 fully, prove it, preload the head fragments
+  stage_half(0u, 0, 0);
+  stage_half(0u, 2, 0);
+  stage_half(0u, 3, 0);
+  stage_half(0u, 1, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  read_a_s(HOFFR(0, 0), 0, 0);  // a0
+  read_b_s(HOFFR(0, 2), 0, 0);  // b0
+
+  const u32 BUFSZ = 4u * 128u * BKH * 2u;
+  for (int t = 0; t < NT; ++t) {
+    const u32 cur = (u32)(t & 1) * BUFSZ, nxt = cur ^ BUFSZ;
+    const bool pre = (t + 1 < NT);
+    if (!PF0 && t > 0) {
+      read_a_s(cur + HOFFR(0, 0), 0, 0);
+      read_b_s(cur + HOFFR(0, 2), 0, 0);
+    }
+    SFENCE();
+    MFMA_RS(0, 0)  // C0
+    SFENCE();
+    if (pre) { stage_half(nxt, 0, t + 1); stage_half(nxt, 2, t + 1); }
+    read_a_s(cur + HOFFR(0, 0), 1, 1);  // a1
+    read_b_s(cur + HOFFR(0, 2), 1, 1);  // b1
+    SFENCE();
+    MFMA_RS(0, 1)  // C1
+    SFENCE();
+    if (pre) { stage_half(nxt, 3, t + 1); stage_half(nxt, 1, t + 1); }
+    read_b_s(cur + HOFFR(0, 3), 0, 0);  // b2 -> bfr[0]
+    SFENCE();
+    MFMA_RS(1, 0)  // C2 (a0 x B1kk0)
+    SFENCE();
+    read_b_s(cur + HOFFR(0, 3), 1, 1);  // b3 -> bfr[1]
+    SFENCE();
+    MFMA_RS(1, 1)  // C3
+    SFENCE();
+    read_a_s(cur + HOFFR(0, 1), 0, 0);  // a2 -> af[0] (A1kk0, own tile)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // t+1 staging landed
+    __builtin_amdgcn_s_barrier();                     // M
+    MFMA_RS(2, 0)  // C4 (A1kk0 x B1kk0)
+    SFENCE();
+    read_a_s(cur + HOFFR(0, 1), 1, 1);  // a3 -> af[1]
+    SFENCE();
+    MFMA_RS(2, 1)  // C5
+    SFENCE();
+    read_b_s(cur + HOFFR(0, 2), 0, 0);  // b0 re-read -> bfr[0]
+    SFENCE();
+    MFMA_RS(3, 0)  // C6 (A1kk0 x B0kk0)
+    SFENCE();
+    read_b_s(cur + HOFFR(0, 2), 1, 1);  // b1 re-read -> bfr[1]
+    SFENCE();
+    MFMA_RS(3, 1)  // C7
+    SFENCE();
+    if (PF0 && pre) {
+      read_a_s(nxt + HOFFR(0, 0), 0, 0);  // a0' (legal: proven at M)
+      read_b_s(nxt + HOFFR(0, 2), 0, 0);  // b0'
+    }
+    __builtin_amdgcn_s_barrier();  // B: bare WAR fence
+  }
+
+  // epilogue: identical to the phased kernel's batched two-pass
+  const u32 crow0 = qr + (lane >> 4) * 4;
+  const u32 ccol0 = qc + (lane & 15);
+  const int QMHR[4] = {0, 0, 1, 1};
+  const int QNHR[4] = {0, 1, 1, 0};
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    float cv[2][4][4];
+    if (!BETA0) {
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          long row = m0 + QMHR[q] * 128 + crow0 + m * 16;
+          long col = n0 + QNHR[q] * 128 + ccol0 + n * 16;
+          const float* cp = C + row * (long)N + col;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) cv[m][n][r] = cp[(long)r * N];
+        }
+    }
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        long row = m0 + QMHR[q] * 128 + crow0 + m * 16;
+        long col = n0 + QNHR[q] * 128 + ccol0 + n * 16;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = (BETA0 ? 0.0f : cv[m][n][r]) + acc[q][m][n][r];
+          if (BF16OUT)
+            Co[(row + r) * (long)N + col] = (bf16)v;
+          else
+            C[(row + r) * (long)N + col] = v;
+        }
+      }
+  }
+}
+#undef MFMA_RS
+#undef SFENCE
+#undef HOFFR
+
 // --------------------------- bf16, 256^2 tile, phased, 32x32x16 MFMA ----
 // Same quadrant-per-phase pipeline as gemm_bt_bf16_acc_256p_k but on
 // v_mfma_f32_32x32x16_bf16 (uarch ceiling 2382 TF vs 2075 for 16x16x32;
@@ -1344,7 +1555,19 @@ extern "C" int launch_gemm_bt_bf16_final(float* C, void* Co, const void* A,
   int nb = (M / BM2) * (N / BN2);
   if (!(M % BM2 == 0 && N % BN2 == 0 && nb >= 192)) return 0;
   size_t lds = 2u * BM2 * BKH * 2 * 2;
-  if (beta0)
+  // ring becomes the default only after the measured A/B (set 14/15)
+  bool ring = (g_gemm_variant == 14 || g_gemm_variant == 15);
+  if (ring && beta0)
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<true, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n, (bf16*)Co);
+  else if (ring)
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<false, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n, (bf16*)Co);
+  else if (beta0)
     hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true, true, true>),
                        dim3((unsigned)nb), dim3(512), lds, stream, C,
                        (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
@@ -1365,10 +1588,16 @@ extern "C" void launch_gemm_bt_bf16_acc_b0(float* C, const void* A,
   int nb = (M / BM2) * (N / BN2);
   if (M % BM2 == 0 && N % BN2 == 0 && nb >= 192) {
     size_t lds = 2u * BM2 * BKH * 2 * 2;
-    hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true, true>),
-                       dim3((unsigned)nb), dim3(512), lds, stream, C,
-                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
-                       g_st_n);
+    if (g_gemm_variant == 14 || g_gemm_variant == 15)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<true, false>),
+                         dim3((unsigned)nb), dim3(512), lds, stream, C,
+                         (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                         g_st_n);
+    else
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256p_k<false, true, true>),
+                         dim3((unsigned)nb), dim3(512), lds, stream, C,
+                         (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                         g_st_n);
     return;
   }
   (void)hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream);
@@ -1422,6 +1651,14 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
           (gemm_bt_bf16_acc_256p_k<false, true, false, false, false, true>),
           grid, dim3(512), lds, stream, C, (const bf16*)A, (const bf16*)Bt,
           M, N, K, g_st_m, g_st_n);
+    else if (v == 14)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<false, false, true>),
+                         grid, dim3(512), lds, stream, C, (const bf16*)A,
+                         (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
+    else if (v == 15)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<false, false, false>),
+                         grid, dim3(512), lds, stream, C, (const bf16*)A,
+                         (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
     else if (v == 12)
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256w4_k, grid, dim3(256), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,

@@ -78,6 +78,27 @@ def spmm_acc(acc: torch.Tensor, a_csr: torch.Tensor,
         acc.add_(a_csr.to_dense().float() @ b.float())
 
 
+def spgemm_acc(acc: torch.Tensor, a_csr: torch.Tensor,
+               b_csr: torch.Tensor) -> bool:
+    """acc(fp32, MxN) += a_csr(M x K) @ b_csr(K x N) - sparse x sparse
+    block product (HIP ds_add_f32 LDS-row kernel on GPU; SURVEY.md §2.1
+    "sparse x sparse block products"). Returns False when the kernel
+    declined (N > 16 KiB LDS row) - caller densifies b instead."""
+    if acc.is_cuda:
+        ext = require_ext()
+        va = a_csr.values()
+        vb = b_csr.values()
+        if va.dtype != vb.dtype:
+            va, vb = va.float(), vb.float()
+        return ext.spgemm_csr_csr_acc(
+            acc, a_csr.crow_indices().to(torch.int32),
+            a_csr.col_indices().to(torch.int32), va.contiguous(),
+            b_csr.crow_indices().to(torch.int32),
+            b_csr.col_indices().to(torch.int32), vb.contiguous())
+    acc.add_(a_csr.to_dense().float() @ b_csr.to_dense().float())
+    return True
+
+
 def _kmult(dtype: torch.dtype) -> int:
     return 64 if dtype == torch.bfloat16 else 32
 

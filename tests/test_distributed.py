@@ -148,6 +148,47 @@ def _body_elem_sparse_summa(s, tmpdir):
     assert torch.allclose(out2.to_dense(), ref2, atol=1e-3)
 
 
+def _body_both_sparse_summa(s, tmpdir):
+    """BOTH operands element-sparse through SUMMA: each panel direction
+    must go over the wire as CSR arrays (nnz-proportional bytes on BOTH
+    panels - VERDICT r1 items 2+3), the local products must run the
+    sparse x sparse path, and the result must match the dense reference."""
+    from matrel_amd.blocks.matrix import BlockMatrix
+    from matrel_amd.blocks.schemes import GRID
+    from matrel_amd.exec import metrics
+    from matrel_amd.exec.summa import summa_matmul
+    metrics.reset()
+    A = BlockMatrix.sparse_random(96, 96, 0.03, seed=25)
+    B = BlockMatrix.sparse_random(96, 64, 0.03, seed=26)
+    out = BlockMatrix(96, 64, A.block_size, GRID, A.dtype, A.device)
+    summa_matmul(A, B, out)
+    ref = A.to_dense() @ B.to_dense()
+    assert torch.allclose(out.to_dense(), ref, atol=1e-3)
+    rep = metrics.report()
+    assert "summa_a_bcast_csr" in rep["collectives"], rep["collectives"]
+    assert "summa_b_bcast_csr" in rep["collectives"], rep["collectives"]
+    # nnz-proportional on BOTH panels: no dense bs^2 payloads at all
+    assert "summa_b_bcast" not in rep["collectives"], rep["collectives"]
+    for key, m in (("summa_a_bcast_csr", A), ("summa_b_bcast_csr", B)):
+        csr_bytes = rep["collectives"][key]["bytes"]
+        dense_equiv = sum(
+            m.block_shape(bi, bj)[0] * m.block_shape(bi, bj)[1]
+            * m.blocks[(bi, bj)].values().element_size()
+            for (bi, bj) in m.blocks)
+        assert csr_bytes < 0.5 * dense_equiv, (key, csr_bytes, dense_equiv)
+    # mixed: sparse A x dense B and dense A x sparse B still correct
+    C = BlockMatrix.randn(96, 64, seed=27)
+    out2 = BlockMatrix(96, 64, A.block_size, GRID, A.dtype, A.device)
+    summa_matmul(A, C, out2)
+    assert torch.allclose(out2.to_dense(), A.to_dense() @ C.to_dense(),
+                          atol=1e-3)
+    D = BlockMatrix.randn(96, 96, seed=28)
+    out3 = BlockMatrix(96, 64, A.block_size, GRID, A.dtype, A.device)
+    summa_matmul(D, B, out3)
+    assert torch.allclose(out3.to_dense(), D.to_dense() @ B.to_dense(),
+                          atol=1e-3)
+
+
 def _body_relational(s, tmpdir):
     """Distributed relational verbs (SURVEY.md §2.1 joins/aggregates over
     matrix data): row-index join (gram), cross-product join + aggregate,
@@ -298,7 +339,8 @@ def _body_ols(s, tmpdir):
 @pytest.mark.parametrize("body", [
     "_body_summa", "_body_bcast_mm", "_body_aggregates", "_body_transpose",
     "_body_repartition", "_body_block_sparse", "_body_sparse_skewed",
-    "_body_elem_sparse_summa", "_body_relational",
+    "_body_elem_sparse_summa", "_body_both_sparse_summa",
+    "_body_relational",
     "_body_deterministic_serialized", "_body_mx8_config",
     "_body_save_load", "_body_partition_aware_local", "_body_ols",
     "_body_maxmin_empty_rank", "_body_ewise_move_smaller",
