@@ -59,6 +59,11 @@ class MatrelConfig:
     # --- sparsity ---
     # A block whose density is below this threshold is stored sparse (CSR).
     sparse_threshold: float = 0.05
+    # GPU compute routing: a CSR block at or above this density computes
+    # faster densified on the MFMA pipe than through the CSR kernels
+    # (measured crossover ~1% at 8-16k blocks - profiles/ spmm isolation
+    # entry). Storage/wire stay CSR; only the local product densifies.
+    spmm_densify_crossover: float = 0.01
 
     # --- execution ---
     # Use a separate HIP stream for collectives so comm overlaps compute.
@@ -102,6 +107,9 @@ class MatrelConfig:
         st = os.environ.get("MATREL_SPARSE_THRESHOLD")
         if st:
             cfg.sparse_threshold = float(st)
+        sc = os.environ.get("MATREL_SPMM_DENSIFY_CROSSOVER")
+        if sc:
+            cfg.spmm_densify_crossover = float(sc)
         cfg.overlap_comm = _env_int("MATREL_OVERLAP_COMM",
                                     int(cfg.overlap_comm)) != 0
         cfg.summa_prefetch = _env_int("MATREL_SUMMA_PREFETCH",

@@ -918,7 +918,16 @@ gemm_bt_bf16_acc_256p_k(float* __restrict__ C, const bf16* __restrict__ A,
 //   vmcnt(0) BARRIER-M | C4:Q2kk0 | R:a3->af1 | C5:Q2kk1 | R:b0->bfr0 |
 //   C6:Q3kk0 | R:b1->bfr1 | C7:Q3kk1 | PF0: R:a0',b0' | BARRIER-B
 
-template <bool BETA0 = false, bool BF16OUT = false, bool PF0 = true>
+// ONEBAR (variant 16): drop the mid-tile barrier entirely - the measured
+// failure of the 2-barrier form (947 vs 1199 TF, gpurun r2 call 1) was
+// the mid-tile vmcnt(0): staging glds issued ~1300 cycles earlier parked
+// every wave under loaded-HBM latency. With ONE barrier per tile the
+// proof moves to the tile-end barrier (~3000+ cycles of slack for the
+// same glds) and the cross-tile prefetch (PF0) is dropped because the
+// nxt buffer is only proven at B; the tile head pays one ~100-cycle LDS
+// drain instead.
+template <bool BETA0 = false, bool BF16OUT = false, bool PF0 = true,
+          bool ONEBAR = false>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_bf16_acc_256r_k(float* __restrict__ C, const bf16* __restrict__ A,
@@ -1037,8 +1046,10 @@ gemm_bt_bf16_acc_256r_k(float* __restrict__ C, const bf16* __restrict__ A,
     MFMA_RS(1, 1)  // C3
     SFENCE();
     read_a_s(cur + HOFFR(0, 1), 0, 0);  // a2 -> af[0] (A1kk0, own tile)
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // t+1 staging landed
-    __builtin_amdgcn_s_barrier();                     // M
+    if constexpr (!ONEBAR) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // staging landed
+      __builtin_amdgcn_s_barrier();                     // M
+    }
     MFMA_RS(2, 0)  // C4 (A1kk0 x B1kk0)
     SFENCE();
     read_a_s(cur + HOFFR(0, 1), 1, 1);  // a3 -> af[1]
@@ -1053,11 +1064,13 @@ gemm_bt_bf16_acc_256r_k(float* __restrict__ C, const bf16* __restrict__ A,
     SFENCE();
     MFMA_RS(3, 1)  // C7
     SFENCE();
-    if (PF0 && pre) {
+    if (!ONEBAR && PF0 && pre) {
       read_a_s(nxt + HOFFR(0, 0), 0, 0);  // a0' (legal: proven at M)
       read_b_s(nxt + HOFFR(0, 2), 0, 0);  // b0'
     }
-    __builtin_amdgcn_s_barrier();  // B: bare WAR fence
+    if constexpr (ONEBAR)  // staging issued ~6 clusters ago: full slack
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // B
   }
 
   // epilogue: identical to the phased kernel's batched two-pass
@@ -1555,9 +1568,21 @@ extern "C" int launch_gemm_bt_bf16_final(float* C, void* Co, const void* A,
   int nb = (M / BM2) * (N / BN2);
   if (!(M % BM2 == 0 && N % BN2 == 0 && nb >= 192)) return 0;
   size_t lds = 2u * BM2 * BKH * 2 * 2;
-  // ring becomes the default only after the measured A/B (set 14/15)
-  bool ring = (g_gemm_variant == 14 || g_gemm_variant == 15);
-  if (ring && beta0)
+  // ring becomes the default only after the measured A/B (set 14/15/16)
+  bool ring = (g_gemm_variant == 14 || g_gemm_variant == 15 ||
+               g_gemm_variant == 16);
+  bool onebar = (g_gemm_variant == 16);
+  if (ring && onebar && beta0)
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<true, true, false, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n, (bf16*)Co);
+  else if (ring && onebar)
+    hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<false, true, false, true>),
+                       dim3((unsigned)nb), dim3(512), lds, stream, C,
+                       (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                       g_st_n, (bf16*)Co);
+  else if (ring && beta0)
     hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<true, true>),
                        dim3((unsigned)nb), dim3(512), lds, stream, C,
                        (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
@@ -1588,7 +1613,12 @@ extern "C" void launch_gemm_bt_bf16_acc_b0(float* C, const void* A,
   int nb = (M / BM2) * (N / BN2);
   if (M % BM2 == 0 && N % BN2 == 0 && nb >= 192) {
     size_t lds = 2u * BM2 * BKH * 2 * 2;
-    if (g_gemm_variant == 14 || g_gemm_variant == 15)
+    if (g_gemm_variant == 16)
+      hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<true, false, false, true>),
+                         dim3((unsigned)nb), dim3(512), lds, stream, C,
+                         (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
+                         g_st_n);
+    else if (g_gemm_variant == 14 || g_gemm_variant == 15)
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<true, false>),
                          dim3((unsigned)nb), dim3(512), lds, stream, C,
                          (const bf16*)A, (const bf16*)Bt, M, N, K, g_st_m,
@@ -1659,6 +1689,11 @@ extern "C" void launch_gemm_bt_bf16_acc(float* C, const void* A,
       hipLaunchKernelGGL((gemm_bt_bf16_acc_256r_k<false, false, false>),
                          grid, dim3(512), lds, stream, C, (const bf16*)A,
                          (const bf16*)Bt, M, N, K, g_st_m, g_st_n);
+    else if (v == 16)
+      hipLaunchKernelGGL(
+          (gemm_bt_bf16_acc_256r_k<false, false, false, true>), grid,
+          dim3(512), lds, stream, C, (const bf16*)A, (const bf16*)Bt, M, N,
+          K, g_st_m, g_st_n);
     else if (v == 12)
       hipLaunchKernelGGL(gemm_bt_bf16_acc_256w4_k, grid, dim3(256), lds,
                          stream, C, (const bf16*)A, (const bf16*)Bt, M, N,

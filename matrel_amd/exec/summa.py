@@ -280,6 +280,19 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
     return True
 
 
+def _csr_density(t: torch.Tensor) -> float:
+    return t._nnz() / max(1, t.shape[0] * t.shape[1])
+
+
+def _route_dense(t: torch.Tensor, device: torch.device) -> bool:
+    """True when a CSR block computes faster densified on the MFMA pipe
+    than through the CSR kernels (measured crossover ~1% density at
+    8-16k blocks; profiles/ spmm isolation entry). Storage and wire stay
+    CSR - this only routes the local product."""
+    return (device.type == "cuda" and blockops.is_sparse(t)
+            and _csr_density(t) >= get_config().spmm_densify_crossover)
+
+
 def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     """Pure-local block multiply: every needed A/B block is already on the
     rank that owns the corresponding C block (single process, or
@@ -292,6 +305,18 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         return
     fused_ok = (out.device.type == "cuda" and out.dtype == torch.bfloat16)
     mx8 = get_config().gemm_precision == "mx8"
+    a_dense_cache: Dict[Tuple[int, int], torch.Tensor] = {}
+
+    def a_routed(bi: int, k: int) -> torch.Tensor:
+        t = a.blocks[(bi, k)]
+        if _route_dense(t, out.device):
+            d = a_dense_cache.get((bi, k))
+            if d is None:
+                d = blockops.densify(t)
+                a_dense_cache[(bi, k)] = d
+            return d
+        return t
+
     for (bi, bj) in out.owned_indices():
         ks = [k for k in range(a.grid_cols)
               if (bi, k) in a.blocks and (k, bj) in b.blocks]
@@ -299,20 +324,21 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             continue
         shape = out.block_shape(bi, bj)
         c = torch.empty(shape, dtype=torch.float32, device=out.device)
-        any_sparse = any(blockops.is_sparse(a.blocks[(bi, k)]) for k in ks)
+        any_sparse = any(blockops.is_sparse(a_routed(bi, k)) for k in ks)
         use_fused = fused_ok and not any_sparse
         outblk = (torch.empty(shape, dtype=out.dtype, device=out.device)
                   if use_fused else None)
         fused_done = False
         for idx, k in enumerate(ks):
-            ab = a.blocks[(bi, k)]
+            ab = a_routed(bi, k)
             bb = b.blocks[(k, bj)]
             first = idx == 0
             last = idx == len(ks) - 1
             if blockops.is_sparse(ab):
                 if first:
                     c.zero_()
-                if blockops.is_sparse(bb):
+                if blockops.is_sparse(bb) and \
+                        not _route_dense(bb, out.device):
                     # sparse x sparse: CSR x CSR kernel, no densified-B
                     # FLOPs or bytes (VERDICT r1 missing #2)
                     if blockops.spgemm_acc(c, ab, bb):
@@ -671,6 +697,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             if depth == 0:
                 _fill(1)
             continue
+        a_dense: Dict[int, torch.Tensor] = {}  # dense-routed CSR-A blocks
         for j in b_cols:
             bsp = b_spinfo.get(j)
             b_csr = None
@@ -704,11 +731,22 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                     acc[(i, j)] = c
                 sp = a_spinfo.get(i)
                 if sp is not None:
-                    if first:
-                        c.zero_()
                     crow, cols, vals, h, w = sp
                     a_csr = torch.sparse_csr_tensor(
                         crow, cols, vals, size=(h, w))
+                    if (out.device.type == "cuda" and len(vals)
+                            >= get_config().spmm_densify_crossover * h * w):
+                        # dense-route: CSR wire (nnz bytes) but MFMA
+                        # compute (faster above the measured crossover)
+                        ad = a_dense.get(i)
+                        if ad is None:
+                            ad = _pad_t(blockops.densify(a_csr), bs)
+                            a_dense[i] = ad
+                        bt_j = _need_bt() if bt is None else bt
+                        blockops.gemm_bt_acc(c, ad, bt_j, beta0=first)
+                        continue
+                    if first:
+                        c.zero_()
                     if b_csr is not None:
                         # both wires sparse: CSR x CSR block product
                         if blockops.spgemm_acc(c[:h], a_csr, b_csr):

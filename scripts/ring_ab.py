@@ -20,7 +20,7 @@ from matrel_amd.ops.backend import require_ext
 
 ext = require_ext()
 ext.set_supertile(4, 4)
-VARIANTS = (14, 15)
+VARIANTS = (14, 15, 16)
 
 
 def gemm(v, a, bt, n):
@@ -91,7 +91,7 @@ def perf_final(n, it, reps):
     c = torch.empty(n, n, dtype=torch.float32, device="cuda")
     ob = torch.empty(n, n, dtype=torch.bfloat16, device="cuda")
     res = {}
-    for v in (6, 14):
+    for v in (6, 16):
         ext.set_gemm_variant(v)
         xs = []
         for rep in range(reps):
