@@ -4,6 +4,7 @@ tracing output, metrics report, debug env."""
 import json
 import os
 
+import pytest
 import torch
 
 
@@ -51,6 +52,33 @@ def test_tracing_chrome_output(session, tmp_path):
     ev = json.load(open(p))["traceEvents"]
     names = {e["name"] for e in ev}
     assert "MatMul" in names and "Agg" in names
+
+
+@pytest.mark.gpu
+def test_tracing_gpu_hipevent_spans(gpu_session, tmp_path):
+    """hipEvent per-op GPU spans: tid-1 '[gpu]' events
+    with nonzero duration on the same timeline as the host spans, no
+    MATREL_TRACE_SYNC serialization."""
+    from matrel_amd.exec import tracing
+    tracing.clear()
+    tracing.enable(True)
+    s = gpu_session
+    A = s.randn(1024, 1024, seed=4)
+    (A @ A.T).sum().compute()
+    p = str(tmp_path / "trace_gpu.json")
+    tracing.dump_trace(p)
+    tracing.enable(False)
+    ev = json.load(open(p))["traceEvents"]
+    gpu = [e for e in ev if e.get("tid") == 1 and e.get("ph") == "X"]
+    host = [e for e in ev if e.get("tid") == 0 and e.get("ph") == "X"]
+    assert any("MatMul" in e["name"] for e in gpu), [e["name"] for e in gpu]
+    mm_gpu = next(e for e in gpu if "MatMul" in e["name"])
+    assert mm_gpu["dur"] > 0
+    # same timeline: GPU span lies within ~the host trace window
+    t0 = min(e["ts"] for e in host) - 1e6
+    t1 = max(e["ts"] + e["dur"] for e in host) + 1e6
+    assert t0 <= mm_gpu["ts"] <= t1
+    tracing.clear()
 
 
 def test_metrics_report(session):
