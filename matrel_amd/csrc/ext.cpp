@@ -74,6 +74,10 @@ int launch_spgemm_csr_csr_f32(float*, const int*, const int*, const float*,
                               const int*, const int*, const float*, long,
                               long, hipStream_t);
 void launch_transpose_f32(float*, const float*, long, long, hipStream_t);
+void launch_ewise_transpose_bf16(int, void*, const void*, const void*, long,
+                                 long, hipStream_t);
+void launch_ewise_transpose_f32(int, float*, const float*, const float*,
+                                long, long, hipStream_t);
 }
 
 namespace {
@@ -389,6 +393,23 @@ void gemm_bt_mx8_acc(torch::Tensor C, torch::Tensor A8, torch::Tensor Asc,
                          beta0 ? 1 : 0, cur_stream());
 }
 
+// out = (a OP b)^T in one pass (fused elementwise + transpose)
+torch::Tensor ewise_transpose(int64_t op, torch::Tensor a, torch::Tensor b) {
+  check_2d(a, "a");
+  check_2d(b, "b");
+  TORCH_CHECK(a.sizes() == b.sizes() && a.scalar_type() == b.scalar_type());
+  auto out = torch::empty({a.size(1), a.size(0)}, a.options());
+  if (is_bf16(a))
+    launch_ewise_transpose_bf16((int)op, out.data_ptr(), a.data_ptr(),
+                                b.data_ptr(), a.size(0), a.size(1),
+                                cur_stream());
+  else
+    launch_ewise_transpose_f32((int)op, out.data_ptr<float>(),
+                               a.data_ptr<float>(), b.data_ptr<float>(),
+                               a.size(0), a.size(1), cur_stream());
+  return out;
+}
+
 torch::Tensor transpose(torch::Tensor a) {
   check_2d(a, "a");
   auto out = torch::empty({a.size(1), a.size(0)}, a.options());
@@ -499,6 +520,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce2d", &reduce2d, "full/row/col reduction");
   m.def("trace", &trace, "diagonal sum");
   m.def("transpose", &transpose, "materialized transpose");
+  m.def("ewise_transpose", &ewise_transpose,
+        "(a OP b)^T fused single pass");
   m.def("gemm_bt_batch", &gemm_bt_batch,
         "execute a block-GEMM task list natively (no Python per launch)");
   m.def("transpose_batch", &transpose_batch, "batched transpose");

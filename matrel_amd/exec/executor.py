@@ -90,40 +90,80 @@ class Executor:
     # --- operators ------------------------------------------------------
 
     def _transpose(self, n: Transpose) -> BlockMatrix:
-        a = self._eval(n.children[0])
+        # Fused (A op B)^T: single-pass ewise+transpose kernel per block
+        #  - 3 HBM streams instead of 5. Applies when
+        # the child is an un-memoized Ewise; a CSE-shared child that is
+        # evaluated later elsewhere simply recomputes the (cheap) ewise.
+        ch = n.children[0]
+        locals_t: dict
+        if (isinstance(ch, Ewise) and id(ch) not in self.cache
+                and ch.op in ("add", "sub", "mul", "div", "min", "max")):
+            a = self._eval(ch.children[0])
+            b = self._eval(ch.children[1])
+            if a.block_size != b.block_size:
+                raise ValueError("elementwise op needs equal block sizes")
+            tgt = self.physical.get(ch).scheme
+            if a.scheme != tgt:
+                a = a.repartition(tgt)
+            if b.scheme != tgt:
+                b = b.repartition(tgt)
+            op = {"add": blockops.EW_ADD, "sub": blockops.EW_SUB,
+                  "mul": blockops.EW_MUL, "div": blockops.EW_DIV,
+                  "min": blockops.EW_MIN, "max": blockops.EW_MAX}[ch.op]
+            if ch.op in ("add", "sub", "min", "max"):
+                keys = set(a.blocks) | set(b.blocks)
+                pm = a.presence() | b.presence()
+            elif ch.op == "mul":
+                keys = set(a.blocks) & set(b.blocks)
+                pm = a.presence() & b.presence()
+            else:  # div (matches _ewise's key rule)
+                keys = set(a.blocks)
+                pm = a.presence().clone()
+            locals_t = {}
+            for (bi, bj) in keys:
+                ta = a.get_block(bi, bj)
+                tb = b.get_block(bi, bj)
+                if ta is None:
+                    ta = a.get_block_dense(bi, bj)
+                if tb is None:
+                    tb = b.get_block_dense(bi, bj)
+                ft = blockops.ewise_transpose(op, ta, tb)
+                # routing receivers allocate out.dtype buffers
+                locals_t[(bj, bi)] = ft.to(a.dtype)
+        else:
+            a = self._eval(ch)
+            pm = a.presence()
+            locals_t = {(bj, bi): blockops.transpose(t)
+                        for (bi, bj), t in a.local_items()}
+
         out = BlockMatrix(a.ncols, a.nrows, a.block_size,
                           a.scheme.transpose(), a.dtype, a.device)
         rt = get_runtime()
         if not rt.is_distributed or a.scheme.is_replicated():
-            for (bi, bj), t in a.local_items():
-                out.blocks[(bj, bi)] = blockops.transpose(t)
+            out.blocks.update(locals_t)
             return out
-        # Distributed: transpose local blocks, then route each to the rank
-        # owning the transposed index under out.scheme.
-        tmp = BlockMatrix(a.ncols, a.nrows, a.block_size, a.scheme.transpose(),
-                          a.dtype, a.device)
-        for (bi, bj), t in a.local_items():
-            tmp.blocks[(bj, bi)] = blockops.transpose(t)
-        # blocks now live where A(j,i)'s owner was; redistribute explicitly
-        pm = a.presence()
+        # Distributed: route each transposed block to the rank owning the
+        # transposed index under out.scheme. pm is the (global) presence
+        # of the UNtransposed source; the holder of block (bj, bi) of the
+        # source is the one carrying transposed block (bi, bj) here.
         out_sch = out.scheme
-        for (bi, bj) in tmp.block_indices():
-            src = a.owner(bj, bi)  # rank holding transposed block (bi,bj)
+        for (bi, bj) in out.block_indices():
+            src = a.scheme.owner(bj, bi, rt.world_size, rt.grid)
             dst = out_sch.owner(bi, bj, rt.world_size, rt.grid)
             if not pm[bj, bi]:
                 continue
             if src == dst:
                 if rt.rank == src:
-                    out.blocks[(bi, bj)] = tmp.blocks[(bi, bj)]
+                    out.blocks[(bi, bj)] = locals_t[(bi, bj)]
             elif dst == -1:
-                t = (tmp.blocks[(bi, bj)] if rt.rank == src else
+                t = (locals_t[(bi, bj)] if rt.rank == src else
                      torch.empty(out.block_shape(bi, bj), dtype=out.dtype,
                                  device=out.device))
                 dist.broadcast(t.contiguous(), src=src)
                 out.blocks[(bi, bj)] = t
             else:
                 if rt.rank == src:
-                    dist.send(tmp.blocks[(bi, bj)].contiguous(), dst=dst)
+                    dist.send(locals_t[(bi, bj)].contiguous(), dst=dst)
                 elif rt.rank == dst:
                     t = torch.empty(out.block_shape(bi, bj), dtype=out.dtype,
                                     device=out.device)

@@ -82,6 +82,26 @@ def _body_transpose(s, tmpdir):
     assert torch.allclose(At.to_tensor(), A.to_tensor().t(), atol=0)
 
 
+def _body_ewise_transpose_fused(s, tmpdir):
+    """(A op B).T takes the fused single-pass path in the executor,
+    including with block-sparse operands (union/intersection presence)
+    and ragged edges, over real routing collectives."""
+    A = s.randn(96, 64, seed=61)
+    B = s.randn(96, 64, seed=62)
+    for expr, ref in (((A + B).T, (A.to_tensor() + B.to_tensor()).t()),
+                      ((A * B).T, (A.to_tensor() * B.to_tensor()).t()),
+                      ((A - B).T, (A.to_tensor() - B.to_tensor()).t())):
+        out = expr.compute()
+        assert torch.allclose(out.to_tensor(), ref, atol=1e-5)
+    S = s.block_sparse_random(96, 64, 0.4, seed=63)
+    out = (S + A).T.compute()
+    assert torch.allclose(out.to_tensor(),
+                          (S.to_tensor() + A.to_tensor()).t(), atol=1e-5)
+    out = (S * A).T.compute()
+    assert torch.allclose(out.to_tensor(),
+                          (S.to_tensor() * A.to_tensor()).t(), atol=1e-5)
+
+
 def _body_repartition(s, tmpdir):
     from matrel_amd.blocks.schemes import BCAST, COL, ROW
     A = s.randn(96, 64, seed=7)
@@ -338,6 +358,7 @@ def _body_ols(s, tmpdir):
 
 @pytest.mark.parametrize("body", [
     "_body_summa", "_body_bcast_mm", "_body_aggregates", "_body_transpose",
+    "_body_ewise_transpose_fused",
     "_body_repartition", "_body_block_sparse", "_body_sparse_skewed",
     "_body_elem_sparse_summa", "_body_both_sparse_summa",
     "_body_relational",

@@ -142,6 +142,22 @@ def test_transpose(ext, shape):
         assert torch.equal(out.float().cpu(), a.float().cpu().t())
 
 
+@pytest.mark.parametrize("shape", [(128, 128), (100, 70), (1, 257)])
+def test_ewise_transpose_fused(ext, shape):
+    """(a OP b)^T single-pass kernel vs torch fp32 reference."""
+    import torch as _t
+    ops = {0: _t.add, 1: _t.sub, 2: _t.mul, 4: _t.minimum, 5: _t.maximum}
+    for dt in (torch.bfloat16, torch.float32):
+        a = _randn(*shape, dtype=dt, seed=31)
+        b = _randn(*shape, dtype=dt, seed=32)
+        for op, fn in ops.items():
+            out = ext.ewise_transpose(op, a, b)
+            ref = fn(a.float().cpu(), b.float().cpu()).t()
+            assert out.shape == (shape[1], shape[0])
+            tol = 0.05 if dt == torch.bfloat16 else 1e-6
+            assert torch.allclose(out.float().cpu(), ref, atol=tol), (op, dt)
+
+
 # --------------------------------------------------------- reductions ----
 
 def test_reduce_full(ext):
