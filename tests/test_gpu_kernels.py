@@ -457,3 +457,31 @@ def test_dsl_agg_pushdown_gpu(gpu_session):
     tr = (A @ C).trace().item()
     ref_tr = (A.to_tensor().float() @ C.to_tensor().float()).trace().item()
     assert abs(tr - ref_tr) <= 1e-3 * abs(ref_tr) + 0.5
+
+
+def test_multiblock_16k_orchestration():
+    """Block orchestration ABOVE toy scale: a
+    32768^2 matmul at the production 16384 block size - 2x2 grid of
+    512-MiB bf16 blocks through local_matmul (bt cache, fused last-k
+    epilogue) - against torch's hipBLASLt product."""
+    import matrel_amd as ma
+    from matrel_amd.exec.runtime import reset_runtime
+    reset_runtime()
+    s = ma.MatrelSession(ma.MatrelConfig(block_size=16384))
+    try:
+        n = 32768
+        A = s.randn(n, n, seed=41)
+        B = s.randn(n, n, seed=42)
+        C = (A @ B).compute()
+        bm = C.blocks
+        assert bm.grid_rows == 2 and bm.grid_cols == 2
+        assert len(bm.blocks) == 4
+        at = A.to_tensor().to(torch.bfloat16)
+        bt = B.to_tensor().to(torch.bfloat16)
+        ref = (at @ bt).float()
+        got = C.to_tensor()
+        rel = (got - ref).abs().max().item() / ref.abs().max().item()
+        # bf16 accumulation-order differences only
+        assert rel < 5e-2, rel
+    finally:
+        reset_runtime()
