@@ -486,6 +486,158 @@ gemm_bt_mx8_acc_k128cv(float* __restrict__ C,
     }
 }
 
+// ---- BK=128 v3: the simple structure + SCALE PREFETCH ------------------
+// The shipped k128 kernel loads 24 strided scale BYTES (Asc/Bsc at
+// stride M/N) at every tile head; the first mfma_scale waits on them
+// (L2 ~200-900 cyc) on top of the kk0 LDS drain - a per-tile head stall
+// the mx8 PMC showed as the 36.2% parked fraction. v3 rotates the scale
+// words one tile ahead: tile t+1's 6 scale words load right after tile
+// t's staging issues, covered by the whole tile body; registers +12.
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx8_acc_k128v3(float* __restrict__ C,
+                       const unsigned char* __restrict__ A,
+                       const unsigned char* __restrict__ Asc,
+                       const unsigned char* __restrict__ Bt,
+                       const unsigned char* __restrict__ Bsc, int M, int N,
+                       int K, int beta0) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned char* As = (unsigned char*)smem;  // [2][256*128]
+  unsigned char* Bs = (unsigned char*)(smem + 2 * MXBM * 128);
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;
+
+  const u32 ntm = (u32)(M / MXBM), ntn = (u32)(N / MXBN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (ntm % 4 == 0 && ntn % 4 == 0) {
+    supertile_map(bid, ntm, ntn, 4, 4, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * MXBM, n0 = (long)tn * MXBN;
+
+  const u32 srow_b = tid >> 3;
+  const u32 sslot = tid & 7;
+
+  auto stage = [&](int buf, int t) {
+    const long k0 = (long)t * 128;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      u32 row = srow_b + p * 64;
+      u32 ss = sslot ^ ((row >> 1) & 7);
+      glds16(A + (m0 + row) * (long)K + k0 + ss * 16,
+             As + buf * MXBM * 128 + row * 128u + sslot * 16u);
+      glds16(Bt + (n0 + row) * (long)K + k0 + ss * 16,
+             Bs + buf * MXBM * 128 + row * 128u + sslot * 16u);
+    }
+  };
+
+  f32x16 acc[4][2] = {};
+  const u32 arow = wm * 128 + (lane & 31);
+  const u32 brow = wn * 64 + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  auto load_scales = [&](int t, int* sa_w, int* sb_w) {
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      long r = m0 + arow + m * 32;
+      sa_w[m] = (int)Asc[(long)(t * 4 + 0) * M + r] |
+                ((int)Asc[(long)(t * 4 + 1) * M + r] << 8) |
+                ((int)Asc[(long)(t * 4 + 2) * M + r] << 16) |
+                ((int)Asc[(long)(t * 4 + 3) * M + r] << 24);
+    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      long r = n0 + brow + n * 32;
+      sb_w[n] = (int)Bsc[(long)(t * 4 + 0) * N + r] |
+                ((int)Bsc[(long)(t * 4 + 1) * N + r] << 8) |
+                ((int)Bsc[(long)(t * 4 + 2) * N + r] << 16) |
+                ((int)Bsc[(long)(t * 4 + 3) * N + r] << 24);
+    }
+  };
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  int sa_cur[4], sb_cur[2], sa_nxt[4], sb_nxt[2];
+  load_scales(0, sa_cur, sb_cur);
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  const int NT = K / 128;
+  for (int t = 0; t < NT; ++t) {
+    const bool pre = (t + 1 < NT);
+    if (pre) {
+      stage(cur ^ 1, t + 1);
+      load_scales(t + 1, sa_nxt, sb_nxt);  // covered by the tile body
+    }
+    const unsigned char* a_base = As + cur * MXBM * 128;
+    const unsigned char* b_base = Bs + cur * MXBM * 128;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      v8i af[4], bf[2];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        u32 row = arow + m * 32;
+        u32 byte = row * 128u + (u32)kk * 64u + kh * 16u;
+        byte ^= ((row >> 1) & 7) << 4;
+        af[m] = v8_from((const char*)a_base + byte,
+                        (const char*)a_base + (byte ^ 32u));
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        u32 row = brow + n * 32;
+        u32 byte = row * 128u + (u32)kk * 64u + kh * 16u;
+        byte ^= ((row >> 1) & 7) << 4;
+        bf[n] = v8_from((const char*)b_base + byte,
+                        (const char*)b_base + (byte ^ 32u));
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af[m], bf[n], acc[m][n], 0, 0, 0,
+              (sa_cur[m] >> (8 * (kk * 2 + (int)kh))) & 0xFF, 0,
+              (sb_cur[n] >> (8 * (kk * 2 + (int)kh))) & 0xFF);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int m = 0; m < 4; ++m) sa_cur[m] = sa_nxt[m];
+#pragma unroll
+    for (int n = 0; n < 2; ++n) sb_cur[n] = sb_nxt[n];
+    cur ^= 1;
+  }
+
+  const long ccol = n0 + wn * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      float cv[16];
+      if (!beta0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                     8 * (r >> 2) + 4 * (lane >> 5);
+          cv[r] = C[row * (long)N + ccol + n * 32];
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                   8 * (r >> 2) + 4 * (lane >> 5);
+        C[row * (long)N + ccol + n * 32] =
+            (beta0 ? 0.0f : cv[r]) + acc[m][n][r];
+      }
+    }
+}
+
 static int g_mx8_variant = 0;  // 0 = auto (simple k128), 2 = counted 2-phase
 extern "C" void set_mx8_variant(int v) { g_mx8_variant = v; }
 
@@ -495,6 +647,15 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                                        int beta0, hipStream_t stream) {
   dim3 grid((M / MXBM) * (N / MXBN));
   if (K % 128 == 0) {
+    if (g_mx8_variant == 3) {
+      size_t lds = 4u * MXBM * 128;  // 128 KiB
+      hipLaunchKernelGGL(gemm_bt_mx8_acc_k128v3, grid, dim3(512), lds,
+                         stream, C, (const unsigned char*)A,
+                         (const unsigned char*)Asc,
+                         (const unsigned char*)Bt,
+                         (const unsigned char*)Bsc, M, N, K, beta0);
+      return;
+    }
     if (g_mx8_variant == 2) {
       size_t lds = 2u * MXCV_BUF;  // 132 KiB
       hipLaunchKernelGGL(gemm_bt_mx8_acc_k128cv, grid, dim3(512), lds,
