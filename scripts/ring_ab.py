@@ -115,8 +115,8 @@ if __name__ == "__main__":
     if not ok:
         print("SCREEN FAILED - not running perf")
         sys.exit(1)
-    perf(8192, 8, 3)
-    perf(16384, 3, 2)
+    perf(8192, 8, 3, vs=(6, 2, 7, 16))
+    perf(16384, 3, 2, vs=(6, 2, 7, 16))
     perf_final(16384, 3, 2)
     if os.environ.get("RING_AB_BIG") == "1":
         perf_final(65536, 1, 2)
