@@ -647,7 +647,9 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                                        int beta0, hipStream_t stream) {
   dim3 grid((M / MXBM) * (N / MXBN));
   if (K % 128 == 0) {
-    if (g_mx8_variant == 3) {
+    // v3 (scale prefetch) is the default: +5% at 8192^3, ties at 16384^3
+    // (gpurun r2 call 2); variant 1 selects the old simple kernel.
+    if (g_mx8_variant == 0 || g_mx8_variant == 3) {
       size_t lds = 4u * MXBM * 128;  // 128 KiB
       hipLaunchKernelGGL(gemm_bt_mx8_acc_k128v3, grid, dim3(512), lds,
                          stream, C, (const unsigned char*)A,
