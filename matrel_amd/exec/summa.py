@@ -34,6 +34,21 @@ def _presence_map(m: BlockMatrix) -> torch.Tensor:
     return m.presence()
 
 
+_prep_stream: torch.cuda.Stream = None
+
+
+def _get_prep_stream():
+    """Dedicated HIP stream for owner-side panel preparation
+    (transposes, pads, mx8 conversions, CSR wire packing): prep work for
+    step k+1 runs UNDER the step-k GEMMs already enqueued on the compute
+    stream instead of serializing between their launches (SURVEY.md §2.2
+    "host DAG executor over HIP streams"; ROADMAP r1 item 3)."""
+    global _prep_stream
+    if _prep_stream is None:
+        _prep_stream = torch.cuda.Stream()
+    return _prep_stream
+
+
 def _cache_has_headroom(nbytes: int) -> bool:
     """Only grow a convenience cache while the device keeps at least 4x
     the entry size PLUS 16 GiB free - caches are a speedup, not worth an
@@ -384,6 +399,33 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     p, q = rt.grid_row, rt.grid_col
     gk = a.grid_cols
     assert gk == b.grid_rows
+    # Owner-side panel prep (pads/transposes/mx8 conversions/CSR packs)
+    # runs on a second HIP stream so it overlaps the already-enqueued
+    # GEMMs of the previous step; broadcasts wait on it via events.
+    use_prep = (a.device.type == "cuda" and get_config().overlap_comm
+                and not get_config().deterministic)
+    prep = _get_prep_stream() if use_prep else None
+    if prep is not None:
+        _ev0 = torch.cuda.Event()
+        _ev0.record()
+        prep.wait_event(_ev0)  # prep ordered after operand production
+
+    def _prep_ctx():
+        import contextlib
+        return (torch.cuda.stream(prep) if prep is not None
+                else contextlib.nullcontext())
+
+    def _prep_reg(*ts):
+        if prep is not None:
+            for t in ts:
+                t.record_stream(prep)
+
+    def _prep_done():
+        if prep is not None:
+            _e = torch.cuda.Event()
+            _e.record(prep)
+            torch.cuda.current_stream().wait_event(_e)
+
     pa, pb = _presence_map(a), _presence_map(b)
     # Global stored-nnz maps: drive the group-uniform dense-vs-CSR wire
     # decision for BOTH panel directions (element-sparse blocks broadcast
@@ -441,12 +483,16 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                                       dtype=torch.uint8, device=a.device)
                 a_sc = torch.empty(len(a_rows), bs // 32, bs,
                                    dtype=torch.uint8, device=a.device)
+                _prep_reg(a_stack, a_sc)
                 if rt.rank == a_src:
-                    for n, i in enumerate(a_rows):
-                        p8, sc = _mx8_padded(a, ("s", i, k),
-                                             a.get_block_dense(i, k), bs)
-                        a_stack[n] = p8
-                        a_sc[n] = sc
+                    with _prep_ctx():
+                        for n, i in enumerate(a_rows):
+                            p8, sc = _mx8_padded(a, ("s", i, k),
+                                                 a.get_block_dense(i, k),
+                                                 bs)
+                            a_stack[n] = p8
+                            a_sc[n] = sc
+                    _prep_done()
                 # NCCL executes ops of one communicator in enqueue order,
                 # so waiting the (later) scale broadcast also fences the
                 # payload broadcast.
@@ -469,13 +515,16 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 # receivers; the source zero-pads only ragged edge blocks
                 a_stack = torch.empty(len(dn_rows), bs, bs, dtype=a.dtype,
                                       device=a.device)
+                _prep_reg(a_stack)
                 if rt.rank == a_src:
-                    for n, i in enumerate(dn_rows):
-                        t = blockops.densify(a.get_block_dense(i, k))
-                        h, w = t.shape
-                        if (h, w) != (bs, bs):
-                            a_stack[n].zero_()
-                        a_stack[n, :h, :w] = t
+                    with _prep_ctx():
+                        for n, i in enumerate(dn_rows):
+                            t = blockops.densify(a.get_block_dense(i, k))
+                            h, w = t.shape
+                            if (h, w) != (bs, bs):
+                                a_stack[n].zero_()
+                            a_stack[n, :h, :w] = t
+                    _prep_done()
                 if len(dn_rows):
                     waits.append(dist.broadcast(a_stack, src=a_src,
                                                 group=rt.row_group(p),
@@ -495,21 +544,24 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                                           device=a.device)
                     val_buf = torch.empty(val_len, dtype=a.dtype,
                                           device=a.device)
+                    _prep_reg(idx_buf, val_buf)
                     if rt.rank == a_src:
-                        io = vo = 0
-                        for i in sp_rows:
-                            t = a.blocks[(i, k)]
-                            if not blockops.is_sparse(t):
-                                t = t.to_sparse_csr()
-                            h = shapes[i][0]
-                            nz = nnzs[i]
-                            idx_buf[io:io + h + 1] = \
-                                t.crow_indices().to(torch.int32)
-                            idx_buf[io + h + 1:io + h + 1 + nz] = \
-                                t.col_indices().to(torch.int32)
-                            val_buf[vo:vo + nz] = t.values()
-                            io += h + 1 + nz
-                            vo += nz
+                        with _prep_ctx():
+                            io = vo = 0
+                            for i in sp_rows:
+                                t = a.blocks[(i, k)]
+                                if not blockops.is_sparse(t):
+                                    t = t.to_sparse_csr()
+                                h = shapes[i][0]
+                                nz = nnzs[i]
+                                idx_buf[io:io + h + 1] = \
+                                    t.crow_indices().to(torch.int32)
+                                idx_buf[io + h + 1:io + h + 1 + nz] = \
+                                    t.col_indices().to(torch.int32)
+                                val_buf[vo:vo + nz] = t.values()
+                                io += h + 1 + nz
+                                vo += nz
+                        _prep_done()
                     waits.append(dist.broadcast(idx_buf, src=a_src,
                                                 group=rt.row_group(p),
                                                 async_op=True))
@@ -552,14 +604,17 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                                       dtype=torch.uint8, device=b.device)
                 b_sc = torch.empty(len(b_cols), bs // 32, bs,
                                    dtype=torch.uint8, device=b.device)
+                _prep_reg(b_stack, b_sc)
                 if rt.rank == b_src:
-                    for n, j in enumerate(b_cols):
-                        t = _pad_t(blockops.densify(
-                            b.get_block_dense(k, j)), bs)
-                        p8, sc = _mx8_padded(b, ("st", k, j),
-                                             blockops.transpose(t), bs)
-                        b_stack[n] = p8
-                        b_sc[n] = sc
+                    with _prep_ctx():
+                        for n, j in enumerate(b_cols):
+                            t = _pad_t(blockops.densify(
+                                b.get_block_dense(k, j)), bs)
+                            p8, sc = _mx8_padded(b, ("st", k, j),
+                                                 blockops.transpose(t), bs)
+                            b_stack[n] = p8
+                            b_sc[n] = sc
+                    _prep_done()
                 waits.append(dist.broadcast(b_stack, src=b_src,
                                             group=rt.col_group(q),
                                             async_op=True))
@@ -575,19 +630,22 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 b_dn_index = {j: n for n, j in enumerate(dn_cols)}
                 b_stack = torch.empty(len(dn_cols), bs, bs, dtype=b.dtype,
                                       device=b.device)
+                _prep_reg(b_stack)
                 if rt.rank == b_src:
-                    for n, j in enumerate(dn_cols):
-                        t = blockops.densify(b.get_block_dense(k, j))
-                        h, w = t.shape
-                        if (h, w) != (bs, bs):
-                            pad = torch.zeros(bs, bs, dtype=t.dtype,
-                                              device=t.device)
-                            pad[:h, :w] = t
-                            t = pad
-                        if t.is_cuda:
-                            b_stack[n] = blockops.transpose(t)
-                        else:
-                            b_stack[n] = t.t()
+                    with _prep_ctx():
+                        for n, j in enumerate(dn_cols):
+                            t = blockops.densify(b.get_block_dense(k, j))
+                            h, w = t.shape
+                            if (h, w) != (bs, bs):
+                                pad = torch.zeros(bs, bs, dtype=t.dtype,
+                                                  device=t.device)
+                                pad[:h, :w] = t
+                                t = pad
+                            if t.is_cuda:
+                                b_stack[n] = blockops.transpose(t)
+                            else:
+                                b_stack[n] = t.t()
+                    _prep_done()
                 if len(dn_cols):
                     waits.append(dist.broadcast(b_stack, src=b_src,
                                                 group=rt.col_group(q),
@@ -604,21 +662,24 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                                           device=b.device)
                     val_buf = torch.empty(val_len, dtype=b.dtype,
                                           device=b.device)
+                    _prep_reg(idx_buf, val_buf)
                     if rt.rank == b_src:
-                        io = vo = 0
-                        for j in sp_cols:
-                            t = b.blocks[(k, j)]
-                            if not blockops.is_sparse(t):
-                                t = t.to_sparse_csr()
-                            h = shapes[j][0]
-                            nz = nnzs[j]
-                            idx_buf[io:io + h + 1] = \
-                                t.crow_indices().to(torch.int32)
-                            idx_buf[io + h + 1:io + h + 1 + nz] = \
-                                t.col_indices().to(torch.int32)
-                            val_buf[vo:vo + nz] = t.values()
-                            io += h + 1 + nz
-                            vo += nz
+                        with _prep_ctx():
+                            io = vo = 0
+                            for j in sp_cols:
+                                t = b.blocks[(k, j)]
+                                if not blockops.is_sparse(t):
+                                    t = t.to_sparse_csr()
+                                h = shapes[j][0]
+                                nz = nnzs[j]
+                                idx_buf[io:io + h + 1] = \
+                                    t.crow_indices().to(torch.int32)
+                                idx_buf[io + h + 1:io + h + 1 + nz] = \
+                                    t.col_indices().to(torch.int32)
+                                val_buf[vo:vo + nz] = t.values()
+                                io += h + 1 + nz
+                                vo += nz
+                        _prep_done()
                     waits.append(dist.broadcast(idx_buf, src=b_src,
                                                 group=rt.col_group(q),
                                                 async_op=True))
