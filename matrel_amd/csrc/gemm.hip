@@ -1548,7 +1548,7 @@ gemm_bt_bf16_acc_256w4p_k(float* __restrict__ C,
 
 static int g_gemm_variant = 0;  // 0=auto, 1=simple 256, 2=phased 256, 3=128
 extern "C" void set_gemm_variant(int v) { g_gemm_variant = v; }
-static unsigned g_st_m = 4, g_st_n = 4;  // supertile cluster dims
+unsigned g_st_m = 4, g_st_n = 4;  // supertile cluster dims (shared with gemm_mx8.hip)
 extern "C" void set_supertile(int sm, int sn) {
   g_st_m = (unsigned)sm;
   g_st_n = (unsigned)sn;

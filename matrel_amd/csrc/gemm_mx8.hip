@@ -23,6 +23,8 @@
 // swizzle on the 64-B fp8 rows, double-buffered LDS.
 #include "common.h"
 
+extern unsigned g_st_m, g_st_n;  // supertile dims (gemm.hip)
+
 typedef int v8i __attribute__((ext_vector_type(8)));
 typedef int v4i __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
@@ -501,7 +503,7 @@ gemm_bt_mx8_acc_k128v3(float* __restrict__ C,
                        const unsigned char* __restrict__ Asc,
                        const unsigned char* __restrict__ Bt,
                        const unsigned char* __restrict__ Bsc, int M, int N,
-                       int K, int beta0) {
+                       int K, int beta0, u32 stm_, u32 stn_) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned char* As = (unsigned char*)smem;  // [2][256*128]
   unsigned char* Bs = (unsigned char*)(smem + 2 * MXBM * 128);
@@ -513,8 +515,8 @@ gemm_bt_mx8_acc_k128v3(float* __restrict__ C,
   const u32 ntm = (u32)(M / MXBM), ntn = (u32)(N / MXBN);
   u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
   u32 tm, tn;
-  if (ntm % 4 == 0 && ntn % 4 == 0) {
-    supertile_map(bid, ntm, ntn, 4, 4, &tm, &tn);
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
   } else {
     tm = bid / ntn;
     tn = bid % ntn;
@@ -655,7 +657,8 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                          stream, C, (const unsigned char*)A,
                          (const unsigned char*)Asc,
                          (const unsigned char*)Bt,
-                         (const unsigned char*)Bsc, M, N, K, beta0);
+                         (const unsigned char*)Bsc, M, N, K, beta0,
+                         g_st_m, g_st_n);
       return;
     }
     if (g_mx8_variant == 2) {

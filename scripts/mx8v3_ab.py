@@ -68,6 +68,30 @@ def perf(n, it=3, reps=3):
     del a, bt, a8, asc, bt8, btsc, c
 
 
+def st_sweep(n, it=3):
+    """Supertile sweep for the (possibly HBM-bound) mx8 v3 kernel: at
+    256 fp8 flops/byte the kernel sits near the bandwidth knee, so L2
+    strip reuse through the supertile shape is a first-order lever."""
+    a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
+    bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
+    a8, asc = blockops.to_mx8(a)
+    bt8, btsc = blockops.to_mx8(bt)
+    c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
+    ext.set_mx8_variant(3)
+    for sm, sn in ((4, 4), (8, 4), (4, 8), (8, 8), (2, 8), (16, 4)):
+        ext.set_supertile(sm, sn)
+        ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(it):
+            ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True)
+        torch.cuda.synchronize()
+        tf = 2 * n ** 3 / ((time.perf_counter() - t0) / it) / 1e12
+        print(f"mx8 st({sm},{sn}) n={n}: {tf:.0f} TF", flush=True)
+    ext.set_supertile(4, 4)
+    del a, bt, a8, asc, bt8, btsc, c
+
+
 if __name__ == "__main__":
     ok = True
     for n in (256, 512, 4096):
@@ -77,5 +101,8 @@ if __name__ == "__main__":
         sys.exit(1)
     perf(8192, it=6)
     perf(16384, it=3)
+    if os.environ.get("MX8_ST_SWEEP") == "1":
+        st_sweep(16384)
+        st_sweep(32768)
     ext.set_mx8_variant(0)
     print("done")

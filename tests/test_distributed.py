@@ -345,6 +345,25 @@ def _body_ewise_move_smaller(s, tmpdir):
     assert moved < 0.75 * dense_bytes, (moved, dense_bytes)
 
 
+def _body_bcast_cost_no_threshold(s, tmpdir):
+    """replication candidacy is cost-based, not a
+    1-GiB constant. A replicable operand far above 1 GiB must still win
+    broadcast-MM when its bytes beat the SUMMA panel estimate (plan
+    shape only - no data is materialized)."""
+    from matrel_amd.blocks.matrix import BlockMatrix
+    from matrel_amd.blocks.schemes import ROW
+    from matrel_amd.plan import logical as L
+    from matrel_amd.plan.physical import plan_physical
+    # A: 2M x 64k row-partitioned (dense hint), B: 64k^2 = 8.6 GB bf16
+    # (>> the old 1-GiB threshold, << the 16-GiB memory cap).
+    A = L.Leaf(BlockMatrix(2_000_000, 65536, block_size=16384,
+                           scheme=ROW), density=1.0)
+    B = L.Leaf(BlockMatrix(65536, 65536, block_size=16384), density=1.0)
+    mm = L.MatMul(A, B)
+    pp = plan_physical(mm)
+    assert pp.get(mm).mm_algo == "bcast_b", pp.get(mm).mm_algo
+
+
 def _body_ols(s, tmpdir):
     X = s.randn(128, 16, seed=11)
     y = s.randn(128, 1, seed=12)
@@ -365,6 +384,7 @@ def _body_ols(s, tmpdir):
     "_body_deterministic_serialized", "_body_mx8_config",
     "_body_save_load", "_body_partition_aware_local", "_body_ols",
     "_body_maxmin_empty_rank", "_body_ewise_move_smaller",
+    "_body_bcast_cost_no_threshold",
 ])
 def test_distributed(body, tmp_path):
     _run(body, tmp_path)
