@@ -106,3 +106,30 @@ if __name__ == "__main__":
         st_sweep(32768)
     ext.set_mx8_variant(0)
     print("done")
+
+
+def zf_diag(n=16384, it=3):
+    """Zero-fill vs random diagnostic (DVFS give-back test, microarch
+    guide "DVFS give-back" item 1): if zero-filled operands run much
+    faster at the same kernel, the random-data number is POWER-bound
+    (toggling energy), not issue/pipe-bound - structural work cannot
+    recover it."""
+    ext.set_mx8_variant(3)
+    for name, make in (("randn", lambda: torch.randn(n, n, device="cuda")
+                        .to(torch.bfloat16)),
+                       ("zeros", lambda: torch.zeros(n, n, device="cuda",
+                                                     dtype=torch.bfloat16))):
+        a = make()
+        bt = make()
+        a8, asc = blockops.to_mx8(a)
+        bt8, btsc = blockops.to_mx8(bt)
+        c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
+        ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(it):
+            ext.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc, True)
+        torch.cuda.synchronize()
+        tf = 2 * n ** 3 / ((time.perf_counter() - t0) / it) / 1e12
+        print(f"mx8 zf-diag {name} n={n}: {tf:.0f} TF", flush=True)
+        del a, bt, a8, asc, bt8, btsc, c

@@ -122,3 +122,24 @@ if __name__ == "__main__":
         perf_final(65536, 1, 2)
     ext.set_gemm_variant(0)
     print("done")
+
+
+def zf_diag(n=16384, it=3):
+    """Zero-fill vs random for v6 (power-bound diagnostic)."""
+    ext.set_gemm_variant(6)
+    for name, make in (("randn", lambda: torch.randn(n, n, device="cuda")
+                        .to(torch.bfloat16)),
+                       ("zeros", lambda: torch.zeros(n, n, device="cuda",
+                                                     dtype=torch.bfloat16))):
+        a = make()
+        bt = make()
+        c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
+        ext.gemm_bt_acc(c, a, bt)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(it):
+            ext.gemm_bt_acc(c, a, bt)
+        torch.cuda.synchronize()
+        tf = 2 * n ** 3 / ((time.perf_counter() - t0) / it) / 1e12
+        print(f"bf16 v6 zf-diag {name} n={n}: {tf:.0f} TF", flush=True)
+        del a, bt, c
