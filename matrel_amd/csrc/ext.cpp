@@ -365,7 +365,12 @@ std::vector<torch::Tensor> mx8_convert(torch::Tensor a) {
   TORCH_CHECK(cols % 32 == 0, "cols must be a multiple of 32");
   auto opts = torch::TensorOptions().dtype(torch::kUInt8).device(a.device());
   auto out8 = torch::empty({rows, cols}, opts);
-  auto sc = torch::empty({cols / 32, rows}, opts);  // block-major
+  // packed-dword scale layout when cols % 128 == 0 (see mxconvert.hip)
+  auto sc = (cols % 128 == 0)
+      ? torch::empty({cols / 128, rows},
+                     torch::TensorOptions().dtype(torch::kInt32)
+                         .device(a.device()))
+      : torch::empty({cols / 32, rows}, opts);
   launch_mx8_convert(out8.data_ptr(), sc.data_ptr(), a.data_ptr(), rows,
                      cols, cur_stream());
   return {out8, sc};
@@ -384,10 +389,22 @@ void gemm_bt_mx8_acc(torch::Tensor C, torch::Tensor A8, torch::Tensor Asc,
   TORCH_CHECK(Bt8.size(1) == K && C.size(0) == M && C.size(1) == N);
   TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
               "mx8 GEMM needs M,N % 256 == 0, K % 64 == 0");
-  TORCH_CHECK(Asc.numel() == (long)M * (K / 32) &&
-              Bsc.numel() == (long)N * (K / 32));
-  TORCH_CHECK(Asc.size(0) == K / 32 && Bsc.size(0) == K / 32,
-              "scales must be block-major [K/32][rows]");
+  if (K % 128 == 0) {
+    // packed-dword layout [K/128][rows] int32 (mx8_convert emits it for
+    // cols % 128 == 0 inputs)
+    TORCH_CHECK(Asc.scalar_type() == torch::kInt32 &&
+                Bsc.scalar_type() == torch::kInt32,
+                "K % 128 == 0 expects packed int32 scales from "
+                "mx8_convert");
+    TORCH_CHECK(Asc.size(0) == K / 128 && Asc.size(1) == M &&
+                Bsc.size(0) == K / 128 && Bsc.size(1) == N,
+                "packed scales must be [K/128][rows] int32");
+  } else {
+    TORCH_CHECK(Asc.numel() == (long)M * (K / 32) &&
+                Bsc.numel() == (long)N * (K / 32));
+    TORCH_CHECK(Asc.size(0) == K / 32 && Bsc.size(0) == K / 32,
+                "scales must be block-major [K/32][rows]");
+  }
   launch_gemm_bt_mx8_acc(C.data_ptr<float>(), A8.data_ptr(), Asc.data_ptr(),
                          Bt8.data_ptr(), Bsc.data_ptr(), M, N, K,
                          beta0 ? 1 : 0, cur_stream());

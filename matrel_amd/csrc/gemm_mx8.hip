@@ -238,22 +238,13 @@ gemm_bt_mx8_acc_k128(float* __restrict__ C,
     const unsigned char* b_base = Bs + cur * MXBM * 128;
     // one dword of scales per fragment row: blocks [4t, 4t+4)
     int sa_w[4], sb_w[2];  // byte b of the word = scale of block 4t+b
+    // packed-dword scales [K/128][rows]: ONE coalesced load per word
 #pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      long r = m0 + arow + m * 32;
-      sa_w[m] = (int)Asc[(long)(t * 4 + 0) * M + r] |
-                ((int)Asc[(long)(t * 4 + 1) * M + r] << 8) |
-                ((int)Asc[(long)(t * 4 + 2) * M + r] << 16) |
-                ((int)Asc[(long)(t * 4 + 3) * M + r] << 24);
-    }
+    for (int m = 0; m < 4; ++m)
+      sa_w[m] = ((const int*)Asc)[(long)t * M + m0 + arow + m * 32];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
-      long r = n0 + brow + n * 32;
-      sb_w[n] = (int)Bsc[(long)(t * 4 + 0) * N + r] |
-                ((int)Bsc[(long)(t * 4 + 1) * N + r] << 8) |
-                ((int)Bsc[(long)(t * 4 + 2) * N + r] << 16) |
-                ((int)Bsc[(long)(t * 4 + 3) * N + r] << 24);
-    }
+    for (int n = 0; n < 2; ++n)
+      sb_w[n] = ((const int*)Bsc)[(long)t * N + n0 + brow + n * 32];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       v8i af[4], bf[2];
@@ -545,22 +536,13 @@ gemm_bt_mx8_acc_k128v3(float* __restrict__ C,
   const u32 kh = lane >> 5;
 
   auto load_scales = [&](int t, int* sa_w, int* sb_w) {
+    // packed-dword scales [K/128][rows]: ONE coalesced load per word
 #pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      long r = m0 + arow + m * 32;
-      sa_w[m] = (int)Asc[(long)(t * 4 + 0) * M + r] |
-                ((int)Asc[(long)(t * 4 + 1) * M + r] << 8) |
-                ((int)Asc[(long)(t * 4 + 2) * M + r] << 16) |
-                ((int)Asc[(long)(t * 4 + 3) * M + r] << 24);
-    }
+    for (int m = 0; m < 4; ++m)
+      sa_w[m] = ((const int*)Asc)[(long)t * M + m0 + arow + m * 32];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
-      long r = n0 + brow + n * 32;
-      sb_w[n] = (int)Bsc[(long)(t * 4 + 0) * N + r] |
-                ((int)Bsc[(long)(t * 4 + 1) * N + r] << 8) |
-                ((int)Bsc[(long)(t * 4 + 2) * N + r] << 16) |
-                ((int)Bsc[(long)(t * 4 + 3) * N + r] << 24);
-    }
+    for (int n = 0; n < 2; ++n)
+      sb_w[n] = ((const int*)Bsc)[(long)t * N + n0 + brow + n * 32];
   };
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -640,6 +622,221 @@ gemm_bt_mx8_acc_k128v3(float* __restrict__ C,
     }
 }
 
+// ---- BK=128 QUADRANT-PHASE (v4): the bf16 flagship structure ported ----
+// to MX-fp8, enabled by two round-2 measurements: (1) the zero-fill
+// diagnostic shows mx8 is NOT power-bound (zeros 1706 vs randn 1596 TF,
+// +7% - vs bf16's +31%), i.e. its ~47% pipe utilization is REAL
+// recoverable headroom; (2) packed-dword scales cut the per-tile scale
+// traffic to 6 coalesced loads, making a counted-vmcnt ladder possible
+// without byte-load queue pollution. Register plan is DELIBERATELY
+// lighter than r1's parked port (which spilled): per quadrant a wave
+// owns a 32 x 64 slice = 1 A fragment x 2 B fragments of 32x32, so the
+// rotating fragment file is af[2](kk) + bfr[2][2] = 48 VGPRs next to
+// the 128-VGPR accumulator file.
+//
+// Phases (quadrant snake Q0=(0,0) Q1=(0,1) Q2=(1,1) Q3=(1,0)); per
+// K-tile of 128, each phase = 2 kk x 2 n mfma_scale_32x32x64. LDS:
+// halves A0/A1/B0/B1 of 128 rows x 128 B = 16 KiB; x2 buffers = 128 K.
+// Staging order A0',B0',B1',A1' at phases 1..4 with scale words for
+// t+1 issued in phase 1 - vmcnt ladder (queue oldest->newest at each
+// wait, 2 glds per stage, 6 scale dwords):
+//   p1-end: prove B1(t):  in flight {A1(t)2, A0'2, sc6} -> vmcnt(10)
+//   p2-end: prove A1(t):  in flight {A0'2, sc6, B0'2}   -> vmcnt(10)
+//   p3-end: none (p4 re-reads B0(t), proven at the tile boundary)
+//   p4-end: prove A0',B0' (sc older than B0'): leave {B1'2, A1'2}
+//           -> vmcnt(4)
+// Last tile (!pre): p1 vmcnt(2), p2 vmcnt(0), p4 none.
+
+#define QHOFF(h) ((u32)(h) * (128u * 128u))
+#define QBUFSZ (4u * 128u * 128u)
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx8_acc_k128q(float* __restrict__ C,
+                      const unsigned char* __restrict__ A,
+                      const unsigned char* __restrict__ Asc,
+                      const unsigned char* __restrict__ Bt,
+                      const unsigned char* __restrict__ Bsc, int M, int N,
+                      int K, int beta0, u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 qr = (wave & 3) * 32;   // row slice within the 128x128 quad
+  const u32 qc = (wave >> 2) * 64;  // col slice
+
+  const u32 ntm = (u32)(M / MXBM), ntn = (u32)(N / MXBN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * MXBM, n0 = (long)tn * MXBN;
+
+  // staging: half = 128 rows x 128 B = 1024 x 16-B chunks; 512 threads
+  // x 2 glds; swizzle slot ^ ((row>>1)&7) as in the k128 kernels.
+  const u32 srow = tid >> 3, sslot = tid & 7;  // rows 0..63 per pass
+
+  auto stage_half = [&](u32 bufoff, int h, int t) {
+    const long k0 = (long)t * 128;
+    const unsigned char* src = (h < 2)
+        ? A + (m0 + (h & 1) * 128) * (long)K
+        : Bt + (n0 + (h & 1) * 128) * (long)K;
+    unsigned char* dst = (unsigned char*)(smem + bufoff + QHOFF(h));
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      u32 row = srow + it * 64;
+      u32 ss = sslot ^ ((row >> 1) & 7);
+      glds16(src + row * (long)K + k0 + ss * 16,
+             dst + row * 128u + sslot * 16u);
+    }
+  };
+
+  f32x16 acc[4][2] = {};  // [quadrant][n-frag]
+  v8i af[2];              // A fragments per kk
+  v8i bfr[2][2];          // B fragments per kk x n
+
+  const u32 arow = qr + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  auto read_a_kk = [&](u32 off, int kk) {
+    u32 byte = arow * 128u + (u32)kk * 64u + kh * 16u;
+    byte ^= ((arow >> 1) & 7) << 4;
+    af[kk] = v8_from((const char*)smem + off + byte,
+                     (const char*)smem + off + (byte ^ 32u));
+  };
+  auto read_b_kk = [&](u32 off, int kk) {
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      u32 row = qc + n * 32 + (lane & 31);
+      u32 byte = row * 128u + (u32)kk * 64u + kh * 16u;
+      byte ^= ((row >> 1) & 7) << 4;
+      bfr[kk][n] = v8_from((const char*)smem + off + byte,
+                           (const char*)smem + off + (byte ^ 32u));
+    }
+  };
+
+  // packed-dword scales: per tile, A words for the two A halves (row
+  // slice per wave) + B words for the two B halves x two n frags.
+  int sa_cur[2], sb_cur[2][2], sa_nxt[2], sb_nxt[2][2];
+  auto load_scales = [&](int t, int* sa, int sb[2][2]) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+      sa[h] = ((const int*)Asc)[(long)t * M + m0 + h * 128 + arow];
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        sb[h][n] = ((const int*)Bsc)[(long)t * N + n0 + h * 128 + qc
+                                     + n * 32 + (lane & 31)];
+  };
+
+#define MFMA_Q(q, ah, bh, kk)                                              \
+  _Pragma("unroll") for (int n = 0; n < 2; ++n)                            \
+      acc[q][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(         \
+          af[kk], bfr[kk][n], acc[q][n], 0, 0, 0,                          \
+          (sa_cur[ah] >> (8 * ((kk) * 2 + (int)kh))) & 0xFF, 0,            \
+          (sb_cur[bh][n] >> (8 * ((kk) * 2 + (int)kh))) & 0xFF);
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  // prologue: stage tile 0 fully, then its scales; prove A0,B0
+  stage_half(0u, 0, 0);
+  stage_half(0u, 2, 0);
+  stage_half(0u, 3, 0);
+  stage_half(0u, 1, 0);
+  load_scales(0, sa_cur, sb_cur);
+  asm volatile("s_waitcnt vmcnt(10)" ::: "memory");  // A0,B0 landed
+  __builtin_amdgcn_s_barrier();
+
+  const int NT = K / 128;
+  for (int t = 0; t < NT; ++t) {
+    const u32 cur = (u32)(t & 1) * QBUFSZ, nxt = cur ^ QBUFSZ;
+    const bool pre = (t + 1 < NT);
+    // ---- phase 1: Q0 = A0 x B0; stage A0'; load scales(t+1)
+    read_a_kk(cur + QHOFF(0), 0);
+    read_b_kk(cur + QHOFF(2), 0);
+    read_a_kk(cur + QHOFF(0), 1);
+    read_b_kk(cur + QHOFF(2), 1);
+    if (pre) {
+      stage_half(nxt, 0, t + 1);
+      load_scales(t + 1, sa_nxt, sb_nxt);
+    }
+    MFMA_Q(0, 0, 0, 0)
+    MFMA_Q(0, 0, 0, 1)
+    if (pre) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 2: Q1 = A0 x B1 (A frags reused); stage B0'
+    read_b_kk(cur + QHOFF(3), 0);
+    read_b_kk(cur + QHOFF(3), 1);
+    if (pre) stage_half(nxt, 2, t + 1);
+    MFMA_Q(1, 0, 1, 0)
+    MFMA_Q(1, 0, 1, 1)
+    if (pre) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 3: Q2 = A1 x B1 (B frags reused); stage B1'
+    read_a_kk(cur + QHOFF(1), 0);
+    read_a_kk(cur + QHOFF(1), 1);
+    if (pre) stage_half(nxt, 3, t + 1);
+    MFMA_Q(2, 1, 1, 0)
+    MFMA_Q(2, 1, 1, 1)
+    // no wait/barrier: phase 4 re-reads B0(cur), proven at tile entry,
+    // and stages into the OTHER buffer
+    // ---- phase 4: Q3 = A1 x B0 (re-read); stage A1'
+    read_b_kk(cur + QHOFF(2), 0);
+    read_b_kk(cur + QHOFF(2), 1);
+    if (pre) stage_half(nxt, 1, t + 1);
+    MFMA_Q(3, 1, 0, 0)
+    MFMA_Q(3, 1, 0, 1)
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      sa_cur[h] = sa_nxt[h];
+#pragma unroll
+      for (int n = 0; n < 2; ++n) sb_cur[h][n] = sb_nxt[h][n];
+    }
+  }
+
+  // epilogue: 32x32 C/D map per quadrant slice
+  const int QMHX[4] = {0, 0, 1, 1};
+  const int QNHX[4] = {0, 1, 1, 0};
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    float cv[2][16];
+    if (!beta0) {
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + QMHX[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                     4 * (lane >> 5);
+          long col = n0 + QNHX[q] * 128 + qc + n * 32 + (lane & 31);
+          cv[n][r] = C[row * (long)N + col];
+        }
+    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + QMHX[q] * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        long col = n0 + QNHX[q] * 128 + qc + n * 32 + (lane & 31);
+        C[row * (long)N + col] =
+            (beta0 ? 0.0f : cv[n][r]) + acc[q][n][r];
+      }
+  }
+}
+#undef MFMA_Q
+#undef QHOFF
+#undef QBUFSZ
+
 static int g_mx8_variant = 0;  // 0 = auto (simple k128), 2 = counted 2-phase
 extern "C" void set_mx8_variant(int v) { g_mx8_variant = v; }
 
@@ -649,6 +846,16 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                                        int beta0, hipStream_t stream) {
   dim3 grid((M / MXBM) * (N / MXBN));
   if (K % 128 == 0) {
+    if (g_mx8_variant == 4) {
+      size_t lds = 2u * 4u * 128u * 128u;  // 128 KiB
+      hipLaunchKernelGGL(gemm_bt_mx8_acc_k128q, grid, dim3(512), lds,
+                         stream, C, (const unsigned char*)A,
+                         (const unsigned char*)Asc,
+                         (const unsigned char*)Bt,
+                         (const unsigned char*)Bsc, M, N, K, beta0,
+                         g_st_m, g_st_n);
+      return;
+    }
     // v3 (scale prefetch) is the default: +5% at 8192^3, ties at 16384^3
     // (gpurun r2 call 2); variant 1 selects the old simple kernel.
     if (g_mx8_variant == 0 || g_mx8_variant == 3) {
@@ -661,7 +868,8 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                          g_st_m, g_st_n);
       return;
     }
-    if (g_mx8_variant == 2) {
+    if (false && g_mx8_variant == 2) {  // cv reads the legacy byte
+      // scale layout; parked (slower anyway - profiles/README.md)
       size_t lds = 2u * MXCV_BUF;  // 132 KiB
       hipLaunchKernelGGL(gemm_bt_mx8_acc_k128cv, grid, dim3(512), lds,
                          stream, C, (const unsigned char*)A,

@@ -56,10 +56,16 @@ __global__ void mx8_convert_k(unsigned char* __restrict__ out8,
     }
   }
   int e;
-  // scales stored BLOCK-MAJOR: sc[block][row] (contiguous rows per
-  // 32-wide k-block) so GEMM kernels can stage a tile's scales with
-  // global_load_lds
-  long sidx = b * rows + r;
+  // Scale layout (round 2): when cols % 128 == 0 (every engine path -
+  // blocks are % 256), scales are PACKED DWORDS sc32[k_tile][row] with
+  // byte b128 = block 4*k_tile + b128 - the BK=128 GEMM kernels then
+  // load ONE coalesced dword per fragment row per K-tile instead of 4
+  // strided bytes (24 strided byte loads/tile was a measured stall).
+  // Otherwise the legacy block-major byte layout sc[block][row] feeds
+  // the BK=64 kernel.
+  long sidx = (cols % 128 == 0)
+      ? ((b >> 2) * rows + r) * 4 + (b & 3)
+      : b * rows + r;
   if (mx == 0.0f) {
     e = 0;
     sc[sidx] = 127;  // scale 1.0; all-zero payload

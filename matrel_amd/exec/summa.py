@@ -481,8 +481,9 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             if mx8:
                 a_stack = torch.empty(len(a_rows), bs, bs,
                                       dtype=torch.uint8, device=a.device)
-                a_sc = torch.empty(len(a_rows), bs // 32, bs,
-                                   dtype=torch.uint8, device=a.device)
+                # packed-dword scales [bs/128][bs] int32 (mxconvert r2)
+                a_sc = torch.empty(len(a_rows), bs // 128, bs,
+                                   dtype=torch.int32, device=a.device)
                 _prep_reg(a_stack, a_sc)
                 if rt.rank == a_src:
                     with _prep_ctx():
@@ -602,8 +603,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             if mx8:
                 b_stack = torch.empty(len(b_cols), bs, bs,
                                       dtype=torch.uint8, device=b.device)
-                b_sc = torch.empty(len(b_cols), bs // 32, bs,
-                                   dtype=torch.uint8, device=b.device)
+                b_sc = torch.empty(len(b_cols), bs // 128, bs,
+                                   dtype=torch.int32, device=b.device)
                 _prep_reg(b_stack, b_sc)
                 if rt.rank == b_src:
                     with _prep_ctx():

@@ -31,12 +31,13 @@ def screen(n, runs=3):
     bt8, btsc = blockops.to_mx8(bt)
     ref = run(0, a8, asc, bt8, btsc, n)
     ok = True
-    for r in range(runs):
-        c = run(3, a8, asc, bt8, btsc, n)
-        if not torch.equal(c, ref):
-            md = (c - ref).abs().max().item()
-            print(f"MISMATCH v3 n={n} run={r}: {md}")
-            ok = False
+    for v in (3, 4):
+        for r in range(runs):
+            c = run(v, a8, asc, bt8, btsc, n)
+            if not torch.equal(c, ref):
+                md = (c - ref).abs().max().item()
+                print(f"MISMATCH v{v} n={n} run={r}: {md}")
+                ok = False
     # sanity vs fp32 torch (loose fp8 band)
     tref = a.float() @ bt.float().t()
     rerr = (ref - tref).abs().max().item() / tref.abs().max().item()
@@ -52,7 +53,7 @@ def perf(n, it=3, reps=3):
     a8, asc = blockops.to_mx8(a)
     bt8, btsc = blockops.to_mx8(bt)
     c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
-    for v in (0, 3):
+    for v in (0, 3, 4):
         ext.set_mx8_variant(v)
         xs = []
         for _ in range(reps):

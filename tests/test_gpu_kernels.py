@@ -301,8 +301,10 @@ def test_mx8_convert_roundtrip(ext):
     """bf16 -> MX-fp8: dequantized values within e4m3 relative error."""
     a = _randn(256, 256, seed=50)
     a8, sc = ext.mx8_convert(a)
-    assert a8.shape == (256, 256) and sc.shape == (8, 256)  # block-major
-    # dequantize on host
+    # packed-dword scale layout for cols % 128 == 0: int32 [K/128][rows],
+    # byte b of word t = e8m0 scale of 32-block 4t+b
+    assert a8.shape == (256, 256)
+    assert sc.shape == (2, 256) and sc.dtype == torch.int32
     import torch as T
     bits = a8.cpu().to(T.int32)
     s = bits >> 7
@@ -311,7 +313,10 @@ def test_mx8_convert_roundtrip(ext):
     val = T.where(be > 0, (2.0 ** (be - 7)) * (1 + m / 8.0),
                   (2.0 ** -6) * (m / 8.0))
     val = T.where(s > 0, -val, val)
-    scale = 2.0 ** (sc.cpu().to(T.int32).float().t() - 127)
+    # unpack: [K/128][rows] int32 -> [K/32][rows] bytes -> [rows][K/32]
+    scb = sc.cpu().view(T.uint8).reshape(2, 256, 4).permute(0, 2, 1) \
+        .reshape(8, 256)
+    scale = 2.0 ** (scb.to(T.int32).float().t() - 127)
     deq = val.float() * scale.repeat_interleave(32, dim=1)
     ref = a.float().cpu()
     err = (deq - ref).abs()
