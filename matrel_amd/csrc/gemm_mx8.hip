@@ -696,32 +696,53 @@ gemm_bt_mx8_acc_k128q(float* __restrict__ C,
   };
 
   f32x16 acc[4][2] = {};  // [quadrant][n-frag]
-  v8i af[2];              // A fragments per kk
-  v8i bfr[2][2];          // B fragments per kk x n
+  // SINGLE-kk fragment sets (24 VGPRs): a 2-kk rotating file spilled.
+  v8i af;       // A fragment (current kk)
+  v8i bfr[2];   // B fragments (current kk)
 
   const u32 arow = qr + (lane & 31);
   const u32 kh = lane >> 5;
 
-  auto read_a_kk = [&](u32 off, int kk) {
-    u32 byte = arow * 128u + (u32)kk * 64u + kh * 16u;
-    byte ^= ((arow >> 1) & 7) << 4;
-    af[kk] = v8_from((const char*)smem + off + byte,
-                     (const char*)smem + off + (byte ^ 32u));
-  };
-  auto read_b_kk = [&](u32 off, int kk) {
+  // SIX per-lane LDS address registers for the whole loop: the swizzled
+  // per-lane byte offsets within a 16-KiB half (lo and ^32 hi for the
+  // A row and the two B rows). Buffer parity and half/kk selection fold
+  // into the ds_read immediate via compile-time constants - the first
+  // two attempts let the compiler keep per-phase address chains and it
+  // spilled acc quadrants (~90 scratch ops in the loop, 358 TF).
+  u32 a_lo, a_hi, b_lo[2], b_hi[2];
+  {
+    u32 ab = arow * 128u + kh * 16u;
+    ab ^= ((arow >> 1) & 7) << 4;
+    a_lo = ab;
+    a_hi = ab ^ 32u;
 #pragma unroll
     for (int n = 0; n < 2; ++n) {
       u32 row = qc + n * 32 + (lane & 31);
-      u32 byte = row * 128u + (u32)kk * 64u + kh * 16u;
-      byte ^= ((row >> 1) & 7) << 4;
-      bfr[kk][n] = v8_from((const char*)smem + off + byte,
-                           (const char*)smem + off + (byte ^ 32u));
+      u32 bb = row * 128u + kh * 16u;
+      bb ^= ((row >> 1) & 7) << 4;
+      b_lo[n] = bb;
+      b_hi[n] = bb ^ 32u;
     }
+  }
+  // h: 0=A0 1=A1 2=B0 3=B1; kk in {0,1}; cur = buffer byte offset
+  auto read_a_kk = [&](u32 cur, int h, int kk) {
+    const char* base = (const char*)smem + cur + a_lo;
+    const char* base_h = (const char*)smem + cur + a_hi;
+    u32 c = QHOFF(h) + (u32)kk * 64u;
+    af = v8_from(base + c, base_h + c);
+  };
+  auto read_b_kk = [&](u32 cur, int h, int kk) {
+    u32 c = QHOFF(h) + (u32)kk * 64u;
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+      bfr[n] = v8_from((const char*)smem + cur + b_lo[n] + c,
+                       (const char*)smem + cur + b_hi[n] + c);
   };
 
-  // packed-dword scales: per tile, A words for the two A halves (row
-  // slice per wave) + B words for the two B halves x two n frags.
-  int sa_cur[2], sb_cur[2][2], sa_nxt[2], sb_nxt[2][2];
+  // packed-dword scales, loaded ONCE per tile at the tile top (6 dword
+  // loads; one short L2 wait per tile beats the 6 extra live registers
+  // + spill traffic the cross-tile rotation cost).
+  int sa_cur[2], sb_cur[2][2];
   auto load_scales = [&](int t, int* sa, int sb[2][2]) {
 #pragma unroll
     for (int h = 0; h < 2; ++h)
@@ -737,7 +758,7 @@ gemm_bt_mx8_acc_k128q(float* __restrict__ C,
 #define MFMA_Q(q, ah, bh, kk)                                              \
   _Pragma("unroll") for (int n = 0; n < 2; ++n)                            \
       acc[q][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(         \
-          af[kk], bfr[kk][n], acc[q][n], 0, 0, 0,                          \
+          af, bfr[n], acc[q][n], 0, 0, 0,                                  \
           (sa_cur[ah] >> (8 * ((kk) * 2 + (int)kh))) & 0xFF, 0,            \
           (sb_cur[bh][n] >> (8 * ((kk) * 2 + (int)kh))) & 0xFF);
 
@@ -749,59 +770,56 @@ gemm_bt_mx8_acc_k128q(float* __restrict__ C,
   stage_half(0u, 2, 0);
   stage_half(0u, 3, 0);
   stage_half(0u, 1, 0);
-  load_scales(0, sa_cur, sb_cur);
-  asm volatile("s_waitcnt vmcnt(10)" ::: "memory");  // A0,B0 landed
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // A0,B0 landed
   __builtin_amdgcn_s_barrier();
 
   const int NT = K / 128;
   for (int t = 0; t < NT; ++t) {
     const u32 cur = (u32)(t & 1) * QBUFSZ, nxt = cur ^ QBUFSZ;
     const bool pre = (t + 1 < NT);
-    // ---- phase 1: Q0 = A0 x B0; stage A0'; load scales(t+1)
-    read_a_kk(cur + QHOFF(0), 0);
-    read_b_kk(cur + QHOFF(2), 0);
-    read_a_kk(cur + QHOFF(0), 1);
-    read_b_kk(cur + QHOFF(2), 1);
-    if (pre) {
-      stage_half(nxt, 0, t + 1);
-      load_scales(t + 1, sa_nxt, sb_nxt);
-    }
+    load_scales(t, sa_cur, sb_cur);  // 6 dwords, short L2 wait
+    // ---- phase 1: Q0 = A0 x B0; stage A0'
+    read_a_kk(cur, 0, 0);
+    read_b_kk(cur, 2, 0);
+    if (pre) stage_half(nxt, 0, t + 1);
     MFMA_Q(0, 0, 0, 0)
+    read_a_kk(cur, 0, 1);
+    read_b_kk(cur, 2, 1);
     MFMA_Q(0, 0, 0, 1)
-    if (pre) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    // ---- phase 2: Q1 = A0 x B1 (A frags reused); stage B0'
-    read_b_kk(cur + QHOFF(3), 0);
-    read_b_kk(cur + QHOFF(3), 1);
+    // ---- phase 2: Q1 = A0 x B1; stage B0'
+    read_a_kk(cur, 0, 0);
+    read_b_kk(cur, 3, 0);
     if (pre) stage_half(nxt, 2, t + 1);
     MFMA_Q(1, 0, 1, 0)
+    read_a_kk(cur, 0, 1);
+    read_b_kk(cur, 3, 1);
     MFMA_Q(1, 0, 1, 1)
-    if (pre) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    // ---- phase 3: Q2 = A1 x B1 (B frags reused); stage B1'
-    read_a_kk(cur + QHOFF(1), 0);
-    read_a_kk(cur + QHOFF(1), 1);
+    // ---- phase 3: Q2 = A1 x B1; stage B1'
+    read_a_kk(cur, 1, 0);
+    read_b_kk(cur, 3, 0);
     if (pre) stage_half(nxt, 3, t + 1);
     MFMA_Q(2, 1, 1, 0)
+    read_a_kk(cur, 1, 1);
+    read_b_kk(cur, 3, 1);
     MFMA_Q(2, 1, 1, 1)
     // no wait/barrier: phase 4 re-reads B0(cur), proven at tile entry,
     // and stages into the OTHER buffer
-    // ---- phase 4: Q3 = A1 x B0 (re-read); stage A1'
-    read_b_kk(cur + QHOFF(2), 0);
-    read_b_kk(cur + QHOFF(2), 1);
+    // ---- phase 4: Q3 = A1 x B0; stage A1'
+    read_a_kk(cur, 1, 0);
+    read_b_kk(cur, 2, 0);
     if (pre) stage_half(nxt, 1, t + 1);
     MFMA_Q(3, 1, 0, 0)
+    read_a_kk(cur, 1, 1);
+    read_b_kk(cur, 2, 1);
     MFMA_Q(3, 1, 0, 1)
     if (pre) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      sa_cur[h] = sa_nxt[h];
-#pragma unroll
-      for (int n = 0; n < 2; ++n) sb_cur[h][n] = sb_nxt[h][n];
-    }
   }
 
   // epilogue: 32x32 C/D map per quadrant slice
@@ -837,6 +855,188 @@ gemm_bt_mx8_acc_k128q(float* __restrict__ C,
 #undef QHOFF
 #undef QBUFSZ
 
+// ---- BK=128 HALF-TILE (v5): 256x128 tile, 2 phases, 64-VGPR acc ------
+// The 256^2 quadrant port (v4) cannot fit: eight 16-register f32x16
+// accumulator tuples + 8-register v8i operands fragment the allocator
+// past 256 VGPRs and it spills ~80 scratch ops into the loop whatever
+// the fragment/address budget (measured 358-389 TF across three
+// register plans). Halving the tile's N dimension halves the
+// accumulator file to FOUR f32x16 (64 VGPRs) and the whole kernel fits
+// with room: per K-tile two 128x128 quadrant phases (rows 0-127 /
+// 128-255 x the full 128-col N), each 2 kk x 2 n mfma_scale_32x32x64
+// on a 32x64 wave slice. LDS ring: halves A0/A1/B of 16 KiB x 2
+// buffers = 96 KiB. Stages: A0' at p1; B',A1' at p2. vmcnt ladder
+// (2 glds per stage): p1-end prove A1(t) -> vmcnt(2); p2-end prove
+// A0',B' -> vmcnt(2); last tile: vmcnt(0) / none.
+
+#define HHOFF(h) ((u32)(h) * (128u * 128u))
+#define HBUFSZ (3u * 128u * 128u)
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx8_acc_k128h(float* __restrict__ C,
+                      const unsigned char* __restrict__ A,
+                      const unsigned char* __restrict__ Asc,
+                      const unsigned char* __restrict__ Bt,
+                      const unsigned char* __restrict__ Bsc, int M, int N,
+                      int K, int beta0, u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 qr = (wave & 3) * 32;   // row slice within a 128-row quad
+  const u32 qc = (wave >> 2) * 64;  // col slice within the 128-col tile
+
+  const u32 ntm = (u32)(M / 256), ntn = (u32)(N / 128);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * 256, n0 = (long)tn * 128;
+
+  const u32 srow = tid >> 3, sslot = tid & 7;
+
+  // h: 0 = A rows 0-127, 1 = A rows 128-255, 2 = B (Bt rows 0-127)
+  auto stage_half = [&](u32 bufoff, int h, int t) {
+    const long k0 = (long)t * 128;
+    const unsigned char* src = (h < 2)
+        ? A + (m0 + (h & 1) * 128) * (long)K
+        : Bt + n0 * (long)K;
+    unsigned char* dst = (unsigned char*)(smem + bufoff + HHOFF(h));
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      u32 row = srow + it * 64;
+      u32 ss = sslot ^ ((row >> 1) & 7);
+      glds16(src + row * (long)K + k0 + ss * 16,
+             dst + row * 128u + sslot * 16u);
+    }
+  };
+
+  f32x16 acc[2][2] = {};  // [quadrant][n-frag]: 64 VGPRs
+  v8i af;
+  v8i bfr[2];
+
+  const u32 arow = qr + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  u32 a_lo, a_hi, b_lo[2], b_hi[2];
+  {
+    u32 ab = arow * 128u + kh * 16u;
+    ab ^= ((arow >> 1) & 7) << 4;
+    a_lo = ab;
+    a_hi = ab ^ 32u;
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      u32 row = qc + n * 32 + (lane & 31);
+      u32 bb = row * 128u + kh * 16u;
+      bb ^= ((row >> 1) & 7) << 4;
+      b_lo[n] = bb;
+      b_hi[n] = bb ^ 32u;
+    }
+  }
+  auto read_a_kk = [&](u32 cur, int h, int kk) {
+    u32 c = HHOFF(h) + (u32)kk * 64u;
+    af = v8_from((const char*)smem + cur + a_lo + c,
+                 (const char*)smem + cur + a_hi + c);
+  };
+  auto read_b_kk = [&](u32 cur, int kk) {
+    u32 c = HHOFF(2) + (u32)kk * 64u;
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+      bfr[n] = v8_from((const char*)smem + cur + b_lo[n] + c,
+                       (const char*)smem + cur + b_hi[n] + c);
+  };
+
+  int sa_cur[2], sb_cur[2];
+  auto load_scales = [&](int t) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+      sa_cur[h] = ((const int*)Asc)[(long)t * M + m0 + h * 128 + arow];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+      sb_cur[n] = ((const int*)Bsc)[(long)t * N + n0 + qc + n * 32
+                                    + (lane & 31)];
+  };
+
+#define MFMA_H(q, kk)                                                      \
+  _Pragma("unroll") for (int n = 0; n < 2; ++n)                            \
+      acc[q][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(         \
+          af, bfr[n], acc[q][n], 0, 0, 0,                                  \
+          (sa_cur[q] >> (8 * ((kk) * 2 + (int)kh))) & 0xFF, 0,             \
+          (sb_cur[n] >> (8 * ((kk) * 2 + (int)kh))) & 0xFF);
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  stage_half(0u, 0, 0);
+  stage_half(0u, 2, 0);
+  stage_half(0u, 1, 0);
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");  // A0,B landed
+  __builtin_amdgcn_s_barrier();
+
+  const int NT = K / 128;
+  for (int t = 0; t < NT; ++t) {
+    const u32 cur = (u32)(t & 1) * HBUFSZ, nxt = cur ^ HBUFSZ;
+    const bool pre = (t + 1 < NT);
+    load_scales(t);  // 4 dwords, short L2 wait
+    // ---- phase 1: Q0 = A0 x B; stage A0'
+    read_a_kk(cur, 0, 0);
+    read_b_kk(cur, 0);
+    if (pre) stage_half(nxt, 0, t + 1);
+    MFMA_H(0, 0)
+    read_a_kk(cur, 0, 1);
+    read_b_kk(cur, 1);
+    MFMA_H(0, 1)
+    if (pre) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 2: Q1 = A1 x B; stage B', A1'
+    read_a_kk(cur, 1, 0);
+    read_b_kk(cur, 0);
+    if (pre) { stage_half(nxt, 2, t + 1); stage_half(nxt, 1, t + 1); }
+    MFMA_H(1, 0)
+    read_a_kk(cur, 1, 1);
+    read_b_kk(cur, 1);
+    MFMA_H(1, 1)
+    if (pre) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: 32x32 C/D map per quadrant
+#pragma unroll
+  for (int q = 0; q < 2; ++q) {
+    float cv[2][16];
+    if (!beta0) {
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + q * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                     4 * (lane >> 5);
+          long col = n0 + qc + n * 32 + (lane & 31);
+          cv[n][r] = C[row * (long)N + col];
+        }
+    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + q * 128 + qr + (r & 3) + 8 * (r >> 2) +
+                   4 * (lane >> 5);
+        long col = n0 + qc + n * 32 + (lane & 31);
+        C[row * (long)N + col] =
+            (beta0 ? 0.0f : cv[n][r]) + acc[q][n][r];
+      }
+  }
+}
+#undef MFMA_H
+#undef HHOFF
+#undef HBUFSZ
+
 static int g_mx8_variant = 0;  // 0 = auto (simple k128), 2 = counted 2-phase
 extern "C" void set_mx8_variant(int v) { g_mx8_variant = v; }
 
@@ -846,6 +1046,18 @@ extern "C" void launch_gemm_bt_mx8_acc(float* C, const void* A,
                                        int beta0, hipStream_t stream) {
   dim3 grid((M / MXBM) * (N / MXBN));
   if (K % 128 == 0) {
+    if (g_mx8_variant == 5 && N % 128 == 0) {
+      size_t lds = 2u * 3u * 128u * 128u;  // 96 KiB
+      hipLaunchKernelGGL(gemm_bt_mx8_acc_k128h,
+                         dim3((unsigned)((M / 256) * (N / 128))),
+                         dim3(512), lds, stream, C,
+                         (const unsigned char*)A,
+                         (const unsigned char*)Asc,
+                         (const unsigned char*)Bt,
+                         (const unsigned char*)Bsc, M, N, K, beta0,
+                         g_st_m, g_st_n);
+      return;
+    }
     if (g_mx8_variant == 4) {
       size_t lds = 2u * 4u * 128u * 128u;  // 128 KiB
       hipLaunchKernelGGL(gemm_bt_mx8_acc_k128q, grid, dim3(512), lds,

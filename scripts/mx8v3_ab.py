@@ -31,7 +31,7 @@ def screen(n, runs=3):
     bt8, btsc = blockops.to_mx8(bt)
     ref = run(0, a8, asc, bt8, btsc, n)
     ok = True
-    for v in (3, 4):
+    for v in (3, 4, 5):
         for r in range(runs):
             c = run(v, a8, asc, bt8, btsc, n)
             if not torch.equal(c, ref):
@@ -53,7 +53,7 @@ def perf(n, it=3, reps=3):
     a8, asc = blockops.to_mx8(a)
     bt8, btsc = blockops.to_mx8(bt)
     c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
-    for v in (0, 3, 4):
+    for v in (0, 3, 4, 5):
         ext.set_mx8_variant(v)
         xs = []
         for _ in range(reps):
