@@ -725,18 +725,24 @@ gemm_bt_mx8_acc_k128q(float* __restrict__ C,
     }
   }
   // h: 0=A0 1=A1 2=B0 3=B1; kk in {0,1}; cur = buffer byte offset
+  // kk selects bit 6 of the swizzled low byte-offset: the st swizzle
+  // XORs bits 4-6, so kk*64 must be XORed into the per-lane part (an
+  // ADD would carry through the swizzled bit pattern - that bug cost
+  // the first attempt its numerics). arow*128 has no bits < 7, so the
+  // flip commutes exactly like the ^32 hi-half flip.
   auto read_a_kk = [&](u32 cur, int h, int kk) {
-    const char* base = (const char*)smem + cur + a_lo;
-    const char* base_h = (const char*)smem + cur + a_hi;
-    u32 c = QHOFF(h) + (u32)kk * 64u;
-    af = v8_from(base + c, base_h + c);
+    u32 sel = (u32)kk * 64u;
+    u32 c = cur + QHOFF(h);
+    af = v8_from((const char*)smem + c + (a_lo ^ sel),
+                 (const char*)smem + c + (a_hi ^ sel));
   };
   auto read_b_kk = [&](u32 cur, int h, int kk) {
-    u32 c = QHOFF(h) + (u32)kk * 64u;
+    u32 sel = (u32)kk * 64u;
+    u32 c = cur + QHOFF(h);
 #pragma unroll
     for (int n = 0; n < 2; ++n)
-      bfr[n] = v8_from((const char*)smem + cur + b_lo[n] + c,
-                       (const char*)smem + cur + b_hi[n] + c);
+      bfr[n] = v8_from((const char*)smem + c + (b_lo[n] ^ sel),
+                       (const char*)smem + c + (b_hi[n] ^ sel));
   };
 
   // packed-dword scales, loaded ONCE per tile at the tile top (6 dword
@@ -938,17 +944,21 @@ gemm_bt_mx8_acc_k128h(float* __restrict__ C,
       b_hi[n] = bb ^ 32u;
     }
   }
+  // kk*64 is XORed into the swizzled per-lane part (the st swizzle
+  // covers bit 6; an ADD would carry through it - see v4's comment)
   auto read_a_kk = [&](u32 cur, int h, int kk) {
-    u32 c = HHOFF(h) + (u32)kk * 64u;
-    af = v8_from((const char*)smem + cur + a_lo + c,
-                 (const char*)smem + cur + a_hi + c);
+    u32 sel = (u32)kk * 64u;
+    u32 c = cur + HHOFF(h);
+    af = v8_from((const char*)smem + c + (a_lo ^ sel),
+                 (const char*)smem + c + (a_hi ^ sel));
   };
   auto read_b_kk = [&](u32 cur, int kk) {
-    u32 c = HHOFF(2) + (u32)kk * 64u;
+    u32 sel = (u32)kk * 64u;
+    u32 c = cur + HHOFF(2);
 #pragma unroll
     for (int n = 0; n < 2; ++n)
-      bfr[n] = v8_from((const char*)smem + cur + b_lo[n] + c,
-                       (const char*)smem + cur + b_hi[n] + c);
+      bfr[n] = v8_from((const char*)smem + c + (b_lo[n] ^ sel),
+                       (const char*)smem + c + (b_hi[n] ^ sel));
   };
 
   int sa_cur[2], sb_cur[2];
