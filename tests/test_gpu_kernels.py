@@ -490,3 +490,34 @@ def test_multiblock_16k_orchestration():
         assert rel < 5e-2, rel
     finally:
         reset_runtime()
+
+
+def test_prep_stream_mechanics():
+    """The owner-side panel-prep pattern used by exec/summa.py (second
+    HIP stream + record_stream + event fence before the consumer): the
+    consumer on the main stream must observe the prep stream's writes.
+    The SUMMA integration itself is multi-GPU-only; this validates the
+    stream/event mechanics it relies on."""
+    from matrel_amd.exec.summa import _get_prep_stream
+    from matrel_amd.ops import blockops
+    prep = _get_prep_stream()
+    for _ in range(5):
+        src = torch.randn(2048, 2048, device="cuda").to(torch.bfloat16)
+        panel = torch.empty(2048, 2048, dtype=torch.bfloat16,
+                            device="cuda")
+        panel.record_stream(prep)
+        # launch noise on the main stream first (the GEMMs prep overlaps)
+        noise_a = torch.randn(4096, 4096, device="cuda")
+        noise = noise_a @ noise_a
+        with torch.cuda.stream(prep):
+            panel.copy_(blockops.transpose(src))
+        ev = torch.cuda.Event()
+        ev.record(prep)
+        torch.cuda.current_stream().wait_event(ev)
+        # consumer on main stream
+        got = panel.float().sum()
+        torch.cuda.synchronize()
+        ref = src.float().t().sum()
+        assert torch.allclose(got, ref, rtol=1e-3), (got.item(),
+                                                     ref.item())
+        del noise
