@@ -75,9 +75,56 @@ def _body_agg_2x2(s, tmpdir):
     assert torch.allclose(rs, t.sum(dim=1), atol=1e-2)
 
 
+def _body_both_sparse_2x2(s, tmpdir):
+    """BOTH operands element-sparse on the 2x2 grid: unlike world 2
+    (2x1 grid, where the A-panel row group has a single member), BOTH
+    CSR wire directions here ride real multi-rank broadcasts - the
+    variable-size idx/val buffer sizing must agree group-uniformly
+    from the nnz maps on every rank."""
+    from matrel_amd.blocks.matrix import BlockMatrix
+    from matrel_amd.blocks.schemes import GRID
+    from matrel_amd.exec import metrics
+    from matrel_amd.exec.summa import summa_matmul
+    metrics.reset()
+    for seed in (61, 62):
+        A = BlockMatrix.sparse_random(160, 128, 0.03, seed=seed)
+        B = BlockMatrix.sparse_random(128, 96, 0.03, seed=seed + 10)
+        out = BlockMatrix(160, 96, A.block_size, GRID, A.dtype, A.device)
+        summa_matmul(A, B, out)
+        ref = A.to_dense() @ B.to_dense()
+        assert torch.allclose(out.to_dense(), ref, atol=1e-3), seed
+    rep = metrics.report()
+    assert "summa_a_bcast_csr" in rep["collectives"]
+    assert "summa_b_bcast_csr" in rep["collectives"]
+    # mixed sparse/dense rows through the same wires
+    A2 = BlockMatrix.sparse_random(128, 128, 0.04, seed=63)
+    if A2.is_local(0, 0):
+        A2.set_block(0, 0, torch.randn(A2.block_shape(0, 0)))
+    B2 = BlockMatrix.randn(128, 64, seed=64)
+    out2 = BlockMatrix(128, 64, A2.block_size, GRID, A2.dtype, A2.device)
+    summa_matmul(A2, B2, out2)
+    assert torch.allclose(out2.to_dense(),
+                          A2.to_dense() @ B2.to_dense(), atol=1e-3)
+
+
+def _body_fused_transpose_2x2(s, tmpdir):
+    """(A op B).T fused path with real 2D routing (both sub-group
+    directions non-trivial)."""
+    A = s.randn(128, 96, seed=71)
+    B = s.randn(128, 96, seed=72)
+    out = (A * B).T.compute()
+    assert torch.allclose(out.to_tensor(),
+                          (A.to_tensor() * B.to_tensor()).t(), atol=1e-5)
+    S = s.block_sparse_random(128, 96, 0.4, seed=73)
+    out2 = (S + A).T.compute()
+    assert torch.allclose(out2.to_tensor(),
+                          (S.to_tensor() + A.to_tensor()).t(), atol=1e-5)
+
+
 @pytest.mark.parametrize("body", [
     "_body_summa_2x2", "_body_chain_2x2", "_body_sparse_2x2",
-    "_body_agg_2x2", "_body_ols_2x2",
+    "_body_agg_2x2", "_body_ols_2x2", "_body_both_sparse_2x2",
+    "_body_fused_transpose_2x2",
 ])
 def test_distributed_world4(body, tmp_path):
     _run(body, tmp_path)
