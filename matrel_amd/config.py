@@ -75,6 +75,9 @@ class MatrelConfig:
     # (serialized debug mode). Panel SIZE is block_size; grid shape is
     # MATREL_GRID - together these are the driver-sweepable SUMMA knobs.
     summa_prefetch: int = 1
+    # Owner-side SUMMA panel prep on a second HIP stream (overlaps the
+    # previous step's GEMMs). Kill switch for triage: MATREL_PREP_STREAM=0.
+    prep_stream: bool = True
     # Memory cap for replicating an operand in broadcast-MM. This is ONLY
     # a memory guard (replication must fit next to the shards in 288 GB
     # HBM3E); the bcast-vs-SUMMA choice itself is by estimated xGMI bytes
@@ -114,6 +117,8 @@ class MatrelConfig:
                                     int(cfg.overlap_comm)) != 0
         cfg.summa_prefetch = _env_int("MATREL_SUMMA_PREFETCH",
                                       cfg.summa_prefetch)
+        cfg.prep_stream = _env_int("MATREL_PREP_STREAM",
+                                   int(cfg.prep_stream)) != 0
         cfg.max_bcast_bytes = _env_int("MATREL_MAX_BCAST_BYTES",
                                        cfg.max_bcast_bytes)
         cfg.deterministic = _env_int("MATREL_DETERMINISTIC",

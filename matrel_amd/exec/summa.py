@@ -403,6 +403,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     # runs on a second HIP stream so it overlaps the already-enqueued
     # GEMMs of the previous step; broadcasts wait on it via events.
     use_prep = (a.device.type == "cuda" and get_config().overlap_comm
+                and get_config().prep_stream
                 and not get_config().deterministic)
     prep = _get_prep_stream() if use_prep else None
     if prep is not None:
