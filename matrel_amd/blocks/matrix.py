@@ -405,7 +405,38 @@ class BlockMatrix:
         if not rt.is_distributed:
             out.blocks.update(self.blocks)
             return out
+        from ..exec.metrics import add_comm
         pm = self.presence()
+        # CSR wire (round 2): element-sparse blocks move as their CSR
+        # arrays - nnz-proportional xGMI bytes, and receivers KEEP the
+        # sparse storage (previously every move densified). The decision
+        # is group-uniform: derived from the globally-synced nnz map with
+        # the same predicate the SUMMA wire uses (only CSR-stored blocks
+        # can satisfy it - dense-stored blocks report h*w).
+        nm = self.nnz_map()
+        thr = get_config().sparse_threshold
+
+        def wire_sparse(bi: int, bj: int) -> bool:
+            h, w = self.block_shape(bi, bj)
+            return int(nm[bi, bj]) < thr * h * w
+
+        def csr_bufs(bi: int, bj: int):
+            """(crow, cols, vals) buffers; source fills, others allocate
+            (sizes agreed from the nnz map + geometry)."""
+            h, w = self.block_shape(bi, bj)
+            nz = int(nm[bi, bj])
+            if rt.rank == self.owner(bi, bj):
+                t = self.blocks[(bi, bj)]
+                if t.layout != torch.sparse_csr:
+                    t = t.to_sparse_csr()
+                return (t.crow_indices().to(torch.int32).contiguous(),
+                        t.col_indices().to(torch.int32).contiguous(),
+                        t.values().contiguous())
+            return (torch.empty(h + 1, dtype=torch.int32,
+                                device=self.device),
+                    torch.empty(nz, dtype=torch.int32, device=self.device),
+                    torch.empty(nz, dtype=self.dtype, device=self.device))
+
         for (bi, bj) in self.block_indices():
             src = self.owner(bi, bj)
             dst = scheme.owner(bi, bj, rt.world_size, rt.grid)
@@ -418,24 +449,47 @@ class BlockMatrix:
                 continue
             if not pm[bi, bj]:
                 continue
+            sp = wire_sparse(bi, bj)
             if dst == -1:
+                if sp:
+                    crow, cols, vals = csr_bufs(bi, bj)
+                    for b in (crow, cols, vals):
+                        dist.broadcast(b, src=src)
+                    add_comm("repartition_bcast_csr",
+                             crow.numel() * 4 + cols.numel() * 4
+                             + vals.numel() * vals.element_size())
+                    out.blocks[(bi, bj)] = torch.sparse_csr_tensor(
+                        crow, cols, vals, size=self.block_shape(bi, bj))
+                    continue
                 t = (_densify(self.blocks[(bi, bj)]).contiguous()
                      if rt.rank == src else
                      torch.empty(self.block_shape(bi, bj), dtype=self.dtype,
                                  device=self.device))
                 dist.broadcast(t, src=src)
-                from ..exec.metrics import add_comm
                 add_comm("repartition_bcast",
                          t.numel() * t.element_size())
                 out.blocks[(bi, bj)] = t
             elif src == dst:
                 if rt.rank == src:
                     out.blocks[(bi, bj)] = self.blocks[(bi, bj)]
+            elif sp:
+                if rt.rank == src:
+                    crow, cols, vals = csr_bufs(bi, bj)
+                    for b in (crow, cols, vals):
+                        dist.send(b, dst=dst)
+                    add_comm("repartition_send_csr",
+                             crow.numel() * 4 + cols.numel() * 4
+                             + vals.numel() * vals.element_size())
+                elif rt.rank == dst:
+                    crow, cols, vals = csr_bufs(bi, bj)
+                    for b in (crow, cols, vals):
+                        dist.recv(b, src=src)
+                    out.blocks[(bi, bj)] = torch.sparse_csr_tensor(
+                        crow, cols, vals, size=self.block_shape(bi, bj))
             else:
                 if rt.rank == src:
                     tt = _densify(self.blocks[(bi, bj)]).contiguous()
                     dist.send(tt, dst=dst)
-                    from ..exec.metrics import add_comm
                     add_comm("repartition_send",
                              tt.numel() * tt.element_size())
                 elif rt.rank == dst:

@@ -111,6 +111,34 @@ def _body_repartition(s, tmpdir):
         assert torch.allclose(moved.to_dense(), bm.to_dense(), atol=0)
 
 
+def _body_repartition_csr_wire(s, tmpdir):
+    """Element-sparse repartition moves CSR arrays (nnz-proportional
+    bytes) and the receivers KEEP sparse storage (round 2; previously
+    every scheme move densified the wire AND the received block)."""
+    from matrel_amd.blocks.schemes import BCAST, COL, ROW
+    from matrel_amd.exec import metrics
+    from matrel_amd.ops import blockops
+    metrics.reset()
+    A = s.sparse_random(128, 96, 0.03, seed=91)
+    bm = A.blocks
+    for sch in (ROW, COL, BCAST):
+        moved = bm.repartition(sch)
+        assert torch.allclose(moved.to_dense(), bm.to_dense(), atol=0)
+        # received blocks stay CSR
+        for _, t in moved.local_items():
+            assert blockops.is_sparse(t), sch
+    rep = metrics.report()
+    keys = rep["collectives"]
+    assert any(k.startswith("repartition") and k.endswith("_csr")
+               for k in keys), keys
+    assert "repartition_bcast" not in keys and \
+        "repartition_send" not in keys, keys
+    # dense matrices still use the dense wire
+    B = s.randn(96, 64, seed=92)
+    moved = B.blocks.repartition(ROW)
+    assert torch.allclose(moved.to_dense(), B.blocks.to_dense(), atol=0)
+
+
 def _body_block_sparse(s, tmpdir):
     A = s.block_sparse_random(128, 128, 0.4, seed=8)
     B = s.randn(128, 64, seed=9)
@@ -378,7 +406,8 @@ def _body_ols(s, tmpdir):
 @pytest.mark.parametrize("body", [
     "_body_summa", "_body_bcast_mm", "_body_aggregates", "_body_transpose",
     "_body_ewise_transpose_fused",
-    "_body_repartition", "_body_block_sparse", "_body_sparse_skewed",
+    "_body_repartition", "_body_repartition_csr_wire",
+    "_body_block_sparse", "_body_sparse_skewed",
     "_body_elem_sparse_summa", "_body_both_sparse_summa",
     "_body_relational",
     "_body_deterministic_serialized", "_body_mx8_config",
