@@ -47,6 +47,22 @@ def test_bench_main_world2(tmp_path):
     assert d["config"]["parallelism"].startswith("summa_grid")
 
 
+def test_bench_main_world8(tmp_path):
+    """The SCALE run's exact rank count (8, 4x2 grid) through bench
+    main() on gloo - rendezvous, grid derivation, MAX-over-ranks
+    reduction and the single rank-0 JSON line."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.spawn(_worker, args=(8, 29954, q), nprocs=8, join=True)
+    outs = dict(q.get() for _ in range(8))
+    lines = [ln for r in outs.values() for ln in r.strip().splitlines()
+             if ln.startswith("{")]
+    assert len(lines) == 1, outs
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "summa_grid4x2"
+
+
 def test_bench_single_process(tmp_path):
     import subprocess
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
