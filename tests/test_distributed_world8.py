@@ -71,8 +71,47 @@ def _worker_grid24(rank: int, world: int, port: int) -> None:
     dist.destroy_process_group()
 
 
+def _worker_stress(rank: int, world: int, port: int) -> None:
+    """SCALE-run rehearsal: maximum collective-ordering pressure on the
+    exact 4x2 grid - many k-steps, mixed block-/element-sparse operands
+    (skipped ka/kb steps + variable-size CSR wires in BOTH directions),
+    prefetch depth 3. A deadlock here is a deadlock in the driver's
+    8-GPU run; a pass pins the enqueue-order reasoning of
+    docs/RCCL_CHECKLIST.md item 2."""
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      HSA_ENABLE_IPC_MODE_LEGACY="0")
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import matrel_amd as ma
+    cfg = ma.MatrelConfig(block_size=16)
+    cfg.summa_prefetch = 3
+    ma.set_config(cfg)
+    s = ma.MatrelSession()
+    assert s.runtime.grid == (4, 2)
+    # 20 k-steps, block-sparse A x element-sparse B
+    A = s.block_sparse_random(256, 320, 0.25, seed=101)
+    B = s.sparse_random(320, 128, 0.04, seed=102)
+    C = (A @ B).compute()
+    ref = A.to_tensor() @ B.to_tensor()
+    assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
+    # element-sparse x block-sparse the other way, chained with an ewise
+    D = s.sparse_random(128, 256, 0.03, seed=103)
+    E = s.block_sparse_random(320, 96, 0.3, seed=104)
+    F = ((D @ A @ E) * 2.0).compute()
+    reff = (D.to_tensor() @ A.to_tensor() @ E.to_tensor()) * 2.0
+    assert torch.allclose(F.to_tensor(), reff, atol=1e-2)
+    dist.destroy_process_group()
+
+
 def test_distributed_world8():
     mp.spawn(_worker, args=(8, 29931), nprocs=8, join=True)
+
+
+def test_distributed_world8_scale_rehearsal():
+    mp.spawn(_worker_stress, args=(8, 29951), nprocs=8, join=True)
 
 
 def test_distributed_world8_grid_2x4():
