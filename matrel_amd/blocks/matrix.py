@@ -21,6 +21,7 @@ Supports ragged edge blocks (m, n need not divide block_size).
 from __future__ import annotations
 
 import math
+import weakref
 from typing import Dict, Iterator, Optional, Tuple
 
 import torch
@@ -45,16 +46,23 @@ class _BlockStore(dict):
     existed only `set_block` invalidated, so a stale nnz map could make the
     SUMMA CSR-wire encoder size broadcast buffers from outdated values
     (ADVICE.md round 1). Now every mutation path is safe.
+
+    The owner backref is a WEAKREF: matrix -> .blocks -> owner as a
+    strong ref would be a reference cycle, and multi-hundred-GB result
+    matrices would then die only at gc intervals instead of by refcount
+    (caught as an OOM regression on the near-capacity 163840^2 bench).
     """
 
     __slots__ = ("_owner",)
 
     def __init__(self, owner: "BlockMatrix", *args) -> None:
         super().__init__(*args)
-        self._owner = owner
+        self._owner = weakref.ref(owner)
 
     def _inv(self) -> None:
-        self._owner._invalidate_stats()
+        o = self._owner()
+        if o is not None:
+            o._invalidate_stats()
 
     def __setitem__(self, k, v) -> None:
         super().__setitem__(k, v)

@@ -97,3 +97,23 @@ def test_maxmin_single_process_all_negative():
     A = s.from_dense(a)
     assert abs(A.max().item() - a.max().item()) < 1e-5
     assert abs(A.min().item() - a.min().item()) < 1e-5
+
+
+def test_blockmatrix_dies_by_refcount():
+    """BlockMatrix must be freed by REFCOUNT, not the cycle collector:
+    a strong blocks->owner backref once made multi-hundred-GB result
+    matrices linger until gc intervals (OOM at the near-capacity
+    163840^2 bench)."""
+    import gc
+    import weakref
+    from matrel_amd.blocks.matrix import BlockMatrix
+    gc.disable()
+    try:
+        m = BlockMatrix.randn(64, 64, seed=9)
+        m.presence()  # populate caches
+        m.blocks[(0, 0)] = torch.zeros(32, 32)  # exercise the store
+        r = weakref.ref(m)
+        del m
+        assert r() is None, "BlockMatrix survived del (reference cycle)"
+    finally:
+        gc.enable()
