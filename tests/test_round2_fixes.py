@@ -117,3 +117,29 @@ def test_blockmatrix_dies_by_refcount():
         assert r() is None, "BlockMatrix survived del (reference cycle)"
     finally:
         gc.enable()
+
+
+def test_no_result_accumulation_across_steps():
+    """Iterative workloads (the bench, power iteration) must not retain
+    discarded step outputs: every previous result's BlockMatrix dies by
+    refcount before the next step completes."""
+    import gc
+    import weakref
+    s = ma.MatrelSession()
+    A = s.randn(128, 128, seed=21)
+    B = s.randn(128, 128, seed=22)
+    gc.disable()
+    try:
+        dead = []
+        prev = None
+        for _ in range(5):
+            C = (A @ B).compute()
+            if prev is not None:
+                dead.append(prev)
+            prev = weakref.ref(C.plan.data)
+        del C
+        dead.append(prev)
+        alive = [r for r in dead if r() is not None]
+        assert not alive, f"{len(alive)} step outputs retained"
+    finally:
+        gc.enable()
