@@ -83,7 +83,14 @@ def spgemm_acc(acc: torch.Tensor, a_csr: torch.Tensor,
     """acc(fp32, MxN) += a_csr(M x K) @ b_csr(K x N) - sparse x sparse
     block product (HIP ds_add_f32 LDS-row kernel on GPU; SURVEY.md §2.1
     "sparse x sparse block products"). Returns False when the kernel
-    declined (N > 16 KiB LDS row) - caller densifies b instead."""
+    declined (N > 16 KiB LDS row) - caller densifies b instead.
+
+    The kernel accumulates via LDS float ATOMICS (lane-order varies), so
+    in deterministic mode (utils/debug.py: fixed-order reductions only)
+    it declines and the caller takes the deterministic densified path."""
+    from ..config import get_config
+    if acc.is_cuda and get_config().deterministic:
+        return False
     if acc.is_cuda:
         ext = require_ext()
         va = a_csr.values()

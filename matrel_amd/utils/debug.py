@@ -15,7 +15,10 @@ design:
     atomics), SUMMA accumulates k-panels in ascending k order on a single
     stream, and the bf16 GEMM accumulates in fp32 with a fixed K-tile
     order. The flag additionally disables comm/compute overlap so
-    collective completion order cannot vary.
+    collective completion order cannot vary, and the sparse x sparse
+    SpGEMM kernel (whose LDS-atomic accumulation is lane-order
+    dependent) declines in favor of the fixed-order densified path
+    (ops/blockops.spgemm_acc).
   * asan: build the extension with MATREL_ASAN=1 to add
     -fsanitize=address on the HOST side of the extension
     (python setup.py build_ext --inplace).

@@ -143,3 +143,17 @@ def test_no_result_accumulation_across_steps():
         assert not alive, f"{len(alive)} step outputs retained"
     finally:
         gc.enable()
+
+
+def test_spgemm_declines_in_deterministic_mode():
+    """deterministic=True must route sparse x sparse away from the
+    LDS-atomic SpGEMM kernel (lane-order-dependent sums) while staying
+    correct through the densified path."""
+    cfg = ma.MatrelConfig(block_size=32)
+    cfg.deterministic = True
+    s = ma.MatrelSession(config=cfg)
+    A = s.sparse_random(96, 96, 0.05, seed=31)
+    B = s.sparse_random(96, 64, 0.05, seed=32)
+    C = (A @ B).compute()
+    ref = A.to_tensor() @ B.to_tensor()
+    assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
