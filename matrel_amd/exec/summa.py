@@ -470,7 +470,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         bytes (config 4). A rank joins the A (B) broadcast only at steps
         in ka (kb) - group-uniform by construction.
         """
-        a_rows, a_stack, wa = [], None, None
+        a_rows, a_stack = [], None
         a_sc = None
         a_spinfo: Dict[int, tuple] = {}
         a_dn_index: Dict[int, int] = {}

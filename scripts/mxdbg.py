@@ -45,7 +45,15 @@ def dequant(a8, sc):
     val = torch.where(be > 0, (2.0 ** (be - 7)) * (1 + m / 8.0),
                       (2.0 ** -6) * (m / 8.0))
     val = torch.where(s > 0, -val, val)
-    scale = 2.0 ** (sc.to(torch.int32).float().t() - 127)
+    if sc.dtype == torch.int32:
+        # packed-dword layout [K/128][rows] (mxconvert round 2): byte b
+        # of word t covers 32-block 4t+b
+        kt, rows = sc.shape
+        scb = sc.cpu().view(torch.uint8).reshape(kt, rows, 4) \
+            .permute(0, 2, 1).reshape(kt * 4, rows).to(sc.device)
+    else:
+        scb = sc
+    scale = 2.0 ** (scb.to(torch.int32).float().t() - 127)
     return val.float() * scale.repeat_interleave(32, dim=1)
 
 
