@@ -1,0 +1,198 @@
+// Standalone MX-fp4 MFMA layout probe (not part of the extension).
+// Verifies the operand/scale lane layout of
+// v_mfma_scale_f32_32x32x64_f8f6f4 with cbsz=blgp=4 (fp4 e2m1).
+//
+// Hypothesis under test (the fp8 map of scripts/mx_probe.hip with bytes
+// -> nibbles): lane l carries 32 e2m1 NIBBLES (16 payload bytes, low
+// half of the v8i operand) of row/col l&31, with
+//   hw_k(l, nib) = (l>>5)*16 + (nib & 15) + (nib >> 4)*32
+// and ONE e8m0 scale byte covering hw block (l>>5). The engine layout
+// then places original k [kh*16,+16) in nibbles 0-15 and [32+kh*16,+16)
+// in nibbles 16-31 so scale blocks stay contiguous in original k.
+//
+// Build: hipcc --offload-arch=gfx950 -O2 scripts/mx4_probe.hip -o scripts/mx4_probe
+#include <hip/hip_runtime.h>
+#include <cmath>
+#include <cstdio>
+#include <cstring>
+
+typedef int v8i __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+__global__ void probe4_k(const unsigned char* __restrict__ A,
+                         const unsigned char* __restrict__ B,
+                         float* __restrict__ C, int sa, int sb, int hi) {
+  int lane = threadIdx.x;
+  v8i a = {}, b = {};
+  // 16 payload bytes per lane; hi=0 -> regs 0..3, hi=1 -> regs 4..7
+  const int* ap = (const int*)(A + lane * 16);
+  const int* bp = (const int*)(B + lane * 16);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    a[i + (hi ? 4 : 0)] = ap[i];
+    b[i + (hi ? 4 : 0)] = bp[i];
+  }
+  f32x16 c = {};
+  c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 4, 4, 0,
+                                                      sa, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    int col = lane & 31;
+    C[row * 32 + col] = c[r];
+  }
+}
+
+// e2m1: [s e e m]; e>0: (1+m/2)*2^(e-1); e==0: m*0.25? (subnormal 0.5*m)
+static unsigned char f2e2m1(float f) {
+  if (f == 0.0f) return 0;
+  unsigned s = f < 0 ? 0x8 : 0;
+  float a = fabsf(f);
+  // representable: 0.5,1,1.5,2,3,4,6
+  const float vals[7] = {0.5f, 1.0f, 1.5f, 2.0f, 3.0f, 4.0f, 6.0f};
+  const unsigned codes[7] = {0x1, 0x2, 0x3, 0x4, 0x5, 0x6, 0x7};
+  int best = 0;
+  float bd = 1e30f;
+  for (int i = 0; i < 7; ++i) {
+    float d = fabsf(a - vals[i]);
+    if (d < bd) { bd = d; best = i; }
+  }
+  return (unsigned char)(s | codes[best]);
+}
+
+static float e2m1val(unsigned char n) {
+  float v[8] = {0.0f, 0.5f, 1.0f, 1.5f, 2.0f, 3.0f, 4.0f, 6.0f};
+  float x = v[n & 7];
+  return (n & 8) ? -x : x;
+}
+
+unsigned char hA[64 * 16], hB[64 * 16];
+float hC[32 * 32];
+unsigned char *dA, *dB;
+float* dC;
+
+static void run(int sa, int sb, int hi) {
+  (void)hipMemcpy(dA, hA, sizeof(hA), hipMemcpyHostToDevice);
+  (void)hipMemcpy(dB, hB, sizeof(hB), hipMemcpyHostToDevice);
+  (void)hipMemset(dC, 0, sizeof(hC));
+  hipLaunchKernelGGL(probe4_k, dim3(1), dim3(64), 0, 0, dA, dB, dC, sa,
+                     sb, hi);
+  (void)hipDeviceSynchronize();
+  (void)hipMemcpy(hC, dC, sizeof(hC), hipMemcpyDeviceToHost);
+}
+
+static void set_nib(unsigned char* buf, int lane, int nib,
+                    unsigned char v) {
+  unsigned char& byte = buf[lane * 16 + (nib >> 1)];
+  if (nib & 1)
+    byte = (unsigned char)((byte & 0x0F) | (v << 4));
+  else
+    byte = (unsigned char)((byte & 0xF0) | v);
+}
+
+int main() {
+  (void)hipMalloc(&dA, sizeof(hA));
+  (void)hipMalloc(&dB, sizeof(hB));
+  (void)hipMalloc(&dC, sizeof(hC));
+  unsigned char one = f2e2m1(1.0f);
+
+  for (int hi = 0; hi < 2; ++hi) {
+    memset(hA, (one << 4) | one, sizeof(hA));
+    memset(hB, (one << 4) | one, sizeof(hB));
+    run(127, 127, hi);
+    printf("[hi=%d] ones: C[0][0]=%g (want 64)\n", hi, hC[0]);
+  }
+  // pick the half that worked for the rest
+  memset(hA, (one << 4) | one, sizeof(hA));
+  memset(hB, (one << 4) | one, sizeof(hB));
+  run(127, 127, 0);
+  int HI = (hC[0] == 64.0f) ? 0 : 1;
+  printf("using hi=%d\n", HI);
+
+  // scale sanity: sa=128 doubles everything?
+  run(128, 127, HI);
+  printf("sa=128: C[0][0]=%g (128 if scale doubles)\n", hC[0]);
+
+  // single A nibble at (lane0, nib0): which row?
+  memset(hA, 0, sizeof(hA));
+  set_nib(hA, 0, 0, one);
+  run(127, 127, HI);
+  for (int i = 0; i < 32 * 32; ++i)
+    if (hC[i] != 0.0f)
+      printf("A(l0,n0): C[%d][%d]=%g\n", i / 32, i % 32, hC[i]);
+
+  // k-map: A one at (l0, nib n); B one at (l0, nib m): hit iff same hw_k
+  int hits = 0, miss = 0;
+  for (int n : {0, 5, 15, 16, 20, 31}) {
+    for (int m : {0, 5, 15, 16, 20, 31}) {
+      memset(hA, 0, sizeof(hA));
+      memset(hB, 0, sizeof(hB));
+      set_nib(hA, 0, n, one);
+      set_nib(hB, 0, m, one);
+      run(127, 127, HI);
+      bool hit = hC[0] != 0.0f;
+      if ((n == m) != hit)
+        printf("UNEXPECTED n=%d m=%d hit=%d\n", n, m, (int)hit);
+      hit ? ++hits : ++miss;
+    }
+  }
+  printf("diag-kmap: hits=%d miss=%d (want 6 / 30)\n", hits, miss);
+
+  // cross-kh: A(l0, n16) vs B(l32, m0): per hypothesis
+  // hw_k(0,16)=0+0+32=32; hw_k(32,0)=16+0+0=16 -> MISS expected;
+  // A(l0,n16)=32 vs B(l32,n16)=16+0+32=48 -> MISS; A(l32,n0)=16 vs
+  // B(l0,n16)=32 MISS; A(l32,n16)=48 vs B(l32,n16)=48 HIT.
+  struct { int la, na, lb, nb, want; } cases[] = {
+      {0, 16, 32, 0, 0}, {0, 16, 32, 16, 0}, {32, 0, 0, 16, 0},
+      {32, 16, 32, 16, 1}, {0, 0, 32, 0, 0}, {32, 5, 32, 5, 1}};
+  for (auto& cse : cases) {
+    memset(hA, 0, sizeof(hA));
+    memset(hB, 0, sizeof(hB));
+    set_nib(hA, cse.la, cse.na, one);
+    set_nib(hB, cse.lb, cse.nb, one);
+    run(127, 127, HI);
+    // A row = la&31; B col = lb&31
+    float got = hC[(cse.la & 31) * 32 + (cse.lb & 31)];
+    printf("cross A(l%d,n%d) B(l%d,n%d): got=%g want %s\n", cse.la,
+           cse.na, cse.lb, cse.nb, got, cse.want ? "hit" : "0");
+  }
+
+  // end-to-end: random e2m1 payloads + random scales vs host reference
+  srand(7);
+  int sa_arr[64], sb_arr[64];
+  for (int l = 0; l < 64; ++l) {
+    for (int nib = 0; nib < 32; ++nib) {
+      set_nib(hA, l, nib, (unsigned char)(rand() & 0xF));
+      set_nib(hB, l, nib, (unsigned char)(rand() & 0xF));
+    }
+  }
+  run(127, 127, HI);
+  // host reference per hypothesized map
+  double maxerr = 0;
+  for (int i = 0; i < 32; ++i)
+    for (int j = 0; j < 32; ++j) {
+      double acc = 0;
+      for (int k = 0; k < 64; ++k) {
+        // find (lane, nib) for row i, hw_k k on A: lane = i + 32*(kh?);
+        int kh_sl = (k & 32) ? 1 : 0;           // hw slice high bit
+        int klo = k & 31;
+        // hw_k = kh*16 + (nib&15) + (nib>>4)*32
+        // invert: nib>>4 = kh_sl; (nib&15) = klo - kh*16 -> need kh:
+        // klo in [kh*16, kh*16+16)
+        int kh = (klo >= 16) ? 1 : 0;
+        int nib = (klo - kh * 16) + kh_sl * 16;
+        int lane_a = i + kh * 32;
+        int lane_b = j + kh * 32;
+        unsigned char na =
+            (hA[lane_a * 16 + (nib >> 1)] >> ((nib & 1) ? 4 : 0)) & 0xF;
+        unsigned char nb =
+            (hB[lane_b * 16 + (nib >> 1)] >> ((nib & 1) ? 4 : 0)) & 0xF;
+        acc += (double)e2m1val(na) * e2m1val(nb);
+      }
+      double d = fabs(acc - hC[i * 32 + j]);
+      if (d > maxerr) maxerr = d;
+    }
+  printf("end-to-end random vs host ref: maxerr=%g (0 expected)\n",
+         maxerr);
+  return 0;
+}
