@@ -51,9 +51,10 @@ class MatrelConfig:
     # Solve/Inverse gather their operand DENSE to every rank (the OLS k x k
     # Gram matrix). Refuse silently-catastrophic gathers above this side.
     solve_max_side: int = 32768
-    # GEMM compute precision: "bf16" (default) or "mx8" (MX-scaled fp8
-    # e4m3 with per-32-element e8m0 block scales - ~2x the bf16 MFMA rate
-    # at ~2-3 decimal digits; opt-in).
+    # GEMM compute precision: "bf16" (default), "mx8" (MX-scaled fp8
+    # e4m3, ~2x the bf16 MFMA rate at ~2-3 decimal digits) or "mx4"
+    # (MX-scaled fp4 e2m1, ~2x the fp8 rate at ~1-2 significant digits -
+    # screening-pass precision). Both opt-in.
     gemm_precision: str = "bf16"
 
     # --- sparsity ---

@@ -17,6 +17,9 @@ int launch_gemm_bt_bf16_final(float*, void*, const void*, const void*, int,
 void launch_gemm_bt_mx8_acc(float*, const void*, const void*, const void*,
                             const void*, int, int, int, int, hipStream_t);
 void launch_mx8_convert(void*, void*, const void*, long, long, hipStream_t);
+void launch_mx4_convert(void*, void*, const void*, long, long, hipStream_t);
+void launch_gemm_bt_mx4_acc(float*, const void*, const void*, const void*,
+                            const void*, int, int, int, int, hipStream_t);
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
                             int, hipStream_t);
 void set_gemm_variant(int);
@@ -376,6 +379,47 @@ std::vector<torch::Tensor> mx8_convert(torch::Tensor a) {
   return {out8, sc};
 }
 
+// bf16 -> MX-fp4: packed e2m1 nibbles [rows][cols/2] + packed-dword
+// e8m0 scales [cols/128][rows] (same scale layout as MX-fp8 round 2)
+std::vector<torch::Tensor> mx4_convert(torch::Tensor a) {
+  check_2d(a, "a");
+  TORCH_CHECK(is_bf16(a), "mx4_convert needs bf16");
+  long rows = a.size(0), cols = a.size(1);
+  TORCH_CHECK(cols % 128 == 0, "cols must be a multiple of 128");
+  auto opts = torch::TensorOptions().dtype(torch::kUInt8).device(a.device());
+  auto out4 = torch::empty({rows, cols / 2}, opts);
+  auto sc = torch::empty({cols / 128, rows},
+                         torch::TensorOptions().dtype(torch::kInt32)
+                             .device(a.device()));
+  launch_mx4_convert(out4.data_ptr(), sc.data_ptr(), a.data_ptr(), rows,
+                     cols, cur_stream());
+  return {out4, sc};
+}
+
+// C(f32) += A4 @ Bt4^T with per-32-block e8m0 scales (MX-fp4, ~2x the
+// MX-fp8 rate at ~1-2 significant digits; opt-in)
+void gemm_bt_mx4_acc(torch::Tensor C, torch::Tensor A4, torch::Tensor Asc,
+                     torch::Tensor Bt4, torch::Tensor Bsc, bool beta0) {
+  check_2d(C, "C");
+  check_2d(A4, "A4");
+  check_2d(Bt4, "Bt4");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(A4.scalar_type() == torch::kUInt8 &&
+              Bt4.scalar_type() == torch::kUInt8);
+  int M = (int)A4.size(0), K = (int)(A4.size(1) * 2), N = (int)Bt4.size(0);
+  TORCH_CHECK(Bt4.size(1) * 2 == K && C.size(0) == M && C.size(1) == N);
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 128 == 0,
+              "mx4 GEMM needs M,N % 256 == 0, K % 128 == 0");
+  TORCH_CHECK(Asc.scalar_type() == torch::kInt32 &&
+              Bsc.scalar_type() == torch::kInt32 &&
+              Asc.size(0) == K / 128 && Asc.size(1) == M &&
+              Bsc.size(0) == K / 128 && Bsc.size(1) == N,
+              "packed scales must be [K/128][rows] int32");
+  launch_gemm_bt_mx4_acc(C.data_ptr<float>(), A4.data_ptr(),
+                         Asc.data_ptr(), Bt4.data_ptr(), Bsc.data_ptr(),
+                         M, N, K, beta0 ? 1 : 0, cur_stream());
+}
+
 // C(f32) += A8 @ Bt8^T with per-32-block e8m0 scales
 void gemm_bt_mx8_acc(torch::Tensor C, torch::Tensor A8, torch::Tensor Asc,
                      torch::Tensor Bt8, torch::Tensor Bsc, bool beta0) {
@@ -546,6 +590,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("spgemm_csr_csr_acc", &spgemm_csr_csr_acc,
         "C(f32) += A_csr @ B_csr (sparse x sparse block product)");
   m.def("mx8_convert", &mx8_convert, "bf16 -> (e4m3, e8m0 block scales)");
+  m.def("mx4_convert", &mx4_convert,
+        "bf16 -> (e2m1 nibbles, e8m0 block scales)");
+  m.def("gemm_bt_mx4_acc", &gemm_bt_mx4_acc,
+        "C(f32) += A4 @ Bt4^T (MX-fp4 scaled MFMA, ~2x the fp8 rate)");
   m.def("gemm_bt_mx8_acc", &gemm_bt_mx8_acc,
         "C(f32) += A8 @ Bt8^T (MX-fp8 scaled MFMA, ~2x bf16 rate)");
 }

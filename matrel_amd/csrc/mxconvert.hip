@@ -100,3 +100,91 @@ extern "C" void launch_mx8_convert(void* out8, void* sc, const void* in,
                      (unsigned char*)out8, (unsigned char*)sc,
                      (const bf16*)in, rows, cols);
 }
+
+// ---- bf16 -> MX-fp4 (e2m1 + e8m0 per-32-block scales) -------------------
+// Layout verified by scripts/mx4_probe.hip: the fp8 byte map at NIBBLE
+// granularity (payload = 16 bytes/lane in the LOW v4i half of the
+// mfma_scale operand). Memory layout: nibble k of a row-block packed 2
+// per byte in original-k order ([rows][cols/2] bytes); scales use the
+// SAME packed-dword layout as fp8 ([cols/128][rows] int32), so the
+// SUMMA wire/caches treat both precisions uniformly. Scale exponent:
+// e = floor(log2 max) - 1 normalizes the block max into [2,4) - inside
+// e2m1's 6.0 range with no clipping (half a bit of headroom traded for
+// clip-free RNE).
+
+__device__ __forceinline__ unsigned char f32_to_e2m1(float f) {
+  unsigned s = (__float_as_uint(f) >> 31) << 3;
+  float a = fabsf(f);
+  if (a != a) return (unsigned char)(s | 0x7);  // NaN -> max mag
+  // representable magnitudes: 0, .5, 1, 1.5, 2, 3, 4, 6 (codes 0..7)
+  // round-to-nearest-even on the midpoints
+  unsigned char m;
+  if (a <= 0.25f) m = 0;       // 0.25 -> even 0
+  else if (a < 0.75f) m = 1;
+  else if (a <= 1.25f) m = 2;       // 0.75..1.25 -> 1.0 (1.25->even 1.0)
+  else if (a < 1.75f) m = 3;
+  else if (a <= 2.5f) m = 4;        // 2.5 -> even 2.0
+  else if (a < 3.5f) m = 5;
+  else if (a <= 5.0f) m = 6;        // 5.0 -> even 4.0
+  else m = 7;
+  return (unsigned char)(s | m);
+}
+
+__global__ void mx4_convert_k(unsigned char* __restrict__ out4,
+                              unsigned char* __restrict__ sc,
+                              const bf16* __restrict__ in, long rows,
+                              long cols) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long kb = cols / 32;
+  if (idx >= rows * kb) return;
+  long r = idx / kb, b = idx - r * kb;
+  const bf16* src = in + r * cols + b * 32;
+  float v[32];
+  float mx = 0.0f;
+#pragma unroll
+  for (int i = 0; i < 32; i += 8) {
+    bf16x8v x = *(const bf16x8v*)(src + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)x[j];
+      v[i + j] = f;
+      float a = fabsf(f);
+      if (a > mx) mx = a;
+    }
+  }
+  int e;
+  // packed-dword scale layout (cols % 128 == 0 enforced by the binding)
+  long sidx = ((b >> 2) * rows + r) * 4 + (b & 3);
+  if (mx == 0.0f) {
+    e = 0;
+    sc[sidx] = 127;
+  } else {
+    e = (int)floorf(log2f(mx)) - 1;  // block max lands in [2, 4)
+    if (e < -127) e = -127;
+    if (e > 127) e = 127;
+    sc[sidx] = (unsigned char)(e + 127);
+  }
+  float inv = exp2f((float)-e);
+  // pack 32 nibbles into 16 bytes (4 dwords), original-k order
+  unsigned int words[4];
+#pragma unroll
+  for (int w = 0; w < 4; ++w) {
+    unsigned int x = 0;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      x |= (unsigned int)f32_to_e2m1(v[w * 8 + j] * inv) << (4 * j);
+    words[w] = x;
+  }
+  typedef unsigned int u32x4c __attribute__((ext_vector_type(4)));
+  *(u32x4c*)(out4 + r * (cols / 2) + b * 16) =
+      u32x4c{words[0], words[1], words[2], words[3]};
+}
+
+extern "C" void launch_mx4_convert(void* out4, void* sc, const void* in,
+                                   long rows, long cols, hipStream_t s) {
+  long n = rows * (cols / 32);
+  long blocks = (n + 255) / 256;
+  hipLaunchKernelGGL(mx4_convert_k, dim3((unsigned)blocks), dim3(256), 0,
+                     s, (unsigned char*)out4, (unsigned char*)sc,
+                     (const bf16*)in, rows, cols);
+}

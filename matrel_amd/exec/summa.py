@@ -79,15 +79,29 @@ def _bt_cached(b: BlockMatrix, k: int, bj: int,
     return bt
 
 
-def _mx8_cached(m: BlockMatrix, key, tensor_bf16: torch.Tensor):
-    """Per-matrix cache of MX-fp8 conversions (fp8 payload is half the
-    bf16 bytes; reused across every multiply touching the block)."""
+def _mxq_convert(prec: str, t: torch.Tensor):
+    return blockops.to_mx4(t) if prec == "mx4" else blockops.to_mx8(t)
+
+
+def _mxq_gemm(prec: str, c, a_p, a_sc, bt_p, bt_sc, beta0: bool) -> None:
+    if prec == "mx4":
+        blockops.gemm_bt_mx4_acc(c, a_p, a_sc, bt_p, bt_sc, beta0=beta0)
+    else:
+        blockops.gemm_bt_mx8_acc(c, a_p, a_sc, bt_p, bt_sc, beta0=beta0)
+
+
+def _mx8_cached(m: BlockMatrix, key, tensor_bf16: torch.Tensor,
+                prec: str = "mx8"):
+    """Per-matrix cache of MX-fp8/fp4 conversions (payloads are 1/2 or
+    1/4 the bf16 bytes; reused across every multiply touching the
+    block)."""
     cache = getattr(m, "_mx8_cache", None)
     if cache is None:
         cache = m._mx8_cache = {}
+    key = (prec,) + (key if isinstance(key, tuple) else (key,))
     ent = cache.get(key)
     if ent is None:
-        ent = blockops.to_mx8(tensor_bf16)
+        ent = _mxq_convert(prec, tensor_bf16)
         if _cache_has_headroom(sum(t.numel() * t.element_size()
                                    for t in ent)):
             cache[key] = ent
@@ -109,7 +123,7 @@ def _try_batch_local(a: BlockMatrix, b: BlockMatrix,
     small blocks (block-sparse workloads); falls back otherwise."""
     if out.device.type != "cuda" or out.dtype != torch.bfloat16:
         return False
-    if get_config().gemm_precision == "mx8":
+    if get_config().gemm_precision in ("mx8", "mx4"):
         return False
     bs = out.block_size
     if bs % 256 != 0 or bs > 8192:
@@ -167,14 +181,16 @@ def _pad_t(t: torch.Tensor, bs: int) -> torch.Tensor:
     return pad
 
 
-def _mx8_padded(m: BlockMatrix, key, t: torch.Tensor, bs: int):
-    """Cached MX-fp8 conversion of a (zero-padded) block."""
+def _mx8_padded(m: BlockMatrix, key, t: torch.Tensor, bs: int,
+                prec: str = "mx8"):
+    """Cached MX-fp8/fp4 conversion of a (zero-padded) block."""
     cache = getattr(m, "_mx8_cache", None)
     if cache is None:
         cache = m._mx8_cache = {}
+    key = (prec,) + (key if isinstance(key, tuple) else (key,))
     ent = cache.get(key)
     if ent is None:
-        ent = blockops.to_mx8(_pad_t(blockops.densify(t), bs))
+        ent = _mxq_convert(prec, _pad_t(blockops.densify(t), bs))
         if _cache_has_headroom(sum(x.numel() * x.element_size()
                                    for x in ent)):
             cache[key] = ent
@@ -209,7 +225,7 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
     launches."""
     if out.device.type != "cuda" or out.dtype != torch.bfloat16:
         return False
-    if get_config().gemm_precision == "mx8":
+    if get_config().gemm_precision in ("mx8", "mx4"):
         return False
     bs = out.block_size
     if bs % 256 != 0 or bs > 4096 or b.grid_cols < 4:
@@ -319,7 +335,9 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     if _try_batch_local(a, b, out):
         return
     fused_ok = (out.device.type == "cuda" and out.dtype == torch.bfloat16)
-    mx8 = get_config().gemm_precision == "mx8"
+    _prec = get_config().gemm_precision
+    mxq = _prec if _prec in ("mx8", "mx4") else None
+    mx8 = mxq is not None
     a_dense_cache: Dict[Tuple[int, int], torch.Tensor] = {}
 
     def a_routed(bi: int, k: int) -> torch.Tensor:
@@ -364,10 +382,9 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             bt = _bt_cached(b, k, bj, bbd)
             if mx8 and _mx8_usable(ab) and _mx8_usable(bt) \
                     and ab.shape[0] % 256 == 0 and bt.shape[0] % 256 == 0:
-                a8, asc = _mx8_cached(a, (bi, k), ab)
-                bt8, btsc = _mx8_cached(b, ("t", k, bj), bt)
-                blockops.gemm_bt_mx8_acc(c, a8, asc, bt8, btsc,
-                                         beta0=first)
+                a8, asc = _mx8_cached(a, (bi, k), ab, mxq)
+                bt8, btsc = _mx8_cached(b, ("t", k, bj), bt, mxq)
+                _mxq_gemm(mxq, c, a8, asc, bt8, btsc, beta0=first)
                 continue
             if use_fused and last:
                 fused_done = blockops.gemm_bt_final(c, outblk, ab, bt,
@@ -437,11 +454,15 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     nmb = b.nnz_map()
     sp_thr = get_config().sparse_threshold
     bs = a.block_size
-    # MX-fp8 SUMMA: owners convert panels to fp8+scales ONCE and the
-    # broadcasts move HALF the bf16 bytes over xGMI (+ 1/32 for scales).
-    mx8 = (get_config().gemm_precision == "mx8"
-           and a.device.type == "cuda" and a.dtype == torch.bfloat16
-           and b.dtype == torch.bfloat16 and bs % 256 == 0)
+    # MX-fp8/fp4 SUMMA: owners convert panels to quantized payloads +
+    # scales ONCE and the broadcasts move 1/2 (fp8) or 1/4 (fp4) of the
+    # bf16 bytes over xGMI (+ 1/32 for scales).
+    _prec = get_config().gemm_precision
+    mxq = (_prec if _prec in ("mx8", "mx4") else None) if (
+        a.device.type == "cuda" and a.dtype == torch.bfloat16
+        and b.dtype == torch.bfloat16 and bs % 256 == 0) else None
+    mx8 = mxq is not None
+    pbytes = bs if mxq != "mx4" else bs // 2  # payload bytes per row
 
     my_is = [i for i in range(out.grid_rows) if i % pr == p]
     my_js = [j for j in range(out.grid_cols) if j % pc == q]
@@ -480,7 +501,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             a_rows = [i for i in my_is if pa[i, k]]
             a_dn_index = {i: n for n, i in enumerate(a_rows)}
             if mx8:
-                a_stack = torch.empty(len(a_rows), bs, bs,
+                a_stack = torch.empty(len(a_rows), bs, pbytes,
                                       dtype=torch.uint8, device=a.device)
                 # packed-dword scales [bs/128][bs] int32 (mxconvert r2)
                 a_sc = torch.empty(len(a_rows), bs // 128, bs,
@@ -491,7 +512,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                         for n, i in enumerate(a_rows):
                             p8, sc = _mx8_padded(a, ("s", i, k),
                                                  a.get_block_dense(i, k),
-                                                 bs)
+                                                 bs, mxq)
                             a_stack[n] = p8
                             a_sc[n] = sc
                     _prep_done()
@@ -602,7 +623,7 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
             b_cols = [j for j in my_js if pb[k, j]]
             b_dn_index = {j: n for n, j in enumerate(b_cols)}
             if mx8:
-                b_stack = torch.empty(len(b_cols), bs, bs,
+                b_stack = torch.empty(len(b_cols), bs, pbytes,
                                       dtype=torch.uint8, device=b.device)
                 b_sc = torch.empty(len(b_cols), bs // 128, bs,
                                    dtype=torch.int32, device=b.device)
@@ -613,7 +634,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                             t = _pad_t(blockops.densify(
                                 b.get_block_dense(k, j)), bs)
                             p8, sc = _mx8_padded(b, ("st", k, j),
-                                                 blockops.transpose(t), bs)
+                                                 blockops.transpose(t),
+                                                 bs, mxq)
                             b_stack[n] = p8
                             b_sc[n] = sc
                     _prep_done()
@@ -823,9 +845,8 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                     continue
                 n_i = a_dn[i]
                 if mx8:
-                    blockops.gemm_bt_mx8_acc(c, a_stack[n_i], a_sc[n_i],
-                                             bt, b_sc[b_dn[j]],
-                                             beta0=first)
+                    _mxq_gemm(mxq, c, a_stack[n_i], a_sc[n_i],
+                              bt, b_sc[b_dn[j]], beta0=first)
                     continue
                 bt_j = _need_bt() if bt is None else bt
                 if fused_ok and last_k.get((i, j)) == k:

@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "spmm.hip"),
     os.path.join(CSRC, "spgemm.hip"),
     os.path.join(CSRC, "gemm_mx8.hip"),
+    os.path.join(CSRC, "gemm_mx4.hip"),
     os.path.join(CSRC, "mxconvert.hip"),
 ]
 

@@ -262,18 +262,19 @@ def _body_relational(s, tmpdir):
 
 
 def _body_mx8_config(s, tmpdir):
-    """gemm_precision='mx8' on a CPU/gloo world: the mx8 branch requires
-    CUDA and must fall back UNIFORMLY on every rank (a rank-divergent
-    decision would deadlock the panel broadcasts)."""
+    """gemm_precision='mx8'/'mx4' on a CPU/gloo world: the quantized
+    branches require CUDA and must fall back UNIFORMLY on every rank (a
+    rank-divergent decision would deadlock the panel broadcasts)."""
     import matrel_amd as ma
     cfg = ma.get_config()
-    cfg.gemm_precision = "mx8"
     try:
-        A = s.randn(96, 96, seed=61)
-        B = s.randn(96, 64, seed=62)
-        C = (A @ B).compute()
-        ref = A.to_tensor() @ B.to_tensor()
-        assert torch.allclose(C.to_tensor(), ref, atol=1e-3)
+        for prec in ("mx8", "mx4"):
+            cfg.gemm_precision = prec
+            A = s.randn(96, 96, seed=61)
+            B = s.randn(96, 64, seed=62)
+            C = (A @ B).compute()
+            ref = A.to_tensor() @ B.to_tensor()
+            assert torch.allclose(C.to_tensor(), ref, atol=1e-3), prec
     finally:
         cfg.gemm_precision = "bf16"
 
