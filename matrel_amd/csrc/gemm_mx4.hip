@@ -5,16 +5,18 @@
 // digits; docs/NUMERICS.md); the engine default stays bf16.
 //
 // Operand semantics verified empirically by scripts/mx4_probe.hip:
-// exactly the fp8 layout of csrc/gemm_mx8.hip at NIBBLE granularity -
 // lane l carries 32 e2m1 nibbles (16 bytes, the LOW v4i of the v8i
 // operand) of row/col l&31 with
 //   hw_k(l, nib) = (l>>5)*16 + (nib & 15) + (nib >> 4)*32,
-// one e8m0 scale byte per lane covering hw block l>>5. The fragment
-// reads below place original k [kh*16,+16) in nibbles 0-15 (memory
-// bytes [kh*8,+8) of the 32-B k-block) and [32+kh*16,+16) in nibbles
-// 16-31 (bytes [16+kh*8,+8)) so scale blocks stay contiguous in
-// original k - the nibble-granularity mirror of the fp8 kernel's
-// byte-half trick.
+// and - UNLIKE the engine's fp8 model - the e8m0 scale applies PER
+// SOURCE LANE (to the lane's own 32 nibbles; probe "scale-granularity"
+// test). The fragment reads therefore load each lane's nibbles with
+// ORIGINAL k CONTIGUOUS: nibble n of lane (row, kh) holds original
+// k = kk*64 + kh*32 + n (one 16-B slot per (kk, kh): a single
+// ds_read_b128). The induced hw_k permutation is identical on A and B,
+// so products still pair equal original k, and the lane scale covers
+// exactly original 32-block 4t + 2*kk + kh - the byte the kernel
+// extracts from the packed scale word.
 //
 // Storage: A4/Bt4 row-major packed nibbles ([rows][K/2] bytes); scales
 // packed-dword [K/128][rows] int32 (same layout as MX-fp8 round 2).
@@ -30,12 +32,11 @@ typedef int v2i4 __attribute__((ext_vector_type(2)));
 typedef float f32x16q __attribute__((ext_vector_type(16)));
 
 // build the v8i operand: payload in the LOW 4 regs (mx4_probe: hi=0)
-__device__ __forceinline__ v8i4 v4_from(const void* p_lo,
-                                        const void* p_hi) {
-  v2i4 lo = *(const v2i4*)p_lo;   // ds_read_b64: nibbles 0-15
-  v2i4 hi = *(const v2i4*)p_hi;   // ds_read_b64: nibbles 16-31
+__device__ __forceinline__ v8i4 v4_from16(const void* p) {
+  typedef int v4i4 __attribute__((ext_vector_type(4)));
+  v4i4 lo = *(const v4i4*)p;      // one ds_read_b128: nibbles 0-31
   v8i4 r = {};
-  r[0] = lo[0]; r[1] = lo[1]; r[2] = hi[0]; r[3] = hi[1];
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
   return r;
 }
 
@@ -127,23 +128,19 @@ gemm_bt_mx4_acc_k128(float* __restrict__ C,
 #pragma unroll
       for (int m = 0; m < 4; ++m) {
         u32 row = arow + m * 32;
-        // base = row*64 + kh*8; kk selects bit 5, hi-half bit 4 - both
-        // inside the swizzle's XOR range, so they XOR in (mx4_probe +
-        // the round-2 bit-6 lesson from the fp8 quadrant port)
-        u32 byte = row * 64u + kh * 8u;
-        byte ^= ((row >> 2) & 3) << 4;
-        u32 b0 = byte ^ ((u32)kk * 32u);
-        af[m] = v4_from((const char*)a_base + b0,
-                        (const char*)a_base + (b0 ^ 16u));
+        // slot = kk*2 + kh of the 64-B row (16 B = the lane's 32
+        // contiguous original-k nibbles); slot bits XOR the staging
+        // swizzle
+        u32 byte = row * 64u + (((u32)kk * 2u + kh) ^ ((row >> 2) & 3))
+                   * 16u;
+        af[m] = v4_from16((const char*)a_base + byte);
       }
 #pragma unroll
       for (int n = 0; n < 2; ++n) {
         u32 row = brow + n * 32;
-        u32 byte = row * 64u + kh * 8u;
-        byte ^= ((row >> 2) & 3) << 4;
-        u32 b0 = byte ^ ((u32)kk * 32u);
-        bf[n] = v4_from((const char*)b_base + b0,
-                        (const char*)b_base + (b0 ^ 16u));
+        u32 byte = row * 64u + (((u32)kk * 2u + kh) ^ ((row >> 2) & 3))
+                   * 16u;
+        bf[n] = v4_from16((const char*)b_base + byte);
       }
 #pragma unroll
       for (int m = 0; m < 4; ++m)

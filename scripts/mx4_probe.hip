@@ -33,8 +33,10 @@ __global__ void probe4_k(const unsigned char* __restrict__ A,
     b[i + (hi ? 4 : 0)] = bp[i];
   }
   f32x16 c = {};
+  int sa_eff = sa;
+  if (sa == 2000) sa_eff = (lane < 32) ? 128 : 127;  // by source kh
   c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 4, 4, 0,
-                                                      sa, 0, sb);
+                                                      sa_eff, 0, sb);
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
@@ -156,6 +158,20 @@ int main() {
     printf("cross A(l%d,n%d) B(l%d,n%d): got=%g want %s\n", cse.la,
            cse.na, cse.lb, cse.nb, got, cse.want ? "hit" : "0");
   }
+
+  // SCALE GRANULARITY: A lane-kh0 nibs 16-31 = 1 (hw_k 32-47), rest 0;
+  // B all ones; sa = 128 on kh0 lanes / 127 on kh1 (per-lane via kernel
+  // mode 2000). Per-SOURCE-LANE scaling -> those products double ->
+  // C[0][0] = 32; per-HW-BLOCK (hw 32-47 = block1 = kh1's scale=127)
+  // -> C[0][0] = 16.
+  memset(hA, 0, sizeof(hA));
+  memset(hB, (one << 4) | one, sizeof(hB));
+  for (int l = 0; l < 32; ++l)          // kh=0 lanes only
+    for (int nib = 16; nib < 32; ++nib)
+      set_nib(hA, l, nib, one);
+  run(2000, 127, HI);
+  printf("scale-granularity: C[0][0]=%g (32 => per-source-lane, "
+         "16 => per-hw-block)\n", hC[0]);
 
   // end-to-end: random e2m1 payloads + random scales vs host reference
   srand(7);

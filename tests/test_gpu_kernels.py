@@ -549,9 +549,17 @@ def test_mx4_convert_roundtrip(ext):
     deq = _dequant_e2m1(a4, sc, 256, 256)
     ref = a.float().cpu()
     err = (deq - ref).abs()
-    # e2m1: 1 mantissa bit in [2,4) normalization -> rel err <= ~25%
-    tol = ref.abs() * 0.30 + 0.02
-    assert (err <= tol).all(), (err / (ref.abs() + 1e-6)).max()
+    # e2m1: 1 mantissa bit in [2,4) normalization -> rel err <= ~20%,
+    # PLUS block-relative underflow: values below 0.25 * 2^e of their
+    # 32-block quantize to ZERO (the format's ~12:1 in-block dynamic
+    # range), so the tolerance floor is 0.25 * blockscale.
+    kt = 256 // 128
+    scb = sc.cpu().view(torch.uint8).reshape(kt, 256, 4) \
+        .permute(0, 2, 1).reshape(kt * 4, 256)
+    bscale = (2.0 ** (scb.to(torch.int32).float() - 127)).t() \
+        .repeat_interleave(32, dim=1)
+    tol = ref.abs() * 0.20 + 0.25 * bscale + 1e-6
+    assert (err <= tol).all(), (err - tol).max()
 
 
 def test_mx4_gemm_exact_vs_dequant(ext):
