@@ -39,7 +39,7 @@ def main():
                     help="override matrix side N")
     ap.add_argument("--block-size", type=int, default=0,
                     help="0 = auto (single block at 1 GPU, 16384 for SUMMA)")
-    ap.add_argument("--precision", choices=["bf16", "mx8", "mx4"],
+    ap.add_argument("--precision", choices=["bf16", "mx8", "mx6", "mx4"],
                     default="bf16",
                     help="GEMM compute precision (mx8 = MX-scaled fp8; "
                          "NOT the headline config - BASELINE is bf16)")
@@ -112,7 +112,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": ({"mx8": "fp8_mx", "mx4": "fp4_mx"}.get(args.precision, "bf16"))
+            "dtype": ({"mx8": "fp8_mx", "mx6": "fp6_mx", "mx4": "fp4_mx"}.get(args.precision, "bf16"))
                      if has_gpu else "fp32",
             "data": "synthetic",
             "config": {

@@ -52,9 +52,10 @@ class MatrelConfig:
     # Gram matrix). Refuse silently-catastrophic gathers above this side.
     solve_max_side: int = 32768
     # GEMM compute precision: "bf16" (default), "mx8" (MX-scaled fp8
-    # e4m3, ~2x the bf16 MFMA rate at ~2-3 decimal digits) or "mx4"
+    # e4m3, ~2x the bf16 MFMA rate at ~2-3 decimal digits), "mx6"
+    # (MX fp6 e2m3: fp8-class accuracy at the fp4 rate) or "mx4"
     # (MX-scaled fp4 e2m1, ~2x the fp8 rate at ~1-2 significant digits -
-    # screening-pass precision). Both opt-in.
+    # screening-pass precision). All opt-in.
     gemm_precision: str = "bf16"
 
     # --- sparsity ---

@@ -18,6 +18,9 @@ void launch_gemm_bt_mx8_acc(float*, const void*, const void*, const void*,
                             const void*, int, int, int, int, hipStream_t);
 void launch_mx8_convert(void*, void*, const void*, long, long, hipStream_t);
 void launch_mx4_convert(void*, void*, const void*, long, long, hipStream_t);
+void launch_mx6_convert(void*, void*, const void*, long, long, hipStream_t);
+void launch_gemm_bt_mx6_acc(float*, const void*, const void*, const void*,
+                            const void*, int, int, int, int, hipStream_t);
 void launch_gemm_bt_mx4_acc(float*, const void*, const void*, const void*,
                             const void*, int, int, int, int, hipStream_t);
 void launch_gemm_bt_f32_acc(float*, const float*, const float*, int, int,
@@ -396,6 +399,48 @@ std::vector<torch::Tensor> mx4_convert(torch::Tensor a) {
   return {out4, sc};
 }
 
+// bf16 -> MX-fp6: packed e2m3 6-bit fields [rows][cols*3/4] + packed-dword
+// e8m0 scales [cols/128][rows] (same scale layout as MX-fp8/fp4)
+std::vector<torch::Tensor> mx6_convert(torch::Tensor a) {
+  check_2d(a, "a");
+  TORCH_CHECK(is_bf16(a), "mx6_convert needs bf16");
+  long rows = a.size(0), cols = a.size(1);
+  TORCH_CHECK(cols % 128 == 0, "cols must be a multiple of 128");
+  auto opts = torch::TensorOptions().dtype(torch::kUInt8).device(a.device());
+  auto out6 = torch::empty({rows, cols * 3 / 4}, opts);
+  auto sc = torch::empty({cols / 128, rows},
+                         torch::TensorOptions().dtype(torch::kInt32)
+                             .device(a.device()));
+  launch_mx6_convert(out6.data_ptr(), sc.data_ptr(), a.data_ptr(), rows,
+                     cols, cur_stream());
+  return {out6, sc};
+}
+
+// C(f32) += A6 @ Bt6^T with per-32-block e8m0 scales (MX-fp6 e2m3:
+// fp8-class accuracy at the fp4 MFMA rate; opt-in)
+void gemm_bt_mx6_acc(torch::Tensor C, torch::Tensor A6, torch::Tensor Asc,
+                     torch::Tensor Bt6, torch::Tensor Bsc, bool beta0) {
+  check_2d(C, "C");
+  check_2d(A6, "A6");
+  check_2d(Bt6, "Bt6");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(A6.scalar_type() == torch::kUInt8 &&
+              Bt6.scalar_type() == torch::kUInt8);
+  int M = (int)A6.size(0), K = (int)(A6.size(1) * 4 / 3),
+      N = (int)Bt6.size(0);
+  TORCH_CHECK(Bt6.size(1) * 4 / 3 == K && C.size(0) == M && C.size(1) == N);
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 128 == 0,
+              "mx6 GEMM needs M,N % 256 == 0, K % 128 == 0");
+  TORCH_CHECK(Asc.scalar_type() == torch::kInt32 &&
+              Bsc.scalar_type() == torch::kInt32 &&
+              Asc.size(0) == K / 128 && Asc.size(1) == M &&
+              Bsc.size(0) == K / 128 && Bsc.size(1) == N,
+              "packed scales must be [K/128][rows] int32");
+  launch_gemm_bt_mx6_acc(C.data_ptr<float>(), A6.data_ptr(),
+                         Asc.data_ptr(), Bt6.data_ptr(), Bsc.data_ptr(),
+                         M, N, K, beta0 ? 1 : 0, cur_stream());
+}
+
 // C(f32) += A4 @ Bt4^T with per-32-block e8m0 scales (MX-fp4, ~2x the
 // MX-fp8 rate at ~1-2 significant digits; opt-in)
 void gemm_bt_mx4_acc(torch::Tensor C, torch::Tensor A4, torch::Tensor Asc,
@@ -594,6 +639,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bf16 -> (e2m1 nibbles, e8m0 block scales)");
   m.def("gemm_bt_mx4_acc", &gemm_bt_mx4_acc,
         "C(f32) += A4 @ Bt4^T (MX-fp4 scaled MFMA, ~2x the fp8 rate)");
+  m.def("mx6_convert", &mx6_convert,
+        "bf16 -> (e2m3 6-bit fields, e8m0 block scales)");
+  m.def("gemm_bt_mx6_acc", &gemm_bt_mx6_acc,
+        "C(f32) += A6 @ Bt6^T (MX-fp6 scaled MFMA at the fp4 rate)");
   m.def("gemm_bt_mx8_acc", &gemm_bt_mx8_acc,
         "C(f32) += A8 @ Bt8^T (MX-fp8 scaled MFMA, ~2x bf16 rate)");
 }

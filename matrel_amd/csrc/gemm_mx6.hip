@@ -1,0 +1,185 @@
+// MX-fp6 (e2m3 + e8m0 block scales) GEMM for gfx950:
+//   C(f32, MxN) += A6(MxK) @ Bt6(NxK)^T
+// on v_mfma_scale_f32_32x32x64_f8f6f4 with cbsz=blgp=2. CDNA4 runs fp6
+// at the FP4 rate (~10 PF dense) - i.e. fp8-class accuracy (3 mantissa
+// bits, ~6% per-element) at ~2x the fp8 GEMM rate. Opt-in
+// (gemm_precision="mx6"); the engine default stays bf16.
+//
+// Operand semantics verified by scripts/mx6_probe.hip: payload = 24
+// bytes/lane in the LOW 6 regs, LSB-first 6-bit fields, field k-map =
+// the fp4 nibble map at field granularity, scale PER SOURCE LANE. As in
+// the fp4 kernel, each lane's fields hold ORIGINAL k contiguous
+// (k = kk*64 + kh*32 + field): the hw permutation cancels between A and
+// B and the lane scale covers original 32-block 4t + 2*kk + kh.
+//
+// Storage: [rows][K*3/4] bytes; scales packed-dword [K/128][rows] int32.
+// Structure: the k128 shape - 256^2 tile, 8 waves, K-tile 128 (row =
+// 96 B), double-buffered LDS (2 x 2 x 24 KiB = 96 KiB), one barrier per
+// tile, scale rotation. No LDS swizzle: the 96-B row stride rotates
+// banks by 24 dwords/row naturally (measured adequate; revisit if PMC
+// shows conflicts).
+#include "common.h"
+
+extern unsigned g_st_m, g_st_n;  // supertile dims (gemm.hip)
+
+typedef int v8i6 __attribute__((ext_vector_type(8)));
+typedef int v4i6 __attribute__((ext_vector_type(4)));
+typedef int v2i6 __attribute__((ext_vector_type(2)));
+typedef float f32x16h __attribute__((ext_vector_type(16)));
+
+__device__ __forceinline__ v8i6 v6_from24(const void* p) {
+  v4i6 lo = *(const v4i6*)p;                    // ds_read_b128
+  v2i6 hi = *(const v2i6*)((const char*)p + 16);  // ds_read_b64
+  v8i6 r = {};
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+  r[4] = hi[0]; r[5] = hi[1];
+  return r;
+}
+
+#define M6BM 256
+#define M6BN 256
+#define M6ROWB 96
+
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx6_acc_k128(float* __restrict__ C,
+                     const unsigned char* __restrict__ A,
+                     const unsigned char* __restrict__ Asc,
+                     const unsigned char* __restrict__ Bt,
+                     const unsigned char* __restrict__ Bsc, int M, int N,
+                     int K, int beta0, u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned char* As = (unsigned char*)smem;  // [2][256*96]
+  unsigned char* Bs = (unsigned char*)(smem + 2 * M6BM * M6ROWB);
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;
+
+  const u32 ntm = (u32)(M / M6BM), ntn = (u32)(N / M6BN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * M6BM, n0 = (long)tn * M6BN;
+  const long KB6 = (long)K * 3 / 4;  // packed row bytes
+
+  // staging: tile = 256 rows x 96 B = 1536 x 16-B chunks; 512 threads
+  // x 3 passes per operand.
+  auto stage = [&](int buf, int t) {
+    const long k0b = (long)t * 96;  // K-tile of 128 = 96 packed bytes
+#pragma unroll
+    for (int p = 0; p < 3; ++p) {
+      u32 c = tid + (u32)p * 512;
+      u32 row = c / 6, slot = c - row * 6;
+      glds16(A + (m0 + row) * KB6 + k0b + slot * 16,
+             As + buf * M6BM * M6ROWB + row * M6ROWB + slot * 16);
+      glds16(Bt + (n0 + row) * KB6 + k0b + slot * 16,
+             Bs + buf * M6BM * M6ROWB + row * M6ROWB + slot * 16);
+    }
+  };
+
+  f32x16h acc[4][2] = {};
+  const u32 arow = wm * 128 + (lane & 31);
+  const u32 brow = wn * 64 + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  auto load_scales = [&](int t, int* sa_w, int* sb_w) {
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      sa_w[m] = ((const int*)Asc)[(long)t * M + m0 + arow + m * 32];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+      sb_w[n] = ((const int*)Bsc)[(long)t * N + n0 + brow + n * 32];
+  };
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  int sa_cur[4], sb_cur[2], sa_nxt[4], sb_nxt[2];
+  load_scales(0, sa_cur, sb_cur);
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  const int NT = K / 128;
+  for (int t = 0; t < NT; ++t) {
+    const bool pre = (t + 1 < NT);
+    if (pre) {
+      stage(cur ^ 1, t + 1);
+      load_scales(t + 1, sa_nxt, sb_nxt);
+    }
+    const unsigned char* a_base = As + cur * M6BM * M6ROWB;
+    const unsigned char* b_base = Bs + cur * M6BM * M6ROWB;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      v8i6 af[4], bf[2];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        u32 row = arow + m * 32;
+        u32 byte = row * M6ROWB + ((u32)kk * 2u + kh) * 24u;
+        af[m] = v6_from24((const char*)a_base + byte);
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        u32 row = brow + n * 32;
+        u32 byte = row * M6ROWB + ((u32)kk * 2u + kh) * 24u;
+        bf[n] = v6_from24((const char*)b_base + byte);
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af[m], bf[n], acc[m][n], 2, 2, 0,
+              (sa_cur[m] >> (8 * (kk * 2 + (int)kh))) & 0xFF, 0,
+              (sb_cur[n] >> (8 * (kk * 2 + (int)kh))) & 0xFF);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int m = 0; m < 4; ++m) sa_cur[m] = sa_nxt[m];
+#pragma unroll
+    for (int n = 0; n < 2; ++n) sb_cur[n] = sb_nxt[n];
+    cur ^= 1;
+  }
+
+  const long ccol = n0 + wn * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      float cv[16];
+      if (!beta0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                     8 * (r >> 2) + 4 * (lane >> 5);
+          cv[r] = C[row * (long)N + ccol + n * 32];
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                   8 * (r >> 2) + 4 * (lane >> 5);
+        C[row * (long)N + ccol + n * 32] =
+            (beta0 ? 0.0f : cv[r]) + acc[m][n][r];
+      }
+    }
+}
+
+extern "C" void launch_gemm_bt_mx6_acc(float* C, const void* A,
+                                       const void* Asc, const void* Bt,
+                                       const void* Bsc, int M, int N,
+                                       int K, int beta0,
+                                       hipStream_t stream) {
+  dim3 grid((M / M6BM) * (N / M6BN));
+  size_t lds = 4u * M6BM * M6ROWB;  // 96 KiB
+  hipLaunchKernelGGL(gemm_bt_mx6_acc_k128, grid, dim3(512), lds, stream,
+                     C, (const unsigned char*)A,
+                     (const unsigned char*)Asc, (const unsigned char*)Bt,
+                     (const unsigned char*)Bsc, M, N, K, beta0, g_st_m,
+                     g_st_n);
+}

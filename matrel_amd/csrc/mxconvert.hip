@@ -188,3 +188,97 @@ extern "C" void launch_mx4_convert(void* out4, void* sc, const void* in,
                      s, (unsigned char*)out4, (unsigned char*)sc,
                      (const bf16*)in, rows, cols);
 }
+
+// ---- bf16 -> MX-fp6 (e2m3 + e8m0 per-32-block scales) -------------------
+// Layout verified by scripts/mx6_probe.hip: LSB-first 6-bit fields
+// (32 per block = 24 bytes), per-SOURCE-LANE scale coverage like fp4.
+// Scale exponent: e = floor(log2 max) - 2 normalizes the block max into
+// [4, 8) (e2m3 tops out at 7.5: only a max element in (7.25, 8) rounds
+// onto 7.5 - <= 6% on that one element). Payload: [rows][cols*3/4]
+// bytes; scales: packed-dword [cols/128][rows] int32 (same as fp8/fp4).
+
+__device__ __forceinline__ unsigned char f32_to_e2m3(float f) {
+  unsigned s = (__float_as_uint(f) >> 31) << 5;
+  float a = fabsf(f);
+  if (a != a) return (unsigned char)(s | 0x1F);
+  // grid: subnormals m/8 (m 0..7), then (1+m/8)*2^(e-1), e 1..3; RNE
+  // equivalently: q = round(a * 8) in [0, 15] covers codes 0..15
+  // (values 0..1.875 step .125); above that the step doubles per octave
+  unsigned c;
+  if (a < 2.0f) {
+    int q = (int)rintf(a * 8.0f);
+    if (q > 15) q = 16;  // 1.9375.. -> 2.0
+    c = (unsigned)q;
+    if (q == 16) c = 0x10;
+  } else if (a < 4.0f) {
+    int q = (int)rintf(a * 4.0f);  // step .25: codes 0x10..0x17
+    if (q > 15) q = 16;
+    c = (unsigned)(0x10 + (q - 8));
+    if (q == 16) c = 0x18;
+  } else {
+    int q = (int)rintf(a * 2.0f);  // step .5: codes 0x18..0x1F
+    if (q > 15) q = 15;            // saturate at 7.5
+    if (q < 8) q = 8;
+    c = (unsigned)(0x18 + (q - 8));
+  }
+  return (unsigned char)(s | c);
+}
+
+__global__ void mx6_convert_k(unsigned char* __restrict__ out6,
+                              unsigned char* __restrict__ sc,
+                              const bf16* __restrict__ in, long rows,
+                              long cols) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long kb = cols / 32;
+  if (idx >= rows * kb) return;
+  long r = idx / kb, b = idx - r * kb;
+  const bf16* src = in + r * cols + b * 32;
+  float v[32];
+  float mx = 0.0f;
+#pragma unroll
+  for (int i = 0; i < 32; i += 8) {
+    bf16x8v x = *(const bf16x8v*)(src + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)x[j];
+      v[i + j] = f;
+      float a = fabsf(f);
+      if (a > mx) mx = a;
+    }
+  }
+  int e;
+  long sidx = ((b >> 2) * rows + r) * 4 + (b & 3);
+  if (mx == 0.0f) {
+    e = 0;
+    sc[sidx] = 127;
+  } else {
+    e = (int)floorf(log2f(mx)) - 2;  // block max lands in [4, 8)
+    if (e < -127) e = -127;
+    if (e > 127) e = 127;
+    sc[sidx] = (unsigned char)(e + 127);
+  }
+  float inv = exp2f((float)-e);
+  // pack 32 x 6-bit fields LSB-first into 24 bytes (6 dwords)
+  unsigned int w[6] = {0, 0, 0, 0, 0, 0};
+#pragma unroll
+  for (int f = 0; f < 32; ++f) {
+    unsigned int code = f32_to_e2m3(v[f] * inv);
+    int bit = f * 6, d = bit >> 5, off = bit & 31;
+    w[d] |= code << off;
+    if (off > 26) w[d + 1] |= code >> (32 - off);
+  }
+  unsigned char* dst = out6 + r * (cols * 3 / 4) + b * 24;
+  typedef unsigned int u32x2c __attribute__((ext_vector_type(2)));
+  ((u32x2c*)dst)[0] = u32x2c{w[0], w[1]};
+  ((u32x2c*)dst)[1] = u32x2c{w[2], w[3]};
+  ((u32x2c*)dst)[2] = u32x2c{w[4], w[5]};
+}
+
+extern "C" void launch_mx6_convert(void* out6, void* sc, const void* in,
+                                   long rows, long cols, hipStream_t s) {
+  long n = rows * (cols / 32);
+  long blocks = (n + 255) / 256;
+  hipLaunchKernelGGL(mx6_convert_k, dim3((unsigned)blocks), dim3(256), 0,
+                     s, (unsigned char*)out6, (unsigned char*)sc,
+                     (const bf16*)in, rows, cols);
+}

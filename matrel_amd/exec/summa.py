@@ -79,15 +79,33 @@ def _bt_cached(b: BlockMatrix, k: int, bj: int,
     return bt
 
 
+_MXQ_PRECS = ("mx8", "mx6", "mx4")
+
+
 def _mxq_convert(prec: str, t: torch.Tensor):
-    return blockops.to_mx4(t) if prec == "mx4" else blockops.to_mx8(t)
+    if prec == "mx4":
+        return blockops.to_mx4(t)
+    if prec == "mx6":
+        return blockops.to_mx6(t)
+    return blockops.to_mx8(t)
 
 
 def _mxq_gemm(prec: str, c, a_p, a_sc, bt_p, bt_sc, beta0: bool) -> None:
     if prec == "mx4":
         blockops.gemm_bt_mx4_acc(c, a_p, a_sc, bt_p, bt_sc, beta0=beta0)
+    elif prec == "mx6":
+        blockops.gemm_bt_mx6_acc(c, a_p, a_sc, bt_p, bt_sc, beta0=beta0)
     else:
         blockops.gemm_bt_mx8_acc(c, a_p, a_sc, bt_p, bt_sc, beta0=beta0)
+
+
+def _mxq_pbytes(prec, bs: int) -> int:
+    """Quantized payload bytes per row of a bs-wide panel."""
+    if prec == "mx4":
+        return bs // 2
+    if prec == "mx6":
+        return bs * 3 // 4
+    return bs
 
 
 def _mx8_cached(m: BlockMatrix, key, tensor_bf16: torch.Tensor,
@@ -123,7 +141,7 @@ def _try_batch_local(a: BlockMatrix, b: BlockMatrix,
     small blocks (block-sparse workloads); falls back otherwise."""
     if out.device.type != "cuda" or out.dtype != torch.bfloat16:
         return False
-    if get_config().gemm_precision in ("mx8", "mx4"):
+    if get_config().gemm_precision in _MXQ_PRECS:
         return False
     bs = out.block_size
     if bs % 256 != 0 or bs > 8192:
@@ -225,7 +243,7 @@ def _try_strip_local(a: BlockMatrix, b: BlockMatrix,
     launches."""
     if out.device.type != "cuda" or out.dtype != torch.bfloat16:
         return False
-    if get_config().gemm_precision in ("mx8", "mx4"):
+    if get_config().gemm_precision in _MXQ_PRECS:
         return False
     bs = out.block_size
     if bs % 256 != 0 or bs > 4096 or b.grid_cols < 4:
@@ -336,7 +354,7 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
         return
     fused_ok = (out.device.type == "cuda" and out.dtype == torch.bfloat16)
     _prec = get_config().gemm_precision
-    mxq = _prec if _prec in ("mx8", "mx4") else None
+    mxq = _prec if _prec in _MXQ_PRECS else None
     mx8 = mxq is not None
     a_dense_cache: Dict[Tuple[int, int], torch.Tensor] = {}
 
@@ -458,11 +476,11 @@ def summa_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
     # scales ONCE and the broadcasts move 1/2 (fp8) or 1/4 (fp4) of the
     # bf16 bytes over xGMI (+ 1/32 for scales).
     _prec = get_config().gemm_precision
-    mxq = (_prec if _prec in ("mx8", "mx4") else None) if (
+    mxq = (_prec if _prec in _MXQ_PRECS else None) if (
         a.device.type == "cuda" and a.dtype == torch.bfloat16
         and b.dtype == torch.bfloat16 and bs % 256 == 0) else None
     mx8 = mxq is not None
-    pbytes = bs if mxq != "mx4" else bs // 2  # payload bytes per row
+    pbytes = _mxq_pbytes(mxq, bs)  # payload bytes per row
 
     my_is = [i for i in range(out.grid_rows) if i % pr == p]
     my_js = [j for j in range(out.grid_cols) if j % pc == q]

@@ -342,6 +342,19 @@ def to_mx4(t_bf16: torch.Tensor):
     return tuple(require_ext().mx4_convert(t_bf16.contiguous()))
 
 
+def to_mx6(t_bf16: torch.Tensor):
+    """bf16 -> (packed e2m3 6-bit fields [rows][cols*3/4], packed e8m0
+    scales)."""
+    return tuple(require_ext().mx6_convert(t_bf16.contiguous()))
+
+
+def gemm_bt_mx6_acc(acc: torch.Tensor, a6, asc, bt6, btsc,
+                    beta0: bool = False) -> None:
+    """acc(fp32) += dequant(a6,asc) @ dequant(bt6,btsc)^T via the MX-fp6
+    scaled MFMA (e2m3: fp8-class accuracy at the fp4 rate)."""
+    require_ext().gemm_bt_mx6_acc(acc, a6, asc, bt6, btsc, beta0)
+
+
 def gemm_bt_mx4_acc(acc: torch.Tensor, a4, asc, bt4, btsc,
                     beta0: bool = False) -> None:
     """acc(fp32) += dequant(a4,asc) @ dequant(bt4,btsc)^T via the MX-fp4

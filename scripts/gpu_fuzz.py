@@ -22,7 +22,7 @@ ma.set_config(ma.MatrelConfig(block_size=BS, gemm_precision=PREC))
 s = ma.MatrelSession()
 # fp8 accumulates larger relative error through deep trees
 ATOL, RTOL = {"bf16": (0.15, 0.05), "mx8": (1.5, 0.2),
-              "mx4": (6.0, 0.75)}[PREC]
+              "mx6": (1.5, 0.2), "mx4": (6.0, 0.75)}[PREC]
 
 leaves = []
 for seed in (1, 2, 3):

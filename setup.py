@@ -26,6 +26,7 @@ sources = [
     os.path.join(CSRC, "spgemm.hip"),
     os.path.join(CSRC, "gemm_mx8.hip"),
     os.path.join(CSRC, "gemm_mx4.hip"),
+    os.path.join(CSRC, "gemm_mx6.hip"),
     os.path.join(CSRC, "mxconvert.hip"),
 ]
 

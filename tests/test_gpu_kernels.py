@@ -606,3 +606,88 @@ def test_mx4_end_to_end():
         assert rel < 0.25, rel
     finally:
         reset_runtime()
+
+# ------------------------------------------------------------- MX-fp6 ----
+
+def _dequant_e2m3(p6: torch.Tensor, sc: torch.Tensor, rows: int,
+                  cols: int) -> torch.Tensor:
+    """Host dequant of packed e2m3 6-bit fields (LSB-first) + packed-dword
+    e8m0 scales."""
+    b = p6.cpu().to(torch.int64).reshape(rows, cols * 3 // 4)
+    bits = torch.stack([(b >> i) & 1 for i in range(8)], dim=-1)
+    bits = bits.reshape(rows, cols, 6)
+    code = sum((bits[..., i] << i) for i in range(6))
+    e = (code >> 3) & 3
+    m = (code & 7).float()
+    mag = torch.where(e > 0, (1.0 + m / 8.0) * (2.0 ** (e - 1).float()),
+                      m / 8.0)
+    val = torch.where((code & 0x20) > 0, -mag, mag)
+    kt = cols // 128
+    scb = sc.cpu().view(torch.uint8).reshape(kt, rows, 4) \
+        .permute(0, 2, 1).reshape(kt * 4, rows)  # [K/32][rows]
+    scale = (2.0 ** (scb.to(torch.int32).float() - 127)).t()
+    return val.float() * scale.repeat_interleave(32, dim=1)
+
+
+def test_mx6_convert_roundtrip(ext):
+    a = _randn(256, 256, seed=70)
+    a6, sc = ext.mx6_convert(a)
+    assert a6.shape == (256, 192) and sc.shape == (2, 256)
+    deq = _dequant_e2m3(a6, sc, 256, 256)
+    ref = a.float().cpu()
+    err = (deq - ref).abs()
+    # e2m3 normalized to [4,8): branch half-steps give rel err <= 1/16
+    # with an absolute subnormal floor of bscale/16.
+    kt = 256 // 128
+    scb = sc.cpu().view(torch.uint8).reshape(kt, 256, 4) \
+        .permute(0, 2, 1).reshape(kt * 4, 256)
+    bscale = (2.0 ** (scb.to(torch.int32).float() - 127)).t() \
+        .repeat_interleave(32, dim=1)
+    tol = ref.abs() * 0.0625 + 0.0625 * bscale + 1e-6
+    assert (err <= tol).all(), (err - tol).max()
+
+
+def test_mx6_gemm_exact_vs_dequant(ext):
+    """The fp6 kernel must EXACTLY compute the dequantized product
+    (modulo fp32 accumulation order) - this pins the probe-verified
+    6-bit-field/scale layout end to end."""
+    M = N = 256
+    K = 512
+    a = _randn(M, K, seed=71)
+    bt = _randn(N, K, seed=72)
+    a6, asc = ext.mx6_convert(a)
+    bt6, btsc = ext.mx6_convert(bt)
+    c = torch.zeros(M, N, dtype=torch.float32, device="cuda")
+    ext.gemm_bt_mx6_acc(c, a6, asc, bt6, btsc, True)
+    da = _dequant_e2m3(a6, asc, M, K)
+    db = _dequant_e2m3(bt6, btsc, N, K)
+    ref = da @ db.t()
+    err = (c.cpu() - ref).abs().max().item()
+    assert err < 1e-2 * ref.abs().max().item() + 1e-3, err
+    # accumulate semantics
+    ext.gemm_bt_mx6_acc(c, a6, asc, bt6, btsc, False)
+    err2 = (c.cpu() - 2 * ref).abs().max().item()
+    assert err2 < 2e-2 * ref.abs().max().item() + 1e-3, err2
+
+
+def test_mx6_end_to_end():
+    """Full engine path at gemm_precision='mx6' (fp8-class accuracy at
+    the fp4 MFMA rate)."""
+    import math as _m
+    import matrel_amd as ma
+    from matrel_amd.exec.runtime import reset_runtime
+    reset_runtime()
+    s = ma.MatrelSession(ma.MatrelConfig(block_size=512,
+                                         gemm_precision="mx6"))
+    try:
+        A = s.randn(1024, 768, seed=73)
+        B = s.randn(768, 512, seed=74)
+        C = (A @ B).compute()
+        ref = A.to_tensor().float() @ B.to_tensor().float()
+        sigma = 0.05 * _m.sqrt(768)
+        err = (C.to_tensor().float() - ref).abs()
+        assert err.max().item() < 6 * sigma, err.max().item()
+        rel = (err.mean() / ref.abs().mean()).item()
+        assert rel < 0.06, rel
+    finally:
+        reset_runtime()
