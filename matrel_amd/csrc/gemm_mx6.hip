@@ -1,8 +1,8 @@
 // MX-fp6 (e2m3 + e8m0 block scales) GEMM for gfx950:
 //   C(f32, MxN) += A6(MxK) @ Bt6(NxK)^T
 // on v_mfma_scale_f32_32x32x64_f8f6f4 with cbsz=blgp=2. CDNA4 runs fp6
-// at the FP4 rate (~10 PF dense) - i.e. fp8-class accuracy (3 mantissa
-// bits, ~6% per-element) at ~2x the fp8 GEMM rate. Opt-in
+// at the FP4 MFMA rate - i.e. fp8-class accuracy (3 mantissa bits,
+// ~6% per-element) at up to ~2x the fp8 GEMM rate. Opt-in
 // (gemm_precision="mx6"); the engine default stays bf16.
 //
 // Operand semantics verified by scripts/mx6_probe.hip: payload = 24
@@ -12,12 +12,26 @@
 // (k = kk*64 + kh*32 + field): the hw permutation cancels between A and
 // B and the lane scale covers original 32-block 4t + 2*kk + kh.
 //
-// Storage: [rows][K*3/4] bytes; scales packed-dword [K/128][rows] int32.
-// Structure: the k128 shape - 256^2 tile, 8 waves, K-tile 128 (row =
-// 96 B), double-buffered LDS (2 x 2 x 24 KiB = 96 KiB), one barrier per
-// tile, scale rotation. No LDS swizzle: the 96-B row stride rotates
-// banks by 24 dwords/row naturally (measured adequate; revisit if PMC
-// shows conflicts).
+// Storage (split-plane; see mxconvert.hip): each row of A6/Bt6 is
+// [K/32 x 16B] (regs 0-3 of each block) then [K/32 x 8B] (regs 4-5);
+// scales packed-dword [K/128][rows] int32. The split keeps both LDS
+// read widths power-of-2:
+//   region L: [row][4 x 16B slots], slot XOR-swizzled by (row>>2)&3
+//     -> the mx4 kernel's proven conflict-free b128 pattern;
+//   region H: [row][2 x 16B pair-chunks], pair-chunk XOR-swizzled by
+//     (row>>3)&1 -> the b64 phase drops from 4-way to 2-way conflicts
+//     (2-way is the floor here: glds16 DMA destinations must be
+//     wave-uniform-base + lane*16, so LDS rows must be a linear 32 B
+//     and only whole 16-B pair-chunks can be permuted - via the GLOBAL
+//     source address, which is per-lane free).
+// A 24-B-contiguous layout instead puts ALL reads at a 24-dword lane
+// stride (2-way b128 + 4-way b64 conflicts) and measured only 1505 TF;
+// the split leaves the inner loop MFMA-bound (b64 is 1/3 of the LDS
+// traffic).
+//
+// Structure: the k128 shape - 256^2 tile, 8 waves, K-tile 128,
+// double-buffered LDS (2 x 2 x (16 KiB + 8 KiB) = 96 KiB), one
+// barrier per tile, scale rotation.
 #include "common.h"
 
 extern unsigned g_st_m, g_st_n;  // supertile dims (gemm.hip)
@@ -27,18 +41,25 @@ typedef int v4i6 __attribute__((ext_vector_type(4)));
 typedef int v2i6 __attribute__((ext_vector_type(2)));
 typedef float f32x16h __attribute__((ext_vector_type(16)));
 
-__device__ __forceinline__ v8i6 v6_from24(const void* p) {
-  v4i6 lo = *(const v4i6*)p;                    // ds_read_b128
-  v2i6 hi = *(const v2i6*)((const char*)p + 16);  // ds_read_b64
+#define M6BM 256
+#define M6BN 256
+#define M6LROW 64   // region L row bytes (4 x 16B)
+#define M6HROW 32   // region H row bytes (2 x 16B pair-chunks)
+#define M6LSZ (M6BM * M6LROW)
+#define M6HSZ (M6BM * M6HROW)
+
+__device__ __forceinline__ v8i6 v6_read(const unsigned char* lbase,
+                                        const unsigned char* hbase,
+                                        u32 row, u32 frag) {
+  u32 lslot = (frag ^ ((row >> 2) & 3u)) * 16u;
+  u32 hslot = ((frag >> 1) ^ ((row >> 3) & 1u)) * 16u + (frag & 1u) * 8u;
+  v4i6 lo = *(const v4i6*)(lbase + row * M6LROW + lslot);  // ds_read_b128
+  v2i6 hi = *(const v2i6*)(hbase + row * M6HROW + hslot);  // ds_read_b64
   v8i6 r = {};
   r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
   r[4] = hi[0]; r[5] = hi[1];
   return r;
 }
-
-#define M6BM 256
-#define M6BN 256
-#define M6ROWB 96
 
 __global__ void
 __launch_bounds__(512)
@@ -49,8 +70,10 @@ gemm_bt_mx6_acc_k128(float* __restrict__ C,
                      const unsigned char* __restrict__ Bsc, int M, int N,
                      int K, int beta0, u32 stm_, u32 stn_) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  unsigned char* As = (unsigned char*)smem;  // [2][256*96]
-  unsigned char* Bs = (unsigned char*)(smem + 2 * M6BM * M6ROWB);
+  unsigned char* AsL = (unsigned char*)smem;            // [2][M6LSZ]
+  unsigned char* BsL = (unsigned char*)(smem + 2 * M6LSZ);
+  unsigned char* AsH = (unsigned char*)(smem + 4 * M6LSZ);  // [2][M6HSZ]
+  unsigned char* BsH = (unsigned char*)(smem + 4 * M6LSZ + 2 * M6HSZ);
 
   const u32 tid = threadIdx.x;
   const u32 lane = tid & 63, wave = tid >> 6;
@@ -67,20 +90,32 @@ gemm_bt_mx6_acc_k128(float* __restrict__ C,
   }
   const long m0 = (long)tm * M6BM, n0 = (long)tn * M6BN;
   const long KB6 = (long)K * 3 / 4;  // packed row bytes
+  const long KL = (long)K / 2;       // L-plane bytes per row
 
-  // staging: tile = 256 rows x 96 B = 1536 x 16-B chunks; 512 threads
-  // x 3 passes per operand.
+  // staging per operand: L = 256 rows x 4 16-B chunks (2 passes of
+  // 512), H = 256 rows x 2 16-B chunks (1 pass). LDS destinations are
+  // strictly LINEAR (glds16 writes wave-base + lane*16); the swizzles
+  // permute the GLOBAL source chunk each lane fetches instead (XOR is
+  // self-inverse, so the read-side slot formulas recover frag order).
+  const u32 lrow = tid >> 2, lslot_st = tid & 3;
+  const u32 lsrc = lslot_st ^ ((lrow >> 2) & 3u);
+  const u32 hrow = tid >> 1, hslot_st = tid & 1;
+  const u32 hsrc = hslot_st ^ ((hrow >> 3) & 1u);
   auto stage = [&](int buf, int t) {
-    const long k0b = (long)t * 96;  // K-tile of 128 = 96 packed bytes
+    const long l0 = (long)t * 64;            // L bytes of this K-tile
+    const long h0 = KL + (long)t * 32;       // H bytes of this K-tile
 #pragma unroll
-    for (int p = 0; p < 3; ++p) {
-      u32 c = tid + (u32)p * 512;
-      u32 row = c / 6, slot = c - row * 6;
-      glds16(A + (m0 + row) * KB6 + k0b + slot * 16,
-             As + buf * M6BM * M6ROWB + row * M6ROWB + slot * 16);
-      glds16(Bt + (n0 + row) * KB6 + k0b + slot * 16,
-             Bs + buf * M6BM * M6ROWB + row * M6ROWB + slot * 16);
+    for (int p = 0; p < 2; ++p) {
+      u32 row = lrow + (u32)p * 128;
+      glds16(A + (m0 + row) * KB6 + l0 + lsrc * 16,
+             AsL + buf * M6LSZ + row * M6LROW + lslot_st * 16u);
+      glds16(Bt + (n0 + row) * KB6 + l0 + lsrc * 16,
+             BsL + buf * M6LSZ + row * M6LROW + lslot_st * 16u);
     }
+    glds16(A + (m0 + hrow) * KB6 + h0 + hsrc * 16,
+           AsH + buf * M6HSZ + hrow * M6HROW + hslot_st * 16u);
+    glds16(Bt + (n0 + hrow) * KB6 + h0 + hsrc * 16,
+           BsH + buf * M6HSZ + hrow * M6HROW + hslot_st * 16u);
   };
 
   f32x16h acc[4][2] = {};
@@ -112,23 +147,18 @@ gemm_bt_mx6_acc_k128(float* __restrict__ C,
       stage(cur ^ 1, t + 1);
       load_scales(t + 1, sa_nxt, sb_nxt);
     }
-    const unsigned char* a_base = As + cur * M6BM * M6ROWB;
-    const unsigned char* b_base = Bs + cur * M6BM * M6ROWB;
+    const unsigned char* al = AsL + cur * M6LSZ;
+    const unsigned char* bl = BsL + cur * M6LSZ;
+    const unsigned char* ah = AsH + cur * M6HSZ;
+    const unsigned char* bh = BsH + cur * M6HSZ;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
+      const u32 frag = (u32)kk * 2u + kh;
       v8i6 af[4], bf[2];
 #pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        u32 row = arow + m * 32;
-        u32 byte = row * M6ROWB + ((u32)kk * 2u + kh) * 24u;
-        af[m] = v6_from24((const char*)a_base + byte);
-      }
+      for (int m = 0; m < 4; ++m) af[m] = v6_read(al, ah, arow + m * 32, frag);
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
-        u32 row = brow + n * 32;
-        u32 byte = row * M6ROWB + ((u32)kk * 2u + kh) * 24u;
-        bf[n] = v6_from24((const char*)b_base + byte);
-      }
+      for (int n = 0; n < 2; ++n) bf[n] = v6_read(bl, bh, brow + n * 32, frag);
 #pragma unroll
       for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -176,7 +206,7 @@ extern "C" void launch_gemm_bt_mx6_acc(float* C, const void* A,
                                        int K, int beta0,
                                        hipStream_t stream) {
   dim3 grid((M / M6BM) * (N / M6BN));
-  size_t lds = 4u * M6BM * M6ROWB;  // 96 KiB
+  size_t lds = 4u * M6LSZ + 4u * M6HSZ;  // 96 KiB
   hipLaunchKernelGGL(gemm_bt_mx6_acc_k128, grid, dim3(512), lds, stream,
                      C, (const unsigned char*)A,
                      (const unsigned char*)Asc, (const unsigned char*)Bt,

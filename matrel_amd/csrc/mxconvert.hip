@@ -267,11 +267,16 @@ __global__ void mx6_convert_k(unsigned char* __restrict__ out6,
     w[d] |= code << off;
     if (off > 26) w[d + 1] |= code >> (32 - off);
   }
-  unsigned char* dst = out6 + r * (cols * 3 / 4) + b * 24;
+  // SPLIT-PLANE row layout: [cols/32 x 16B] (payload bytes 0-15 of each
+  // 24-B block = mfma regs 0-3) then [cols/32 x 8B] (bytes 16-23 = regs
+  // 4-5). Keeps every GEMM LDS read power-of-2 so both the b128 and b64
+  // phases can be laid out bank-conflict-free (24-B-contiguous blocks
+  // force a 24-dword lane stride that collides 4-way on the b64 reads).
+  unsigned char* rowp = out6 + r * (cols * 3 / 4);
+  typedef unsigned int u32x4c __attribute__((ext_vector_type(4)));
   typedef unsigned int u32x2c __attribute__((ext_vector_type(2)));
-  ((u32x2c*)dst)[0] = u32x2c{w[0], w[1]};
-  ((u32x2c*)dst)[1] = u32x2c{w[2], w[3]};
-  ((u32x2c*)dst)[2] = u32x2c{w[4], w[5]};
+  *(u32x4c*)(rowp + b * 16) = u32x4c{w[0], w[1], w[2], w[3]};
+  *(u32x2c*)(rowp + cols / 2 + b * 8) = u32x2c{w[4], w[5]};
 }
 
 extern "C" void launch_mx6_convert(void* out6, void* sc, const void* in,
