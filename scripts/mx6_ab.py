@@ -77,4 +77,27 @@ if __name__ == "__main__":
     perf(8192, it=6)
     perf(16384, it=3)
     perf(32768, it=3)
+    if os.environ.get("MX6_ST_SWEEP") == "1":
+        st_sweep(16384)
+        st_sweep(32768)
     print("done")
+
+
+def st_sweep(n, it=3):
+    a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
+    bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
+    a6, asc = blockops.to_mx6(a)
+    bt6, btsc = blockops.to_mx6(bt)
+    c = torch.zeros(n, n, dtype=torch.float32, device="cuda")
+    for sm, sn in ((4, 4), (8, 4), (4, 8), (8, 8), (2, 8), (16, 4)):
+        ext.set_supertile(sm, sn)
+        ext.gemm_bt_mx6_acc(c, a6, asc, bt6, btsc, True)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(it):
+            ext.gemm_bt_mx6_acc(c, a6, asc, bt6, btsc, True)
+        torch.cuda.synchronize()
+        tf = 2 * n ** 3 / ((time.perf_counter() - t0) / it) / 1e12
+        print(f"mx6 st({sm},{sn}) n={n}: {tf:.0f} TF", flush=True)
+    ext.set_supertile(4, 4)
+    del a, bt, a6, asc, bt6, btsc, c
