@@ -66,23 +66,6 @@ def conv_perf(n, it=5):
     del a
 
 
-if __name__ == "__main__":
-    ok = True
-    for n in (256, 512, 4096):
-        ok = screen(n) and ok
-    if not ok:
-        print("SCREEN FAILED")
-        sys.exit(1)
-    conv_perf(16384)
-    perf(8192, it=6)
-    perf(16384, it=3)
-    perf(32768, it=3)
-    if os.environ.get("MX6_ST_SWEEP") == "1":
-        st_sweep(16384)
-        st_sweep(32768)
-    print("done")
-
-
 def st_sweep(n, it=3):
     a = torch.randn(n, n, device="cuda").to(torch.bfloat16)
     bt = torch.randn(n, n, device="cuda").to(torch.bfloat16)
@@ -101,3 +84,20 @@ def st_sweep(n, it=3):
         print(f"mx6 st({sm},{sn}) n={n}: {tf:.0f} TF", flush=True)
     ext.set_supertile(4, 4)
     del a, bt, a6, asc, bt6, btsc, c
+
+
+if __name__ == "__main__":
+    ok = True
+    for n in (256, 512, 4096):
+        ok = screen(n) and ok
+    if not ok:
+        print("SCREEN FAILED")
+        sys.exit(1)
+    conv_perf(16384)
+    perf(8192, it=6)
+    perf(16384, it=3)
+    perf(32768, it=3)
+    if os.environ.get("MX6_ST_SWEEP") == "1":
+        st_sweep(16384)
+        st_sweep(32768)
+    print("done")
