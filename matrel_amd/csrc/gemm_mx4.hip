@@ -184,6 +184,13 @@ gemm_bt_mx4_acc_k128(float* __restrict__ C,
     }
 }
 
+__global__ void gemm_bt_mx4_acc_k128r3(float* __restrict__,
+                                        const unsigned char* __restrict__,
+                                        const unsigned char* __restrict__,
+                                        const unsigned char* __restrict__,
+                                        const unsigned char* __restrict__,
+                                        int, int, int, int, u32, u32);
+
 extern "C" void launch_gemm_bt_mx4_acc(float* C, const void* A,
                                        const void* Asc, const void* Bt,
                                        const void* Bsc, int M, int N,
@@ -196,4 +203,197 @@ extern "C" void launch_gemm_bt_mx4_acc(float* C, const void* A,
                      (const unsigned char*)Asc, (const unsigned char*)Bt,
                      (const unsigned char*)Bsc, M, N, K, beta0, g_st_m,
                      g_st_n);
+}
+
+// ------------------------- triple-buffered ring (overlapped staging) ----
+// The double-buffered kernel above serializes staging with compute: the
+// compiler's memory legalizer inserts s_waitcnt vmcnt(0) before the
+// fragment ds_reads (it cannot prove the in-flight global_load_lds
+// writes target the OTHER buffer), so every K-tile pays
+// mfma-time + staging-time. Measured decomposition at 16384^3
+// (scripts/mx4_floor_ab.py): full 2865 TF vs staging-removed 4358 TF.
+// This variant stages tile t+2 during tile t through a 3-buffer LDS
+// ring (96 KiB) with v6-style counted waits (gemm.hip): per tile,
+//   reads (v6 phase order: ds_reads BEFORE this tile's DMA issue) ->
+//   [scales t+1 (6 dwords), stage t+2 (4 glds)] -> mfmas ->
+//   s_waitcnt vmcnt(4) (proves S_{t+1}+G_{t+1}, leaves G_{t+2} in
+//   flight - never drains to 0 in the main loop) -> raw s_barrier.
+// The wait is UNCONDITIONAL straight-line code (a branchy wait splits
+// the block and the legalizer's conservative merge re-adds a vmcnt(0)
+// drain) and carries no "memory" clobber (the legalizer treats
+// clobbering asm as a memory access and drains before it). Scale banks
+// alternate by tile parity (2x unroll, literal banks): register copies
+// or %2 indexing both re-serialize (the copies make the legalizer
+// drain for the pending scale loads; dynamic indexing spills).
+// Buffer identity rotates through three pointer pairs.
+// The end-of-tile barrier separates tile t's reads of its buffer from
+// tile t+1's DMA into that same buffer (t+3 = t mod 3). Sync-structure
+// change => race-screened (multi-run bit-exactness vs the
+// double-buffered kernel); MATREL_MX4_V=1 keeps the old kernel.
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx4_acc_k128r3(float* __restrict__ C,
+                       const unsigned char* __restrict__ A,
+                       const unsigned char* __restrict__ Asc,
+                       const unsigned char* __restrict__ Bt,
+                       const unsigned char* __restrict__ Bsc, int M,
+                       int N, int K, int beta0, u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned char* As3 = (unsigned char*)smem;              // [3][16 KiB]
+  unsigned char* Bs3 = (unsigned char*)(smem + 3 * 16384);
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;
+
+  const u32 ntm = (u32)(M / M4BM), ntn = (u32)(N / M4BN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * M4BM, n0 = (long)tn * M4BN;
+  const long KB2 = K / 2;
+
+  const u32 srow = tid >> 2;
+  const u32 sslot = tid & 3;
+  const u32 src_slot = sslot ^ ((srow >> 2) & 3);
+
+  // per-thread staging source pointers as STABLE loop-carried registers
+  // STEPPED by += 64 per tile (no per-call offset addition): an addr
+  // temp computed per tile gets register-reused as a ds_read dest, and
+  // the waitcnt pass then protects the pending DMA's address regs with
+  // a vmcnt(0) drain - the serialization this kernel exists to remove.
+  // per-thread staging source pointers as STABLE loop-carried registers
+  // STEPPED by += 64 per tile (no per-call offset addition): an addr
+  // temp computed per tile gets register-reused as a ds_read dest, and
+  // the waitcnt pass then protects the pending DMA's address regs with
+  // a vmcnt(0) drain in the middle of the tile.
+  const unsigned char* sa0 = A + (m0 + srow) * KB2 + src_slot * 16;
+  const unsigned char* sa1 = A + (m0 + srow + 128) * KB2 + src_slot * 16;
+  const unsigned char* sb0 = Bt + (n0 + srow) * KB2 + src_slot * 16;
+  const unsigned char* sb1 = Bt + (n0 + srow + 128) * KB2 + src_slot * 16;
+
+  f32x16q acc[4][2] = {};
+  const u32 arow = wm * 128 + (lane & 31);
+  const u32 brow = wn * 64 + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  auto loadsc = [&](int t, int* sa_w, int* sb_w) {
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      sa_w[m] = ((const int*)Asc)[(long)t * M + m0 + arow + m * 32];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+      sb_w[n] = ((const int*)Bsc)[(long)t * N + n0 + brow + n * 32];
+  };
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  const int NT = K / 128;
+  // buffer ring as three wave-uniform 32-bit offsets (SGPRs; 64-bit
+  // pointer pairs cost 12 VGPRs and spilled the loop)
+  u32 oc = 0, on = 16384, os2 = 32768;
+  int sa[4], sb[2];
+
+  // prologue: [G0, G1] in flight; scales are loaded per tile (no
+  // prefetch bank: bank rotation needs copies or a 2x unroll, and both
+  // re-serialize - copies make the legalizer drain for the pending
+  // loads, the unroll spills; a per-tile scale load costs only its L2
+  // latency and its counted wait leaves the DMA ring in flight).
+  auto stage_to = [&](u32 buf, long k0b) {
+    glds16(sa0 + k0b, As3 + buf + srow * 64u + sslot * 16u);
+    glds16(sa1 + k0b, As3 + buf + (srow + 128) * 64u + sslot * 16u);
+    glds16(sb0 + k0b, Bs3 + buf + srow * 64u + sslot * 16u);
+    glds16(sb1 + k0b, Bs3 + buf + (srow + 128) * 64u + sslot * 16u);
+  };
+  stage_to(oc, 0);
+  if (NT > 1) stage_to(on, 64);
+  // pre-advance to tile 2: in-loop staging reads the pointers directly
+  sa0 += 128; sa1 += 128; sb0 += 128; sb1 += 128;
+  asm volatile("s_waitcnt vmcnt(4)");  // tile 0 proven (G1 in flight)
+  __builtin_amdgcn_s_barrier();
+
+#define MX4R3_PHASE(KK, ISSUE)                                              \
+  {                                                                         \
+    v8i4 af[4], bf[2];                                                      \
+    _Pragma("unroll")                                                       \
+    for (int m = 0; m < 4; ++m) {                                           \
+      u32 row = arow + m * 32;                                              \
+      u32 byte = row * 64u +                                                \
+                 (((u32)(KK) * 2u + kh) ^ ((row >> 2) & 3)) * 16u;          \
+      af[m] = v4_from16((const char*)As3 + oc + byte);                      \
+    }                                                                       \
+    _Pragma("unroll")                                                       \
+    for (int n = 0; n < 2; ++n) {                                           \
+      u32 row = brow + n * 32;                                              \
+      u32 byte = row * 64u +                                                \
+                 (((u32)(KK) * 2u + kh) ^ ((row >> 2) & 3)) * 16u;          \
+      bf[n] = v4_from16((const char*)Bs3 + oc + byte);                      \
+    }                                                                       \
+    ISSUE;                                                                  \
+    _Pragma("unroll")                                                       \
+    for (int m = 0; m < 4; ++m)                                             \
+      _Pragma("unroll")                                                     \
+      for (int n = 0; n < 2; ++n)                                           \
+        acc[m][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(        \
+            af[m], bf[n], acc[m][n], 4, 4, 0,                               \
+            (sa[m] >> (8 * ((KK) * 2 + (int)kh))) & 0xFF, 0,                \
+            (sb[n] >> (8 * ((KK) * 2 + (int)kh))) & 0xFF);                  \
+  }
+
+  loadsc(0, sa, sb);
+  // tile body: reads+mfmas first; ALL new memory issue at tile END so
+  // every glds -> ds_read path crosses the barrier (the structure the
+  // legalizer accepts without a mid-tile vmcnt(0) drain; scales BEFORE
+  // glds so next tile's scale extraction is a counted wait). The
+  // pointer steps on the backedge still cost one vmcnt(0) there -
+  // but by then the staging DMA has had the whole mfma cluster to
+  // complete, which is the overlap this kernel exists to create.
+  for (int t = 0; t < NT; ++t) {
+    MX4R3_PHASE(0, {});
+    __builtin_amdgcn_sched_barrier(0);
+    MX4R3_PHASE(1, {});
+    __builtin_amdgcn_sched_barrier(0);
+    if (t + 1 < NT) loadsc(t + 1, sa, sb);
+    if (t + 2 < NT) {
+      glds16(sa0, As3 + os2 + srow * 64u + sslot * 16u);
+      glds16(sa1, As3 + os2 + (srow + 128) * 64u + sslot * 16u);
+      glds16(sb0, Bs3 + os2 + srow * 64u + sslot * 16u);
+      glds16(sb1, Bs3 + os2 + (srow + 128) * 64u + sslot * 16u);
+    }
+    __builtin_amdgcn_sched_barrier(0);
+    asm volatile("s_waitcnt vmcnt(10)");
+    __builtin_amdgcn_s_barrier();
+    u32 tmp_ = oc; oc = on; on = os2; os2 = tmp_;
+    sa0 += 64; sa1 += 64; sb0 += 64; sb1 += 64;
+  }
+#undef MX4R3_PHASE
+
+  const long ccol = n0 + wn * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      float cv[16];
+      if (!beta0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                     8 * (r >> 2) + 4 * (lane >> 5);
+          cv[r] = C[row * (long)N + ccol + n * 32];
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                   8 * (r >> 2) + 4 * (lane >> 5);
+        C[row * (long)N + ccol + n * 32] =
+            (beta0 ? 0.0f : cv[r]) + acc[m][n][r];
+      }
+    }
 }

@@ -18,21 +18,20 @@
 // read widths power-of-2:
 //   region L: [row][4 x 16B slots], slot XOR-swizzled by (row>>2)&3
 //     -> the mx4 kernel's proven conflict-free b128 pattern;
-//   region H: [row][2 x 16B pair-chunks], pair-chunk XOR-swizzled by
-//     (row>>3)&1 -> the b64 phase drops from 4-way to 2-way conflicts
-//     (2-way is the floor here: glds16 DMA destinations must be
-//     wave-uniform-base + lane*16, so LDS rows must be a linear 32 B
-//     and only whole 16-B pair-chunks can be permuted - via the GLOBAL
-//     source address, which is per-lane free).
+//   region H: [frag][256 rows][8B] mirroring the global H plane's
+//     BLOCK-major order - staging reads each K-tile's H bytes as four
+//     contiguous 2-KiB runs (full-128-B-line coalesced DMA; a
+//     row-major H plane fragments into 32-B segments and cost ~25%
+//     end to end - see profiles/README.md), and the b64 reads sit on
+//     a conflict-free 2-dword lane stride with no swizzle at all.
 // A 24-B-contiguous layout instead puts ALL reads at a 24-dword lane
-// stride (2-way b128 + 4-way b64 conflicts) and measured only 1505 TF;
-// the split leaves the inner loop MFMA-bound (b64 is 1/3 of the LDS
-// traffic).
+// stride (2-way b128 + 4-way b64 conflicts) and measured only 1505 TF.
 //
 // Structure: the k128 shape - 256^2 tile, 8 waves, K-tile 128,
 // double-buffered LDS (2 x 2 x (16 KiB + 8 KiB) = 96 KiB), one
 // barrier per tile, scale rotation.
 #include "common.h"
+#include <cstdlib>
 
 extern unsigned g_st_m, g_st_n;  // supertile dims (gemm.hip)
 
@@ -44,23 +43,27 @@ typedef float f32x16h __attribute__((ext_vector_type(16)));
 #define M6BM 256
 #define M6BN 256
 #define M6LROW 64   // region L row bytes (4 x 16B)
-#define M6HROW 32   // region H row bytes (2 x 16B pair-chunks)
+#define M6HFRAG 2048  // region H bytes per frag (256 rows x 8B)
 #define M6LSZ (M6BM * M6LROW)
-#define M6HSZ (M6BM * M6HROW)
+#define M6HSZ (4 * M6HFRAG)
 
+template <int NOHI>
 __device__ __forceinline__ v8i6 v6_read(const unsigned char* lbase,
                                         const unsigned char* hbase,
                                         u32 row, u32 frag) {
   u32 lslot = (frag ^ ((row >> 2) & 3u)) * 16u;
-  u32 hslot = ((frag >> 1) ^ ((row >> 3) & 1u)) * 16u + (frag & 1u) * 8u;
   v4i6 lo = *(const v4i6*)(lbase + row * M6LROW + lslot);  // ds_read_b128
-  v2i6 hi = *(const v2i6*)(hbase + row * M6HROW + hslot);  // ds_read_b64
   v8i6 r = {};
   r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
-  r[4] = hi[0]; r[5] = hi[1];
+  if (NOHI == 0) {  // NOHI>=1: perf probe only - result is WRONG (regs 4-5
+                // zero); isolates the b64 stream's cost
+    v2i6 hi = *(const v2i6*)(hbase + frag * M6HFRAG + row * 8u);  // ds_read_b64
+    r[4] = hi[0]; r[5] = hi[1];
+  }
   return r;
 }
 
+template <int NOHI>
 __global__ void
 __launch_bounds__(512)
 gemm_bt_mx6_acc_k128(float* __restrict__ C,
@@ -89,8 +92,9 @@ gemm_bt_mx6_acc_k128(float* __restrict__ C,
     tn = bid % ntn;
   }
   const long m0 = (long)tm * M6BM, n0 = (long)tn * M6BN;
-  const long KB6 = (long)K * 3 / 4;  // packed row bytes
   const long KL = (long)K / 2;       // L-plane bytes per row
+  const long HA = (long)M * KL;      // A's H-plane base (block-major)
+  const long HB = (long)N * KL;      // Bt's H-plane base
 
   // staging per operand: L = 256 rows x 4 16-B chunks (2 passes of
   // 512), H = 256 rows x 2 16-B chunks (1 pass). LDS destinations are
@@ -99,23 +103,24 @@ gemm_bt_mx6_acc_k128(float* __restrict__ C,
   // self-inverse, so the read-side slot formulas recover frag order).
   const u32 lrow = tid >> 2, lslot_st = tid & 3;
   const u32 lsrc = lslot_st ^ ((lrow >> 2) & 3u);
-  const u32 hrow = tid >> 1, hslot_st = tid & 1;
-  const u32 hsrc = hslot_st ^ ((hrow >> 3) & 1u);
+  const u32 hblk = tid >> 7, hi_ = tid & 127;  // chunk = rows 2i, 2i+1
   auto stage = [&](int buf, int t) {
     const long l0 = (long)t * 64;            // L bytes of this K-tile
-    const long h0 = KL + (long)t * 32;       // H bytes of this K-tile
 #pragma unroll
     for (int p = 0; p < 2; ++p) {
       u32 row = lrow + (u32)p * 128;
-      glds16(A + (m0 + row) * KB6 + l0 + lsrc * 16,
+      glds16(A + (m0 + row) * KL + l0 + lsrc * 16,
              AsL + buf * M6LSZ + row * M6LROW + lslot_st * 16u);
-      glds16(Bt + (n0 + row) * KB6 + l0 + lsrc * 16,
+      glds16(Bt + (n0 + row) * KL + l0 + lsrc * 16,
              BsL + buf * M6LSZ + row * M6LROW + lslot_st * 16u);
     }
-    glds16(A + (m0 + hrow) * KB6 + h0 + hsrc * 16,
-           AsH + buf * M6HSZ + hrow * M6HROW + hslot_st * 16u);
-    glds16(Bt + (n0 + hrow) * KB6 + h0 + hsrc * 16,
-           BsH + buf * M6HSZ + hrow * M6HROW + hslot_st * 16u);
+    if (NOHI < 2) {  // NOHI=2 perf probe also drops the H staging DMA
+      const long hb = (long)(4 * t + hblk);
+      glds16(A + HA + (hb * M + m0 + hi_ * 2) * 8,
+             AsH + buf * M6HSZ + tid * 16u);
+      glds16(Bt + HB + (hb * N + n0 + hi_ * 2) * 8,
+             BsH + buf * M6HSZ + tid * 16u);
+    }
   };
 
   f32x16h acc[4][2] = {};
@@ -156,9 +161,11 @@ gemm_bt_mx6_acc_k128(float* __restrict__ C,
       const u32 frag = (u32)kk * 2u + kh;
       v8i6 af[4], bf[2];
 #pragma unroll
-      for (int m = 0; m < 4; ++m) af[m] = v6_read(al, ah, arow + m * 32, frag);
+      for (int m = 0; m < 4; ++m)
+        af[m] = v6_read<NOHI>(al, ah, arow + m * 32, frag);
 #pragma unroll
-      for (int n = 0; n < 2; ++n) bf[n] = v6_read(bl, bh, brow + n * 32, frag);
+      for (int n = 0; n < 2; ++n)
+        bf[n] = v6_read<NOHI>(bl, bh, brow + n * 32, frag);
 #pragma unroll
       for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -207,9 +214,23 @@ extern "C" void launch_gemm_bt_mx6_acc(float* C, const void* A,
                                        hipStream_t stream) {
   dim3 grid((M / M6BM) * (N / M6BN));
   size_t lds = 4u * M6LSZ + 4u * M6HSZ;  // 96 KiB
-  hipLaunchKernelGGL(gemm_bt_mx6_acc_k128, grid, dim3(512), lds, stream,
-                     C, (const unsigned char*)A,
-                     (const unsigned char*)Asc, (const unsigned char*)Bt,
-                     (const unsigned char*)Bsc, M, N, K, beta0, g_st_m,
-                     g_st_n);
+  const char* probe = getenv("MATREL_MX6_PROBE_NOHI");
+  if (probe && probe[0] == '2')  // perf probe: WRONG results
+    hipLaunchKernelGGL(gemm_bt_mx6_acc_k128<2>, grid, dim3(512), lds,
+                       stream, C, (const unsigned char*)A,
+                       (const unsigned char*)Asc, (const unsigned char*)Bt,
+                       (const unsigned char*)Bsc, M, N, K, beta0, g_st_m,
+                       g_st_n);
+  else if (probe)  // perf probe: WRONG results
+    hipLaunchKernelGGL(gemm_bt_mx6_acc_k128<1>, grid, dim3(512), lds,
+                       stream, C, (const unsigned char*)A,
+                       (const unsigned char*)Asc, (const unsigned char*)Bt,
+                       (const unsigned char*)Bsc, M, N, K, beta0, g_st_m,
+                       g_st_n);
+  else
+    hipLaunchKernelGGL(gemm_bt_mx6_acc_k128<0>, grid, dim3(512), lds,
+                       stream, C, (const unsigned char*)A,
+                       (const unsigned char*)Asc, (const unsigned char*)Bt,
+                       (const unsigned char*)Bsc, M, N, K, beta0, g_st_m,
+                       g_st_n);
 }

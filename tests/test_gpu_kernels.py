@@ -612,12 +612,13 @@ def test_mx4_end_to_end():
 def _dequant_e2m3(p6: torch.Tensor, sc: torch.Tensor, rows: int,
                   cols: int) -> torch.Tensor:
     """Host dequant of split-plane e2m3 6-bit fields (LSB-first) +
-    packed-dword e8m0 scales. Row layout: [cols/32 x 16B] block bytes
-    0-15, then [cols/32 x 8B] block bytes 16-23 (see mxconvert.hip)."""
-    raw = p6.cpu().to(torch.int64).reshape(rows, cols * 3 // 4)
+    packed-dword e8m0 scales. Layout: L plane row-major
+    [rows][cols/32 x 16B] (block bytes 0-15), then H plane BLOCK-major
+    [cols/32][rows][8B] (bytes 16-23) - see mxconvert.hip."""
     nb = cols // 32
-    lo = raw[:, : nb * 16].reshape(rows, nb, 16)
-    hi = raw[:, nb * 16:].reshape(rows, nb, 8)
+    flat = p6.cpu().to(torch.int64).reshape(-1)
+    lo = flat[: rows * nb * 16].reshape(rows, nb, 16)
+    hi = flat[rows * nb * 16:].reshape(nb, rows, 8).permute(1, 0, 2)
     b = torch.cat([lo, hi], dim=-1).reshape(rows, cols * 3 // 4)
     bits = torch.stack([(b >> i) & 1 for i in range(8)], dim=-1)
     bits = bits.reshape(rows, cols, 6)
