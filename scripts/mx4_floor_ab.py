@@ -1,5 +1,7 @@
-"""mx4 staging-vs-compute decomposition probe (results WRONG in probe
-modes; perf only)."""
+"""mx4 staging-vs-compute decomposition probe. HISTORICAL: the NOSTAGE
+kernel path was removed after the measurement (full 2865 / floor 4358
+TF at 16384^3, recorded in profiles/README.md \"ring-kernel study\");
+without it this script now measures the full kernel twice."""
 import os, sys, time
 import torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
