@@ -190,6 +190,12 @@ __global__ void gemm_bt_mx4_acc_k128r3(float* __restrict__,
                                         const unsigned char* __restrict__,
                                         const unsigned char* __restrict__,
                                         int, int, int, int, u32, u32);
+__global__ void gemm_bt_mx4_acc_k256(float* __restrict__,
+                                     const unsigned char* __restrict__,
+                                     const unsigned char* __restrict__,
+                                     const unsigned char* __restrict__,
+                                     const unsigned char* __restrict__,
+                                     int, int, int, int, u32, u32);
 
 extern "C" void launch_gemm_bt_mx4_acc(float* C, const void* A,
                                        const void* Asc, const void* Bt,
@@ -373,6 +379,156 @@ gemm_bt_mx4_acc_k128r3(float* __restrict__ C,
     sa0 += 64; sa1 += 64; sb0 += 64; sb1 += 64;
   }
 #undef MX4R3_PHASE
+
+  const long ccol = n0 + wn * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      float cv[16];
+      if (!beta0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                     8 * (r >> 2) + 4 * (lane >> 5);
+          cv[r] = C[row * (long)N + ccol + n * 32];
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        long row = m0 + arow + m * 32 - (lane & 31) + (r & 3) +
+                   8 * (r >> 2) + 4 * (lane >> 5);
+        C[row * (long)N + ccol + n * 32] =
+            (beta0 ? 0.0f : cv[r]) + acc[m][n][r];
+      }
+    }
+}
+
+// ------------------------- K-tile 256 (halved per-tile sync) ----------
+// Same proven k128 LDS layout, but each buffer holds TWO k128 sub-tiles
+// (2 x 2 x 32 KiB = 128 KiB): one barrier + one staging drain per 256
+// of K instead of per 128, halving the per-K sync overhead (the
+// request-throughput study bounds staging itself; this trims the fixed
+// cost 'a'). MATREL_MX4_V=2. Race-screened bit-exact vs the default.
+__global__ void
+__launch_bounds__(512)
+gemm_bt_mx4_acc_k256(float* __restrict__ C,
+                     const unsigned char* __restrict__ A,
+                     const unsigned char* __restrict__ Asc,
+                     const unsigned char* __restrict__ Bt,
+                     const unsigned char* __restrict__ Bsc, int M,
+                     int N, int K, int beta0, u32 stm_, u32 stn_) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned char* As = (unsigned char*)smem;  // [2][2][16 KiB]
+  unsigned char* Bs = (unsigned char*)(smem + 2 * 32768);
+
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & 63, wave = tid >> 6;
+  const u32 wm = wave >> 2, wn = wave & 3;
+
+  const u32 ntm = (u32)(M / M4BM), ntn = (u32)(N / M4BN);
+  u32 bid = xcd_swizzle(blockIdx.x, ntm * ntn);
+  u32 tm, tn;
+  if (stm_ && ntm % stm_ == 0 && ntn % stn_ == 0) {
+    supertile_map(bid, ntm, ntn, stm_, stn_, &tm, &tn);
+  } else {
+    tm = bid / ntn;
+    tn = bid % ntn;
+  }
+  const long m0 = (long)tm * M4BM, n0 = (long)tn * M4BN;
+  const long KB2 = K / 2;
+
+  const u32 srow = tid >> 2;
+  const u32 sslot = tid & 3;
+  const u32 src_slot = sslot ^ ((srow >> 2) & 3);
+
+  // stage one K-256 tile = two k128 sub-tiles per operand (8 glds)
+  auto stage = [&](int buf, int t) {
+    const long k0b = (long)t * 128;  // K-tile of 256 = 128 packed bytes
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        u32 row = srow + half * 128;
+        glds16(A + (m0 + row) * KB2 + k0b + sub * 64 + src_slot * 16,
+               As + buf * 32768u + sub * 16384u + row * 64u + sslot * 16u);
+        glds16(Bt + (n0 + row) * KB2 + k0b + sub * 64 + src_slot * 16,
+               Bs + buf * 32768u + sub * 16384u + row * 64u + sslot * 16u);
+      }
+  };
+
+  f32x16q acc[4][2] = {};
+  const u32 arow = wm * 128 + (lane & 31);
+  const u32 brow = wn * 64 + (lane & 31);
+  const u32 kh = lane >> 5;
+
+  // packed-dword scales: K-256 tile spans TWO scale words per row
+  auto load_scales = [&](int t, int* sa_w, int* sb_w) {
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        sa_w[q * 4 + m] =
+            ((const int*)Asc)[(long)(2 * t + q) * M + m0 + arow + m * 32];
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        sb_w[q * 2 + n] =
+            ((const int*)Bsc)[(long)(2 * t + q) * N + n0 + brow + n * 32];
+    }
+  };
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  int sa_cur[8], sb_cur[4], sa_nxt[8], sb_nxt[4];
+  load_scales(0, sa_cur, sb_cur);
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  const int NT = K / 256;
+  for (int t = 0; t < NT; ++t) {
+    const bool pre = (t + 1 < NT);
+    if (pre) {
+      stage(cur ^ 1, t + 1);
+      load_scales(t + 1, sa_nxt, sb_nxt);
+    }
+    const unsigned char* a_base = As + cur * 32768u;
+    const unsigned char* b_base = Bs + cur * 32768u;
+#pragma unroll
+    for (int kq = 0; kq < 4; ++kq) {  // sub-tile = kq>>1, inner kk = kq&1
+      const u32 sub = (u32)kq >> 1, kk = (u32)kq & 1;
+      v8i4 af[4], bf[2];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        u32 row = arow + m * 32;
+        u32 byte = sub * 16384u + row * 64u +
+                   ((kk * 2u + kh) ^ ((row >> 2) & 3)) * 16u;
+        af[m] = v4_from16((const char*)a_base + byte);
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        u32 row = brow + n * 32;
+        u32 byte = sub * 16384u + row * 64u +
+                   ((kk * 2u + kh) ^ ((row >> 2) & 3)) * 16u;
+        bf[n] = v4_from16((const char*)b_base + byte);
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af[m], bf[n], acc[m][n], 4, 4, 0,
+              (sa_cur[sub * 4 + m] >> (8 * ((int)kk * 2 + (int)kh))) & 0xFF,
+              0,
+              (sb_cur[sub * 2 + n] >> (8 * ((int)kk * 2 + (int)kh))) & 0xFF);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int m = 0; m < 8; ++m) sa_cur[m] = sa_nxt[m];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) sb_cur[n] = sb_nxt[n];
+    cur ^= 1;
+  }
 
   const long ccol = n0 + wn * 64 + (lane & 31);
 #pragma unroll
