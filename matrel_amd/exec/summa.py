@@ -126,10 +126,14 @@ def _mx8_cached(m: BlockMatrix, key, tensor_bf16: torch.Tensor,
     return ent
 
 
-def _mx8_usable(t: torch.Tensor) -> bool:
+def _mx8_usable(t: torch.Tensor, prec: str = "mx8") -> bool:
+    # mx8 has a legacy byte-scale layout + BK=64 kernel for cols % 32;
+    # the packed-scale mx6/mx4 converters and their k128 kernels need
+    # cols % 128 (a ragged edge of width 64 mod 128 passes the mx8 rule
+    # but raises in mx6_convert - caught by the refinement example).
+    kmod = 32 if prec == "mx8" else 128
     return (t.is_cuda and t.dtype == torch.bfloat16
-            and t.shape[0] % 256 == 0 and t.shape[1] % 64 == 0
-            and t.shape[1] % 32 == 0)
+            and t.shape[0] % 256 == 0 and t.shape[1] % kmod == 0)
 
 
 def _try_batch_local(a: BlockMatrix, b: BlockMatrix,
@@ -398,7 +402,7 @@ def local_matmul(a: BlockMatrix, b: BlockMatrix, out: BlockMatrix) -> None:
                 continue
             bbd = blockops.densify(bb)
             bt = _bt_cached(b, k, bj, bbd)
-            if mx8 and _mx8_usable(ab) and _mx8_usable(bt) \
+            if mx8 and _mx8_usable(ab, mxq) and _mx8_usable(bt, mxq) \
                     and ab.shape[0] % 256 == 0 and bt.shape[0] % 256 == 0:
                 a8, asc = _mx8_cached(a, (bi, k), ab, mxq)
                 bt8, btsc = _mx8_cached(b, ("t", k, bj), bt, mxq)

@@ -697,3 +697,28 @@ def test_mx6_end_to_end():
         assert rel < 0.06, rel
     finally:
         reset_runtime()
+
+
+def test_mxq_ragged_edge_64():
+    """Regression: a ragged edge block of width == 64 (mod 128) passes
+    the mx8 usability rule (cols % 32) but the packed-scale mx6/mx4
+    converters need cols % 128 - the engine must fall back to bf16 for
+    that block instead of raising (found by
+    examples/mixed_precision_refinement.py: 1e6 % 16384 = 576)."""
+    import matrel_amd as ma
+    from matrel_amd.exec.runtime import reset_runtime
+    for prec in ("mx6", "mx4"):
+        reset_runtime()
+        s = ma.MatrelSession(ma.MatrelConfig(block_size=512,
+                                             gemm_precision=prec))
+        try:
+            n = 2 * 512 + 64  # edge 64: % 32 == 0 but % 128 != 0
+            A = s.randn(1024, n, seed=80)
+            B = s.randn(n, 512, seed=81)
+            C = (A @ B).compute()
+            ref = A.to_tensor().float() @ B.to_tensor().float()
+            err = (C.to_tensor().float() - ref).abs()
+            sigma = (0.05 if prec == "mx6" else 0.17) * (n ** 0.5)
+            assert err.max().item() < 6 * sigma, err.max().item()
+        finally:
+            reset_runtime()
