@@ -102,7 +102,11 @@ def spgemm_acc(acc: torch.Tensor, a_csr: torch.Tensor,
             a_csr.col_indices().to(torch.int32), va.contiguous(),
             b_csr.crow_indices().to(torch.int32),
             b_csr.col_indices().to(torch.int32), vb.contiguous())
-    acc.add_(a_csr.to_dense().float() @ b_csr.to_dense().float())
+    d = a_csr.to_dense().float() @ b_csr.to_dense().float()
+    # ragged blocks: the product can be narrower/shorter than the
+    # (block-size-wide) accumulator slice - missing columns/rows are
+    # zero contributions (found by scripts/cpu_fuzz_sweep.py)
+    acc[: d.shape[0], : d.shape[1]].add_(d)
     return True
 
 

@@ -76,7 +76,7 @@ def _worker(rank, world, port, iters=25):
     # ewise densify, SUMMA CSR-wire panels, select masks, transpose
     leaves.append((s.sparse_random(N, N, 0.02, seed=4), None))
     leaves = [(m, m.to_tensor()) for m, _ in leaves]
-    rng = random.Random(77 + world)
+    rng = random.Random(int(os.environ.get("MATREL_FUZZ_SEED", "77")) + world)
     for i in range(iters):
         spec = _gen_spec(rng, 4)
         m, ref = _build(spec, leaves)

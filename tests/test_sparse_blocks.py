@@ -127,3 +127,22 @@ def test_sparse_end_to_end_gpu(gpu_session):
     out = (A @ B).compute()
     ref = A.to_tensor() @ B.to_tensor()
     assert (out.to_tensor() - ref).abs().max().item() < 0.5
+
+
+def test_both_sparse_ragged_blocks():
+    """Regression (cpu_fuzz_sweep): both-operands-sparse matmul with a
+    RAGGED trailing block (n % bs != 0) - the spgemm fallback must add
+    the narrower product into the padded accumulator slice."""
+    import matrel_amd as ma
+    from matrel_amd.exec.runtime import reset_runtime
+    reset_runtime()
+    ma.set_config(ma.MatrelConfig(block_size=24))
+    s = ma.MatrelSession()
+    try:
+        a = s.sparse_random(64, 64, 0.05, seed=11)
+        b = s.sparse_random(64, 64, 0.05, seed=12)
+        out = (a @ b).compute().to_tensor()
+        ref = a.to_tensor() @ b.to_tensor()
+        assert torch.allclose(out, ref, atol=2e-3, rtol=2e-3)
+    finally:
+        reset_runtime()
