@@ -77,8 +77,9 @@ def _worker(rank, world, port, iters=25):
     leaves.append((s.sparse_random(N, N, 0.02, seed=4), None))
     leaves = [(m, m.to_tensor()) for m, _ in leaves]
     rng = random.Random(int(os.environ.get("MATREL_FUZZ_SEED", "77")) + world)
+    depth = int(os.environ.get("MATREL_FUZZ_DEPTH", "4"))
     for i in range(iters):
-        spec = _gen_spec(rng, 4)
+        spec = _gen_spec(rng, depth)
         m, ref = _build(spec, leaves)
         out = m.compute().to_tensor()
         assert torch.allclose(out, ref, atol=2e-3, rtol=2e-3), \
