@@ -143,6 +143,8 @@ def test_both_sparse_ragged_blocks():
         b = s.sparse_random(64, 64, 0.05, seed=12)
         out = (a @ b).compute().to_tensor()
         ref = a.to_tensor() @ b.to_tensor()
-        assert torch.allclose(out, ref, atol=2e-3, rtol=2e-3)
+        tol = 2e-3 if ref.dtype == torch.float32 else 2e-2  # bf16 on GPU
+        assert torch.allclose(out.float(), ref.float(), atol=tol,
+                              rtol=tol)
     finally:
         reset_runtime()
