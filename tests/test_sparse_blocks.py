@@ -143,7 +143,7 @@ def test_both_sparse_ragged_blocks():
         b = s.sparse_random(64, 64, 0.05, seed=12)
         out = (a @ b).compute().to_tensor()
         ref = a.to_tensor() @ b.to_tensor()
-        tol = 2e-3 if ref.dtype == torch.float32 else 2e-2  # bf16 on GPU
+        tol = 2e-2 if out.is_cuda else 2e-3  # GPU sessions hold bf16
         assert torch.allclose(out.float(), ref.float(), atol=tol,
                               rtol=tol)
     finally:
