@@ -90,3 +90,10 @@ def _worker(rank, world, port, iters=30):
 
 def test_distributed_fuzz_shapes():
     mp.spawn(_worker, args=(2, 29840), nprocs=2, join=True)
+
+
+def test_distributed_fuzz_shapes_world8():
+    """Same shape-directed trees on the 4x2 grid (the driver's 8-GPU
+    SCALE shape) - vector/ragged operands across a grid where most
+    ranks own zero blocks of the small leaves."""
+    mp.spawn(_worker, args=(8, 29845, 12), nprocs=8, join=True)
