@@ -10,6 +10,8 @@ hand-written cases.
 
 import pytest
 import torch
+import os
+
 from hypothesis import given, settings, strategies as st
 
 import matrel_amd as ma
@@ -102,7 +104,8 @@ def _build(spec, leaves):
     raise ValueError(kind)
 
 
-@settings(max_examples=150, deadline=None, derandomize=True)
+@settings(max_examples=int(os.environ.get("MATREL_HYP_EXAMPLES", "150")),
+          deadline=None, derandomize=True)
 @given(spec=_expr(4))
 def test_random_expression_matches_torch(spec, _session):
     leaves = _leaves(_session)
