@@ -67,7 +67,8 @@ def _worker(rank, world, port, iters=30):
     import torch.distributed as dist
     dist.init_process_group("gloo", rank=rank, world_size=world)
     import matrel_amd as ma
-    ma.set_config(ma.MatrelConfig(block_size=24))
+    bs = int(os.environ.get("MATREL_FUZZ_BS", "24"))
+    ma.set_config(ma.MatrelConfig(block_size=bs))
     s = ma.MatrelSession()
     seed = int(os.environ.get("MATREL_FUZZ_SEED", "101")) + world
     rng = random.Random(seed)
